@@ -1,0 +1,156 @@
+"""LocalEngine: on-node prefill+decode for one opponent model.
+
+This is the MI355X replacement for the reference's remote `completion()`
+(SURVEY.md §2.4 K1): chat templating, single-shot long-context prefill,
+sampled decode with early stop on the protocol close tag, and exact token
+accounting from our own tokenizer (replacing the len//4 estimate,
+reference models.py:444-447).
+
+Engines are cached per (model name, weights, device): weights stay resident
+in HBM3E between rounds, so round k+1 pays only prefill+decode.
+"""
+
+from __future__ import annotations
+
+import threading
+import time
+from typing import Any, Optional
+
+import torch
+
+from .. import ops
+from ..models import LlamaModel, get_config
+from ..utils.timing import PhaseTimer
+from .tokenizer import build_tokenizer
+
+# `[/SPEC]` seen in the decoded tail stops decode early — the critique is
+# complete by protocol.
+_STOP_SUBSTR = "[/SPEC]"
+
+
+def _seed_from_name(name: str) -> int:
+    h = 2166136261
+    for ch in name:
+        h = ((h ^ ord(ch)) * 16777619) & 0x7FFFFFFF
+    return h
+
+
+class LocalEngine:
+    def __init__(self, spec: dict[str, Any], device: Optional[str] = None) -> None:
+        self.name = spec.get("name", spec["arch"])
+        self.arch = spec["arch"]
+        self.config = get_config(self.arch)
+        if device is None:
+            gpu = spec.get("gpu")
+            if torch.cuda.is_available():
+                device = f"cuda:{gpu}" if gpu is not None else "cuda:0"
+            else:
+                device = "cpu"
+        self.device = torch.device(device)
+        self.model = LlamaModel(
+            self.config, device=self.device, seed=_seed_from_name(self.name)
+        )
+        weights = spec.get("weights")
+        if weights:
+            self.model.load_safetensors(weights)
+            self.tokenizer = build_tokenizer(self.config.vocab_size, weights)
+        else:
+            self.model.init_random()
+            self.tokenizer = build_tokenizer(self.config.vocab_size)
+        self._gen_lock = threading.Lock()
+        self._sample_gen = torch.Generator().manual_seed(_seed_from_name(self.name) ^ 0x5EED)
+        self._seed_counter = _seed_from_name(self.name) ^ 0x5EED
+
+    def _fit_prompt(self, ids: list[int], reserve: int) -> list[int]:
+        """Clamp a prompt into the context window, dropping the middle."""
+        budget = self.config.max_seq_len - reserve - 8
+        if len(ids) <= budget:
+            return ids
+        head = budget * 2 // 3
+        tail = budget - head
+        return ids[:head] + ids[len(ids) - tail :]
+
+    def generate(
+        self,
+        system_prompt: str,
+        user_message: str,
+        max_tokens: int = 8000,
+        temperature: float = 0.7,
+        timeout: float = 600.0,
+        top_p: float = 1.0,
+    ) -> tuple[str, int, int, dict[str, float]]:
+        """Returns (text, input_tokens, output_tokens, phase timings ms)."""
+        with self._gen_lock:
+            return self._generate_locked(
+                system_prompt, user_message, max_tokens, temperature, timeout, top_p
+            )
+
+    def _generate_locked(self, system_prompt, user_message, max_tokens,
+                         temperature, timeout, top_p):
+        deadline = time.monotonic() + timeout
+        timer = PhaseTimer(sync_cuda=self.device.type == "cuda")
+        ids = self.tokenizer.render_chat(system_prompt, user_message)
+        reserve = min(max_tokens, self.config.max_seq_len // 2)
+        ids = self._fit_prompt(ids, reserve)
+        max_new = min(max_tokens, self.config.max_seq_len - len(ids) - 1)
+        stop_ids = self.tokenizer.stop_ids()
+
+        prev_device = None
+        if self.device.type == "cuda":
+            prev_device = torch.cuda.current_device()
+            torch.cuda.set_device(self.device)
+        try:
+            cache = self.model.new_cache(len(ids) + max_new + 8)
+            tokens = torch.tensor(ids, device=self.device, dtype=torch.long)
+            with timer.phase("prefill"):
+                logits = self.model.prefill(tokens, cache)
+
+            out_ids: list[int] = []
+            tail = ""  # rolling decoded tail for stop-substring detection
+            with timer.phase("decode"):
+                for _ in range(max_new):
+                    self._seed_counter = (self._seed_counter * 6364136223846793005 + 1) & 0x7FFFFFFF
+                    tok = ops.sample(
+                        logits, temperature=temperature, top_p=top_p,
+                        generator=self._sample_gen, seed=self._seed_counter,
+                    )
+                    if tok in stop_ids:
+                        break
+                    out_ids.append(tok)
+                    if 0 <= tok < 256:
+                        tail = (tail + chr(tok))[-16:]
+                        if tail.endswith(_STOP_SUBSTR):
+                            break
+                    if time.monotonic() > deadline:
+                        # deadline reached: return the partial critique
+                        # rather than erroring the opponent (the reference's
+                        # --timeout kills the remote call; locally a partial
+                        # decode is still a usable critique).
+                        break
+                    logits = self.model.decode_one(tok, cache)
+        finally:
+            if prev_device is not None:
+                torch.cuda.set_device(prev_device)
+
+        text = self.tokenizer.decode(out_ids)
+        return text, len(ids), len(out_ids), timer.as_dict()
+
+
+_ENGINES: dict[tuple, LocalEngine] = {}
+_ENGINES_LOCK = threading.Lock()
+
+
+def get_engine(spec: dict[str, Any], device: Optional[str] = None) -> LocalEngine:
+    """Process-wide engine cache: weights stay HBM-resident across rounds."""
+    key = (spec.get("name"), spec.get("arch"), spec.get("weights"), device or spec.get("gpu"))
+    with _ENGINES_LOCK:
+        eng = _ENGINES.get(key)
+        if eng is None:
+            eng = LocalEngine(spec, device=device)
+            _ENGINES[key] = eng
+        return eng
+
+
+def clear_engines() -> None:
+    with _ENGINES_LOCK:
+        _ENGINES.clear()

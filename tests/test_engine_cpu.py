@@ -1,0 +1,128 @@
+"""Engine tests on CPU: tokenizer, tiny model, generation, registry glue."""
+
+import torch
+
+from adversarial_spec_amd.engine.local import LocalEngine, clear_engines, get_engine
+from adversarial_spec_amd.engine.tokenizer import ByteTokenizer
+from adversarial_spec_amd.models import LlamaModel, get_config
+from adversarial_spec_amd.utils.synth import synthetic_spec
+
+
+class TestByteTokenizer:
+    def test_roundtrip_ascii(self):
+        tok = ByteTokenizer()
+        assert tok.decode(tok.encode("hello [SPEC] world")) == "hello [SPEC] world"
+
+    def test_roundtrip_utf8(self):
+        tok = ByteTokenizer()
+        s = "naïve — ünïcödé ✓"
+        assert tok.decode(tok.encode(s)) == s
+
+    def test_token_count_is_exact_bytes(self):
+        tok = ByteTokenizer()
+        assert len(tok.encode("abcd")) == 4
+
+    def test_specials_in_decode(self):
+        tok = ByteTokenizer()
+        text = tok.decode([tok.bos_id] + tok.encode("x") + [tok.eot_id])
+        assert "<|begin_of_text|>" in text and "<|eot_id|>" in text
+
+    def test_small_vocab_remap(self):
+        tok = ByteTokenizer(1024)
+        assert 256 <= tok.bos_id < 1024
+        assert tok.eot_id < 1024
+        assert tok.decode(tok.encode("ok")) == "ok"
+
+    def test_chat_render_structure(self):
+        tok = ByteTokenizer(1024)
+        ids = tok.render_chat("SYS", "USER MSG")
+        assert ids[0] == tok.bos_id
+        text = tok.decode(ids)
+        assert "SYS" in text and "USER MSG" in text
+        assert text.index("system") < text.index("SYS") < text.index("user")
+
+    def test_out_of_range_ids_dropped(self):
+        tok = ByteTokenizer(1024)
+        assert tok.decode([65, 999, 66]) == "AB"
+
+
+class TestTinyModel:
+    def setup_method(self):
+        self.cfg = get_config("tiny")
+        self.model = LlamaModel(self.cfg, device="cpu", seed=42).init_random()
+
+    def test_prefill_shape_and_finite(self):
+        cache = self.model.new_cache(64)
+        tokens = torch.arange(10)
+        logits = self.model.prefill(tokens, cache)
+        assert logits.shape == (self.cfg.vocab_size,)
+        assert torch.isfinite(logits).all()
+        assert cache.seq_len == 10
+
+    def test_decode_matches_prefill(self):
+        """Prefill of n+1 tokens == prefill of n then decode of 1 (same logits)."""
+        toks = torch.arange(1, 12)
+        c1 = self.model.new_cache(64)
+        full = self.model.prefill(toks, c1)
+
+        c2 = self.model.new_cache(64)
+        self.model.prefill(toks[:-1], c2)
+        step = self.model.decode_one(int(toks[-1]), c2)
+        assert torch.allclose(full, step, atol=1e-4)
+
+    def test_deterministic_init(self):
+        m2 = LlamaModel(self.cfg, device="cpu", seed=42).init_random()
+        assert torch.equal(self.model.embed, m2.embed)
+        m3 = LlamaModel(self.cfg, device="cpu", seed=43).init_random()
+        assert not torch.equal(self.model.embed, m3.embed)
+
+    def test_param_count_sane(self):
+        pc = self.cfg.param_count()
+        assert 1e6 < pc < 1e8
+
+
+class TestLocalEngine:
+    def test_generate_tiny(self):
+        eng = LocalEngine({"name": "t", "arch": "tiny"}, device="cpu")
+        text, in_tok, out_tok, timings = eng.generate(
+            "sys", "This is round 1 of adversarial spec development.\n\nspec",
+            max_tokens=16, temperature=0.7, timeout=30,
+        )
+        assert in_tok > 0
+        assert 0 <= out_tok <= 16
+        assert "prefill" in timings and "decode" in timings
+
+    def test_generate_deterministic_greedy(self):
+        eng = LocalEngine({"name": "t2", "arch": "tiny"}, device="cpu")
+        a = eng.generate("s", "u", max_tokens=8, temperature=0.0, timeout=30)
+        b = eng.generate("s", "u", max_tokens=8, temperature=0.0, timeout=30)
+        assert a[0] == b[0]
+
+    def test_prompt_fitting(self):
+        eng = LocalEngine({"name": "t3", "arch": "tiny"}, device="cpu")
+        huge = "x" * 10000  # tiny max_seq_len is 2048
+        text, in_tok, out_tok, _ = eng.generate(
+            "s", huge, max_tokens=4, temperature=0.0, timeout=30
+        )
+        assert in_tok <= eng.config.max_seq_len
+
+    def test_engine_cache(self):
+        clear_engines()
+        e1 = get_engine({"name": "c", "arch": "tiny", "weights": None, "gpu": None})
+        e2 = get_engine({"name": "c", "arch": "tiny", "weights": None, "gpu": None})
+        assert e1 is e2
+        clear_engines()
+
+
+class TestSyntheticSpec:
+    def test_exact_length(self):
+        s = synthetic_spec(4096, seed=1)
+        assert len(s) == 4096
+
+    def test_deterministic(self):
+        assert synthetic_spec(1000, seed=5) == synthetic_spec(1000, seed=5)
+        assert synthetic_spec(1000, seed=5) != synthetic_spec(1000, seed=6)
+
+    def test_looks_like_markdown(self):
+        s = synthetic_spec(2000, seed=0)
+        assert s.startswith("# ") and "## " in s
