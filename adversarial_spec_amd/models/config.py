@@ -61,9 +61,9 @@ PRESETS: dict[str, LlamaConfig] = {
         name="tiny", dim=256, n_layers=4, n_heads=8, n_kv_heads=2,
         ffn_dim=688, vocab_size=1024, max_seq_len=2048, rope_theta=10000.0,
     ),
-    # A mid-size config for single-GPU kernel shakeout (fits fast init).
+    # A mid-size config for single-GPU kernel shakeout (hd=128 MFMA path).
     "debug-1b": LlamaConfig(
-        name="debug-1b", dim=2048, n_layers=16, n_heads=32, n_kv_heads=8,
+        name="debug-1b", dim=2048, n_layers=16, n_heads=16, n_kv_heads=8,
         ffn_dim=8192, vocab_size=128256, max_seq_len=8192, rope_theta=500000.0,
     ),
 }

@@ -142,9 +142,13 @@ def sample(
     seed: Optional[int] = None,
 ) -> int:
     if _on_gpu(logits):
-        # Device-side fused temperature/softmax/top-p sampling kernel; the
-        # random draw comes from a host-provided seed counter for
-        # reproducibility across graph replays.
+        if top_p < 1.0:
+            # exact nucleus needs a sorted vocab: cold path (default decode
+            # samples at top_p == 1, reference models.py:626)
+            return torch_ref.sample(logits.float().cpu(), temperature, top_p, generator)
+        # Device-side fused temperature/softmax sampling kernel; the random
+        # draw comes from a host-provided seed counter so HIP-graph replays
+        # stay deterministic.
         if seed is None:
             seed = int(torch.randint(0, 2**31 - 1, (1,)).item())
         return int(_require_hip().sample(logits, temperature, top_p, seed))

@@ -1,0 +1,326 @@
+// Attention kernels for CDNA4 (gfx950).
+//
+//  - attn_decode_split / attn_decode_combine: single-token decode over the
+//    paged KV cache, flash-decoding style (split-KV online softmax with a
+//    combine pass). Memory-bound by the KV read: each KV row is read ONCE
+//    per kv-head and serves all `group` GQA query heads; vectorized 16 B
+//    per lane; splits spread the read over the whole chip.
+//
+//  - attn_prefill_simple: correctness-first causal prefill over the fresh
+//    contiguous K/V of the prompt (one 16-lane group per query row, online
+//    softmax in registers). It is the numerics anchor; the MFMA-tiled
+//    prefill (attn_prefill_mfma.hip) replaces it on the hot path.
+//
+// Layout contracts (ops/torch_ref.py):
+//   q        [tq, hq, hd] bf16
+//   k,v      [tk, kh, hd] bf16 (contiguous, prefill)
+//   kc,vc    [n_pages, page, kh, hd] bf16 (decode)
+//   page_table int32
+// State is fp32 throughout; hd in {32, 64, 128}; group = hq/kh <= 8.
+
+#include "common.h"
+
+#define MAXG 8
+
+// 16-lane-group dot reduce (LPP = lanes per position = hd/8)
+template <int LPP>
+DEVINL float group_reduce_sum(float v) {
+#pragma unroll
+  for (int off = LPP / 2; off > 0; off >>= 1) v += __shfl_xor(v, off, WAVE);
+  return v;
+}
+
+// ---------------------------------------------------------------------------
+// Decode split kernel.
+// grid.x = kh, grid.y = n_splits, block = 256 (4 waves).
+// Each wave covers SPLIT_LEN/4 = 64 consecutive positions of the split's
+// 256-position range; a wave runs 64/LPP positions concurrently (one per
+// 16-lane group at hd=128).
+//
+// Workspace (fp32): ws_m, ws_l: [kh, n_splits, group]
+//                   ws_acc:     [kh, n_splits, group, hd]
+// ---------------------------------------------------------------------------
+
+template <int LPP>
+__global__ void __launch_bounds__(256) attn_decode_split_kernel(
+    const ushort_t *__restrict__ q,   // [hq, hd]
+    const ushort_t *__restrict__ kc, const ushort_t *__restrict__ vc,
+    const int *__restrict__ page_table, int seq_len, float scale,
+    int kh, int group, int hd, int page, int split_len,
+    float *__restrict__ ws_m, float *__restrict__ ws_l,
+    float *__restrict__ ws_acc) {
+  const int g = blockIdx.x;          // kv head
+  const int split = blockIdx.y;
+  const int n_splits = gridDim.y;
+  const int lane = threadIdx.x & (WAVE - 1);
+  const int wid = threadIdx.x / WAVE;
+  const int sub = lane / LPP;        // position sub-group within wave
+  const int sl = lane % LPP;         // dim slice within position
+  const int pos_per_wave = split_len / 4;
+  const int subs = WAVE / LPP;       // concurrent positions per wave
+
+  const int start = split * split_len + wid * pos_per_wave;
+  const int limit = min(seq_len, min(start + pos_per_wave,
+                                     (split + 1) * split_len));
+
+  // q fragments: [group][8] for this lane's dim slice
+  float qf[MAXG][8];
+#pragma unroll
+  for (int gi = 0; gi < MAXG; ++gi) {
+    if (gi < group) {
+      const bf16x8 qv =
+          ((const bf16x8 *)(q + ((size_t)(g * group + gi)) * hd))[sl];
+      f32x8 qd = unpack8(qv);
+#pragma unroll
+      for (int j = 0; j < 8; ++j) qf[gi][j] = qd.v[j];
+    }
+  }
+
+  float m[MAXG], l[MAXG], acc[MAXG][8];
+#pragma unroll
+  for (int gi = 0; gi < MAXG; ++gi) {
+    m[gi] = -INFINITY;
+    l[gi] = 0.f;
+#pragma unroll
+    for (int j = 0; j < 8; ++j) acc[gi][j] = 0.f;
+  }
+
+  for (int p = start + sub; p < limit; p += subs) {
+    const int phys = page_table[p / page];
+    const size_t row = ((size_t)phys * page + (p % page)) * kh * hd;
+    const f32x8 kd = unpack8(((const bf16x8 *)(kc + row + (size_t)g * hd))[sl]);
+    const f32x8 vd = unpack8(((const bf16x8 *)(vc + row + (size_t)g * hd))[sl]);
+
+#pragma unroll
+    for (int gi = 0; gi < MAXG; ++gi) {
+      if (gi >= group) break;
+      float dot = 0.f;
+#pragma unroll
+      for (int j = 0; j < 8; ++j) dot += qf[gi][j] * kd.v[j];
+      const float s = group_reduce_sum<LPP>(dot) * scale;
+      const float m_new = fmaxf(m[gi], s);
+      const float alpha = __expf(m[gi] - m_new);
+      const float pex = __expf(s - m_new);
+      l[gi] = l[gi] * alpha + pex;
+#pragma unroll
+      for (int j = 0; j < 8; ++j)
+        acc[gi][j] = acc[gi][j] * alpha + pex * vd.v[j];
+      m[gi] = m_new;
+    }
+  }
+
+  // merge position sub-groups within the wave (lanes xor LPP, 2*LPP, ...)
+#pragma unroll
+  for (int off = LPP; off < WAVE; off <<= 1) {
+#pragma unroll
+    for (int gi = 0; gi < MAXG; ++gi) {
+      if (gi >= group) break;
+      const float mo = __shfl_xor(m[gi], off, WAVE);
+      const float lo = __shfl_xor(l[gi], off, WAVE);
+      float ao[8];
+#pragma unroll
+      for (int j = 0; j < 8; ++j) ao[j] = __shfl_xor(acc[gi][j], off, WAVE);
+      const float mn = fmaxf(m[gi], mo);
+      const float a = (m[gi] == -INFINITY && mo == -INFINITY) ? 0.f : __expf(m[gi] - mn);
+      const float b = (m[gi] == -INFINITY && mo == -INFINITY) ? 0.f : __expf(mo - mn);
+      l[gi] = l[gi] * a + lo * b;
+#pragma unroll
+      for (int j = 0; j < 8; ++j) acc[gi][j] = acc[gi][j] * a + ao[j] * b;
+      m[gi] = mn;
+    }
+  }
+
+  // merge the 4 waves via LDS, then write this split's partial state.
+  // LDS: per wave: m[group], l[group], acc[group][hd]
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  float *lm = (float *)smem;                       // [4][MAXG]
+  float *ll = lm + 4 * MAXG;                       // [4][MAXG]
+  float *lacc = ll + 4 * MAXG;                     // [4][MAXG][hd]
+
+  if (lane < LPP) {  // one lane per dim slice (sub==0 lanes)
+#pragma unroll
+    for (int gi = 0; gi < MAXG; ++gi) {
+      if (gi >= group) break;
+      if (lane == 0) {
+        lm[wid * MAXG + gi] = m[gi];
+        ll[wid * MAXG + gi] = l[gi];
+      }
+#pragma unroll
+      for (int j = 0; j < 8; ++j)
+        lacc[(wid * MAXG + gi) * hd + sl * 8 + j] = acc[gi][j];
+    }
+  }
+  __syncthreads();
+
+  // threads [0, group*hd) combine the 4 wave-states for one (gi, dim)
+  const int tid = threadIdx.x;
+  if (tid < group * hd) {
+    const int gi = tid / hd, dd = tid % hd;
+    float M = -INFINITY;
+#pragma unroll
+    for (int w = 0; w < 4; ++w) M = fmaxf(M, lm[w * MAXG + gi]);
+    float L = 0.f, A = 0.f;
+#pragma unroll
+    for (int w = 0; w < 4; ++w) {
+      const float mw = lm[w * MAXG + gi];
+      const float sc = (mw == -INFINITY) ? 0.f : __expf(mw - M);
+      L += ll[w * MAXG + gi] * sc;
+      A += lacc[(w * MAXG + gi) * hd + dd] * sc;
+    }
+    const size_t base = ((size_t)g * n_splits + split) * group + gi;
+    if (dd == 0) {
+      ws_m[base] = M;
+      ws_l[base] = L;
+    }
+    ws_acc[base * hd + dd] = A;
+  }
+}
+
+// ---------------------------------------------------------------------------
+// Decode combine kernel: grid.x = hq, block = hd threads.
+// out[h, :] = sum_splits(acc * exp(m - M)) / L_total
+// ---------------------------------------------------------------------------
+
+extern "C" __global__ void __launch_bounds__(128)
+attn_decode_combine_kernel(const float *__restrict__ ws_m,
+                           const float *__restrict__ ws_l,
+                           const float *__restrict__ ws_acc,
+                           ushort_t *__restrict__ out, int n_splits,
+                           int group, int hd) {
+  const int h = blockIdx.x;
+  const int g = h / group, gi = h % group;
+  const int dd = threadIdx.x;
+  if (dd >= hd) return;
+
+  float M = -INFINITY;
+  for (int s = 0; s < n_splits; ++s)
+    M = fmaxf(M, ws_m[((size_t)g * n_splits + s) * group + gi]);
+  float L = 0.f, A = 0.f;
+  for (int s = 0; s < n_splits; ++s) {
+    const size_t base = ((size_t)g * n_splits + s) * group + gi;
+    const float mw = ws_m[base];
+    const float sc = (mw == -INFINITY) ? 0.f : __expf(mw - M);
+    L += ws_l[base] * sc;
+    A += ws_acc[base * hd + dd] * sc;
+  }
+  out[(size_t)h * hd + dd] = f32_to_bf16(A / L);
+}
+
+// ---------------------------------------------------------------------------
+// Correctness-first causal prefill.
+// grid.x = hq, grid.y = ceil(tq / rows_per_block), block = 256 (4 waves).
+// Each 16-lane (LPP-lane) group owns ONE query row and walks keys 0..row.
+// ---------------------------------------------------------------------------
+
+template <int LPP>
+__global__ void __launch_bounds__(256) attn_prefill_simple_kernel(
+    const ushort_t *__restrict__ q, const ushort_t *__restrict__ k,
+    const ushort_t *__restrict__ v, ushort_t *__restrict__ out,
+    int tq, int tk, int kv_offset, float scale, int hq, int kh, int hd,
+    int causal) {
+  const int h = blockIdx.x;
+  const int g = h / (hq / kh);
+  const int lane = threadIdx.x & (WAVE - 1);
+  const int wid = threadIdx.x / WAVE;
+  const int sub = lane / LPP, sl = lane % LPP;
+  const int rows_per_block = 4 * (WAVE / LPP);
+  const int row = blockIdx.y * rows_per_block + wid * (WAVE / LPP) + sub;
+  if (row >= tq) return;
+
+  const int limit = causal ? min(tk, kv_offset + row + 1) : tk;
+
+  float qf[8];
+  {
+    const f32x8 qd =
+        unpack8(((const bf16x8 *)(q + ((size_t)row * hq + h) * hd))[sl]);
+#pragma unroll
+    for (int j = 0; j < 8; ++j) qf[j] = qd.v[j];
+  }
+
+  float m = -INFINITY, l = 0.f, acc[8];
+#pragma unroll
+  for (int j = 0; j < 8; ++j) acc[j] = 0.f;
+
+  for (int p = 0; p < limit; ++p) {
+    const size_t kr = ((size_t)p * kh + g) * hd;
+    const f32x8 kd = unpack8(((const bf16x8 *)(k + kr))[sl]);
+    float dot = 0.f;
+#pragma unroll
+    for (int j = 0; j < 8; ++j) dot += qf[j] * kd.v[j];
+    const float s = group_reduce_sum<LPP>(dot) * scale;
+    const f32x8 vd = unpack8(((const bf16x8 *)(v + kr))[sl]);
+    const float mn = fmaxf(m, s);
+    const float alpha = __expf(m - mn);
+    const float pex = __expf(s - mn);
+    l = l * alpha + pex;
+#pragma unroll
+    for (int j = 0; j < 8; ++j) acc[j] = acc[j] * alpha + pex * vd.v[j];
+    m = mn;
+  }
+
+  const float inv = (l > 0.f) ? 1.0f / l : 0.f;
+  bf16x8 o;
+#pragma unroll
+  for (int j = 0; j < 8; ++j) o.u[j] = f32_to_bf16(acc[j] * inv);
+  ((bf16x8 *)(out + ((size_t)row * hq + h) * hd))[sl] = o;
+}
+
+// ---------------------------------------------------------------------------
+// C-linkage launch wrappers (templated LPP dispatch)
+// ---------------------------------------------------------------------------
+
+extern "C" void launch_attn_decode_split(
+    const ushort_t *q, const ushort_t *kc, const ushort_t *vc,
+    const int *page_table, int seq_len, float scale, int kh, int group,
+    int hd, int page, int split_len, int n_splits, float *ws_m, float *ws_l,
+    float *ws_acc, hipStream_t stream) {
+  dim3 grid(kh, n_splits);
+  const int lds = (4 * MAXG * 2 + 4 * MAXG * hd) * sizeof(float);
+  switch (hd / 8) {
+    case 4:
+      attn_decode_split_kernel<4><<<grid, 256, lds, stream>>>(
+          q, kc, vc, page_table, seq_len, scale, kh, group, hd, page,
+          split_len, ws_m, ws_l, ws_acc);
+      break;
+    case 8:
+      attn_decode_split_kernel<8><<<grid, 256, lds, stream>>>(
+          q, kc, vc, page_table, seq_len, scale, kh, group, hd, page,
+          split_len, ws_m, ws_l, ws_acc);
+      break;
+    case 16:
+      attn_decode_split_kernel<16><<<grid, 256, lds, stream>>>(
+          q, kc, vc, page_table, seq_len, scale, kh, group, hd, page,
+          split_len, ws_m, ws_l, ws_acc);
+      break;
+  }
+}
+
+extern "C" void launch_attn_decode_combine(
+    const float *ws_m, const float *ws_l, const float *ws_acc, ushort_t *out,
+    int hq, int n_splits, int group, int hd, hipStream_t stream) {
+  attn_decode_combine_kernel<<<dim3(hq), dim3(hd), 0, stream>>>(
+      ws_m, ws_l, ws_acc, out, n_splits, group, hd);
+}
+
+extern "C" void launch_attn_prefill_simple(
+    const ushort_t *q, const ushort_t *k, const ushort_t *v, ushort_t *out,
+    int tq, int tk, int kv_offset, float scale, int hq, int kh, int hd,
+    int causal, hipStream_t stream) {
+  const int lpp = hd / 8;
+  const int rows_per_block = 4 * (WAVE / lpp);
+  dim3 grid(hq, (tq + rows_per_block - 1) / rows_per_block);
+  switch (lpp) {
+    case 4:
+      attn_prefill_simple_kernel<4><<<grid, 256, 0, stream>>>(
+          q, k, v, out, tq, tk, kv_offset, scale, hq, kh, hd, causal);
+      break;
+    case 8:
+      attn_prefill_simple_kernel<8><<<grid, 256, 0, stream>>>(
+          q, k, v, out, tq, tk, kv_offset, scale, hq, kh, hd, causal);
+      break;
+    case 16:
+      attn_prefill_simple_kernel<16><<<grid, 256, 0, stream>>>(
+          q, k, v, out, tq, tk, kv_offset, scale, hq, kh, hd, causal);
+      break;
+  }
+}

@@ -1,0 +1,251 @@
+#include "hip/hip_runtime.h"
+// Flash-style causal prefill attention on CDNA4 MFMA (gfx950), bf16 I/O,
+// fp32 accumulation. This is the hot prefill kernel for hd=128 models
+// (Llama-3 family); hd∈{32,64} falls back to attention.hip's simple kernel.
+//
+// Structure (cdna_hip_programming.md §5/§B: LDS-staged K/V tiles, online
+// softmax, MFMA 16x16x32):
+//   grid  = (hq, ceil(tq/128));  block = 512 threads (8 waves)
+//   each wave owns a 16-row Q tile; the block shares K/V tiles of 32 keys
+//   staged in LDS (double-barrier loop, v1).
+//
+// MFMA fragment maps used (verified on hardware by tests/test_ops_gpu.py's
+// mfma probe — the C/D map is documented in the CDNA4 guide §3; A/B are the
+// standard CDNA maps with K doubled for gfx950):
+//   mfma_f32_16x16x32_bf16:
+//     A: lane l holds A[row = l%16][k = (l/16)*8 + j], j = 0..7
+//     B: lane l holds B[k = (l/16)*8 + j][col = l%16]
+//     C/D: lane l, reg r holds C[row = (l>>4)*4 + r][col = l&15]
+
+#include "common.h"
+
+typedef __attribute__((__vector_size__(8 * sizeof(short)))) short bf16x8v;
+typedef __attribute__((__vector_size__(4 * sizeof(float)))) float f32x4v;
+
+#define QROWS 16   // q rows per wave
+#define KVBLK 32   // keys per staged tile
+#define NWAVE 8    // waves per block
+#define HD 128     // head dim (this kernel is hd=128 only)
+
+extern "C" __global__ void __launch_bounds__(512, 1)
+attn_prefill_mfma_kernel(const ushort_t *__restrict__ q,
+                         const ushort_t *__restrict__ k,
+                         const ushort_t *__restrict__ v,
+                         ushort_t *__restrict__ out, int tq, int tk,
+                         int kv_offset, float scale, int hq, int kh,
+                         int causal) {
+  const int h = blockIdx.x;
+  const int g = h / (hq / kh);   // kv head
+  const int qblock = blockIdx.y; // 128 q rows per block
+  const int tid = threadIdx.x;
+  const int lane = tid & (WAVE - 1);
+  const int wid = tid / WAVE;
+
+  const int qb0 = qblock * (NWAVE * QROWS);
+  const int q0 = qb0 + wid * QROWS; // this wave's first q row
+
+  // LDS: K tile + V tile (bf16 [KVBLK][HD]), plus per-wave P tiles
+  // ([QROWS][KVBLK] bf16). Single __shared__ object (compiler trap #4a).
+  __shared__ __attribute__((aligned(16))) ushort_t lds[
+      2 * KVBLK * HD + NWAVE * QROWS * KVBLK];
+  ushort_t *ldsK = lds;
+  ushort_t *ldsV = lds + KVBLK * HD;
+  ushort_t *ldsP = lds + 2 * KVBLK * HD + wid * QROWS * KVBLK;
+
+  const int lrow = lane & 15;       // 0..15
+  const int lhi = lane >> 4;        // 0..3
+
+  // ---- load Q fragments (persistent): 4 k-chunks of 32 dims ----
+  bf16x8v qfrag[4];
+  {
+    const int qrow = q0 + lrow;
+    const bool ok = qrow < tq;
+    const ushort_t *qr = q + ((size_t)(ok ? qrow : 0) * hq + h) * HD;
+#pragma unroll
+    for (int c = 0; c < 4; ++c) {
+      bf16x8 t;
+      if (ok) {
+        t = ((const bf16x8 *)qr)[c * 4 + lhi];  // dims c*32 + lhi*8 .. +8
+      } else {
+#pragma unroll
+        for (int j = 0; j < 8; ++j) t.u[j] = 0;
+      }
+#pragma unroll
+      for (int j = 0; j < 8; ++j) qfrag[c][j] = (short)t.u[j];
+    }
+  }
+
+  // ---- state ----
+  float m[4], lsum[4];
+  f32x4v o[8];  // O accumulators: 8 dim-chunks of 16
+#pragma unroll
+  for (int r = 0; r < 4; ++r) { m[r] = -INFINITY; lsum[r] = 0.f; }
+#pragma unroll
+  for (int d = 0; d < 8; ++d) o[d] = (f32x4v){0.f, 0.f, 0.f, 0.f};
+
+  // causal upper bound on keys for this block (kv-position space)
+  int kmax = tk;
+  if (causal) kmax = min(tk, kv_offset + qb0 + NWAVE * QROWS);
+
+  for (int kt = 0; kt < kmax; kt += KVBLK) {
+    // ---- stage K/V tile (coalesced 16B per thread) ----
+    {
+      // KVBLK*HD bf16 = 8 KB = 512 * 16 B
+      const int flat = tid;  // one 16B piece each
+      const int krow = flat / (HD / 8);
+      const int kcol8 = flat % (HD / 8);
+      const int key = kt + krow;
+      bf16x8 kk, vv;
+      if (key < tk) {
+        kk = ((const bf16x8 *)(k + ((size_t)key * kh + g) * HD))[kcol8];
+        vv = ((const bf16x8 *)(v + ((size_t)key * kh + g) * HD))[kcol8];
+      } else {
+#pragma unroll
+        for (int j = 0; j < 8; ++j) { kk.u[j] = 0; vv.u[j] = 0; }
+      }
+      __syncthreads();  // previous iteration's reads done
+      ((bf16x8 *)ldsK)[flat] = kk;
+      ((bf16x8 *)ldsV)[flat] = vv;
+      __syncthreads();
+    }
+
+    // ---- S = Q K^T for two 16-key subtiles ----
+    f32x4v s[2];
+#pragma unroll
+    for (int n = 0; n < 2; ++n) {
+      s[n] = (f32x4v){0.f, 0.f, 0.f, 0.f};
+      // B frag: B[kd][col] = K[n*16+col][c*32+kd8]
+      const int keyr = n * 16 + lrow;
+#pragma unroll
+      for (int c = 0; c < 4; ++c) {
+        bf16x8v bfr;
+        const bf16x8 t = ((const bf16x8 *)(ldsK + keyr * HD))[c * 4 + lhi];
+#pragma unroll
+        for (int j = 0; j < 8; ++j) bfr[j] = (short)t.u[j];
+        s[n] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(qfrag[c], bfr, s[n], 0, 0, 0);
+      }
+    }
+
+    // ---- online softmax update (4 q rows per lane: row=(lhi)*4+r) ----
+    float rmax[4];
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+      // scale + causal mask
+      const int qrow_abs = kv_offset + q0 + lhi * 4 + r;
+#pragma unroll
+      for (int n = 0; n < 2; ++n) {
+        const int key = kt + n * 16 + lrow;
+        float sv = s[n][r] * scale;
+        if (key >= tk || (causal && key > qrow_abs)) sv = -INFINITY;
+        s[n][r] = sv;
+      }
+      float mx = fmaxf(s[0][r], s[1][r]);
+#pragma unroll
+      for (int off = 8; off > 0; off >>= 1)
+        mx = fmaxf(mx, __shfl_xor(mx, off, WAVE));
+      rmax[r] = mx;
+    }
+
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+      const float mn = fmaxf(m[r], rmax[r]);
+      const float alpha =
+          (m[r] == -INFINITY && mn == -INFINITY) ? 0.f : __expf(m[r] - mn);
+      float psum = 0.f;
+#pragma unroll
+      for (int n = 0; n < 2; ++n) {
+        const float p = (s[n][r] == -INFINITY) ? 0.f : __expf(s[n][r] - mn);
+        s[n][r] = p;
+        psum += p;
+      }
+#pragma unroll
+      for (int off = 8; off > 0; off >>= 1)
+        psum += __shfl_xor(psum, off, WAVE);
+      lsum[r] = lsum[r] * alpha + psum;
+      m[r] = mn;
+      // rescale O rows
+#pragma unroll
+      for (int d = 0; d < 8; ++d) o[d][r] *= alpha;
+    }
+
+    // ---- P -> LDS (C layout -> A layout bounce) ----
+#pragma unroll
+    for (int n = 0; n < 2; ++n) {
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        ldsP[(lhi * 4 + r) * KVBLK + n * 16 + lrow] = f32_to_bf16(s[n][r]);
+      }
+    }
+    // wave-local LDS dependency; compiler inserts lgkmcnt before reads.
+
+    // ---- O += P V ----
+    bf16x8v pfrag;
+    {
+      const bf16x8 t = *(const bf16x8 *)(ldsP + lrow * KVBLK + lhi * 8);
+#pragma unroll
+      for (int j = 0; j < 8; ++j) pfrag[j] = (short)t.u[j];
+    }
+#pragma unroll
+    for (int d = 0; d < 8; ++d) {
+      // B frag: B[kd][col] = V[kd8][d*16+col] (strided LDS read, v1)
+      bf16x8v vfr;
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        vfr[j] = (short)ldsV[(lhi * 8 + j) * HD + d * 16 + lrow];
+      }
+      o[d] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(pfrag, vfr, o[d], 0, 0, 0);
+    }
+  }
+
+  // ---- epilogue: O / l ----
+#pragma unroll
+  for (int r = 0; r < 4; ++r) {
+    const int qrow = q0 + lhi * 4 + r;
+    if (qrow >= tq) continue;
+    const float inv = (lsum[r] > 0.f) ? 1.0f / lsum[r] : 0.f;
+    ushort_t *orow = out + ((size_t)qrow * hq + h) * HD;
+#pragma unroll
+    for (int d = 0; d < 8; ++d) {
+      orow[d * 16 + lrow] = f32_to_bf16(o[d][r] * inv);
+    }
+  }
+}
+
+extern "C" void launch_attn_prefill_mfma(const ushort_t *q, const ushort_t *k,
+                                         const ushort_t *v, ushort_t *out,
+                                         int tq, int tk, int kv_offset,
+                                         float scale, int hq, int kh,
+                                         int hd, hipStream_t stream, int *ok) {
+  if (hd != HD) {
+    *ok = 0;
+    return;
+  }
+  dim3 grid(hq, (tq + NWAVE * QROWS - 1) / (NWAVE * QROWS));
+ hipLaunchKernelGGL(( attn_prefill_mfma_kernel), dim3(grid), dim3(NWAVE * WAVE), 0, stream, 
+      q, k, v, out, tq, tk, kv_offset, scale, hq, kh, 1);
+  *ok = 1;
+}
+
+// ---------------------------------------------------------------------------
+// MFMA layout probe: D = A @ B for A[16,32], B[32,16] bf16 — used by the GPU
+// test suite to pin the fragment maps against torch.matmul before trusting
+// the attention kernel.
+// ---------------------------------------------------------------------------
+
+extern "C" __global__ void __launch_bounds__(64)
+mfma_probe_16x16x32(const ushort_t *__restrict__ a,  // [16][32]
+                    const ushort_t *__restrict__ b,  // [32][16]
+                    float *__restrict__ d) {         // [16][16]
+  const int lane = threadIdx.x & (WAVE - 1);
+  const int lrow = lane & 15, lhi = lane >> 4;
+  bf16x8v af, bf;
+#pragma unroll
+  for (int j = 0; j < 8; ++j) {
+    af[j] = (short)a[lrow * 32 + lhi * 8 + j];
+    bf[j] = (short)b[(lhi * 8 + j) * 16 + lrow];
+  }
+  f32x4v c = (f32x4v){0.f, 0.f, 0.f, 0.f};
+  c = __builtin_amdgcn_mfma_f32_16x16x32_bf16(af, bf, c, 0, 0, 0);
+#pragma unroll
+  for (int r = 0; r < 4; ++r) d[(lhi * 4 + r) * 16 + lrow] = c[r];
+}

@@ -1,0 +1,260 @@
+// Torch extension bindings for the gfx950 kernels.
+//
+// Contracts mirror ops/torch_ref.py exactly (shapes/dtypes documented
+// there). All activations are bf16, RoPE tables f32, workspace fp32.
+
+#include <torch/extension.h>
+#include <c10/hip/HIPStream.h>
+#include <hip/hip_runtime.h>
+
+using ushort_t = unsigned short;
+
+// launch wrappers defined in the .hip translation units
+extern "C" {
+void launch_attn_decode_split(const ushort_t*, const ushort_t*, const ushort_t*,
+                              const int*, int, float, int, int, int, int, int,
+                              int, float*, float*, float*, hipStream_t);
+void launch_attn_decode_combine(const float*, const float*, const float*,
+                                ushort_t*, int, int, int, int, hipStream_t);
+void launch_attn_prefill_simple(const ushort_t*, const ushort_t*, const ushort_t*,
+                                ushort_t*, int, int, int, float, int, int, int,
+                                int, hipStream_t);
+void launch_attn_prefill_mfma(const ushort_t*, const ushort_t*, const ushort_t*,
+                              ushort_t*, int, int, int, float, int, int, int,
+                              hipStream_t, int* ok);
+__global__ void mfma_probe_16x16x32(const ushort_t*, const ushort_t*, float*);
+__global__ void rmsnorm_kernel(const ushort_t*, const ushort_t*, ushort_t*, int, float);
+__global__ void add_rmsnorm_kernel(const ushort_t*, const ushort_t*, const ushort_t*,
+                                   ushort_t*, ushort_t*, int, float);
+__global__ void rope_kernel(ushort_t*, ushort_t*, const float*, const float*,
+                            int, int, int, int, int);
+__global__ void swiglu_kernel(const ushort_t*, const ushort_t*, ushort_t*, int, int, int);
+__global__ void kv_write_kernel(const ushort_t*, const ushort_t*, ushort_t*, ushort_t*,
+                                const int*, int, int, int, int, int);
+__global__ void sample_kernel(const ushort_t*, int, float, uint32_t, int*);
+}
+
+static hipStream_t cur_stream() {
+  return c10::hip::getCurrentHIPStream().stream();
+}
+
+#define CHECK_BF16_CUDA(t)                                                 \
+  TORCH_CHECK((t).is_cuda(), #t " must be on GPU");                        \
+  TORCH_CHECK((t).scalar_type() == at::kBFloat16, #t " must be bf16");
+
+static const ushort_t* uptr(const torch::Tensor& t) {
+  return reinterpret_cast<const ushort_t*>(t.data_ptr());
+}
+static ushort_t* uptr_mut(torch::Tensor& t) {
+  return reinterpret_cast<ushort_t*>(t.data_ptr());
+}
+
+// --------------------------------------------------------------------------
+
+torch::Tensor rmsnorm(torch::Tensor x, torch::Tensor w, double eps) {
+  CHECK_BF16_CUDA(x);
+  CHECK_BF16_CUDA(w);
+  auto xc = x.contiguous();
+  auto wc = w.contiguous();
+  const int d = xc.size(-1);
+  const int t = xc.numel() / d;
+  TORCH_CHECK(d % 8 == 0, "rmsnorm: d must be a multiple of 8");
+  auto y = torch::empty_like(xc);
+  rmsnorm_kernel<<<t, 256, 0, cur_stream()>>>(uptr(xc), uptr(wc), uptr_mut(y),
+                                              d, (float)eps);
+  return y;
+}
+
+std::tuple<torch::Tensor, torch::Tensor> add_rmsnorm(torch::Tensor resid,
+                                                     torch::Tensor delta,
+                                                     torch::Tensor w,
+                                                     double eps) {
+  CHECK_BF16_CUDA(resid);
+  CHECK_BF16_CUDA(delta);
+  CHECK_BF16_CUDA(w);
+  auto rc = resid.contiguous();
+  auto dc = delta.contiguous();
+  auto wc = w.contiguous();
+  const int d = rc.size(-1);
+  const int t = rc.numel() / d;
+  TORCH_CHECK(d % 8 == 0, "add_rmsnorm: d must be a multiple of 8");
+  auto r_out = torch::empty_like(rc);
+  auto y = torch::empty_like(rc);
+  add_rmsnorm_kernel<<<t, 256, 0, cur_stream()>>>(
+      uptr(rc), uptr(dc), uptr(wc), uptr_mut(r_out), uptr_mut(y), d, (float)eps);
+  return {r_out, y};
+}
+
+void rope_inplace(torch::Tensor q, torch::Tensor k, torch::Tensor cost,
+                  torch::Tensor sint, int64_t pos0) {
+  CHECK_BF16_CUDA(q);
+  CHECK_BF16_CUDA(k);
+  TORCH_CHECK(q.is_contiguous() && k.is_contiguous(), "rope: q/k contiguous");
+  TORCH_CHECK(cost.scalar_type() == at::kFloat, "rope: cos table f32");
+  const int t = q.size(0), hq = q.size(1), hd = q.size(2);
+  const int hk = k.size(1);
+  TORCH_CHECK(hd % 2 == 0);
+  const int waves = t * (hq + hk);
+  const int blocks = (waves * 64 + 255) / 256;
+  rope_kernel<<<blocks, 256, 0, cur_stream()>>>(
+      uptr_mut(q), uptr_mut(k), cost.data_ptr<float>(), sint.data_ptr<float>(),
+      t, hq, hk, hd, (int)pos0);
+}
+
+torch::Tensor swiglu(torch::Tensor gate, torch::Tensor up) {
+  CHECK_BF16_CUDA(gate);
+  CHECK_BF16_CUDA(up);
+  TORCH_CHECK(gate.dim() == 2 && up.dim() == 2);
+  TORCH_CHECK(gate.stride(1) == 1 && up.stride(1) == 1,
+              "swiglu: last dim must be contiguous");
+  TORCH_CHECK(gate.stride(0) == up.stride(0),
+              "swiglu: gate/up must share a row stride (gate_up halves)");
+  const int t = gate.size(0), f = gate.size(1);
+  TORCH_CHECK(f % 8 == 0);
+  auto out = torch::empty({t, f}, gate.options());
+  const long nvec = (long)t * f / 8;
+  const int blocks = (int)std::min<long>((nvec + 255) / 256, 8192);
+  swiglu_kernel<<<blocks, 256, 0, cur_stream()>>>(
+      uptr(gate), uptr(up), uptr_mut(out), t, f, (int)gate.stride(0));
+  return out;
+}
+
+void kv_write(torch::Tensor kc, torch::Tensor vc, torch::Tensor page_table,
+              int64_t pos0, torch::Tensor k, torch::Tensor v) {
+  CHECK_BF16_CUDA(kc);
+  CHECK_BF16_CUDA(k);
+  TORCH_CHECK(page_table.scalar_type() == at::kInt);
+  auto kcc = kc;  // [np, page, kh, hd] must already be contiguous
+  TORCH_CHECK(kc.is_contiguous() && vc.is_contiguous(), "cache contiguous");
+  auto kq = k.contiguous();
+  auto vq = v.contiguous();
+  const int t = kq.size(0), kh = kq.size(1), hd = kq.size(2);
+  const int page = kc.size(1);
+  TORCH_CHECK((kh * hd) % 8 == 0);
+  kv_write_kernel<<<t, 256, 0, cur_stream()>>>(
+      uptr(kq), uptr(vq), uptr_mut(kcc), uptr_mut(vc),
+      page_table.data_ptr<int>(), (int)pos0, t, kh, hd, page);
+}
+
+torch::Tensor attn_prefill(torch::Tensor q, torch::Tensor k, torch::Tensor v,
+                           double scale, bool causal, int64_t kv_offset) {
+  CHECK_BF16_CUDA(q);
+  CHECK_BF16_CUDA(k);
+  CHECK_BF16_CUDA(v);
+  auto qc = q.contiguous();
+  auto kc = k.contiguous();
+  auto vc = v.contiguous();
+  const int tq = qc.size(0), hq = qc.size(1), hd = qc.size(2);
+  const int tk = kc.size(0), kh = kc.size(1);
+  TORCH_CHECK(hq % kh == 0, "GQA: hq % kh == 0");
+  TORCH_CHECK(hd == 32 || hd == 64 || hd == 128, "hd in {32,64,128}");
+  auto out = torch::empty_like(qc);
+  // hd=128 causal takes the MFMA-tiled kernel; other head dims and the
+  // non-causal path use the simple online-softmax kernel.
+  int used_mfma = 0;
+  if (hd == 128 && causal) {
+    launch_attn_prefill_mfma(uptr(qc), uptr(kc), uptr(vc), uptr_mut(out), tq,
+                             tk, (int)kv_offset, (float)scale, hq, kh, hd,
+                             cur_stream(), &used_mfma);
+  }
+  if (!used_mfma) {
+    launch_attn_prefill_simple(uptr(qc), uptr(kc), uptr(vc), uptr_mut(out), tq,
+                               tk, (int)kv_offset, (float)scale, hq, kh, hd,
+                               causal ? 1 : 0, cur_stream());
+  }
+  return out;
+}
+
+// Force the non-MFMA prefill (A/B + numerics anchor for tests).
+torch::Tensor attn_prefill_simple(torch::Tensor q, torch::Tensor k,
+                                  torch::Tensor v, double scale, bool causal,
+                                  int64_t kv_offset) {
+  CHECK_BF16_CUDA(q);
+  auto qc = q.contiguous();
+  auto kc = k.contiguous();
+  auto vc = v.contiguous();
+  const int tq = qc.size(0), hq = qc.size(1), hd = qc.size(2);
+  const int tk = kc.size(0), kh = kc.size(1);
+  auto out = torch::empty_like(qc);
+  launch_attn_prefill_simple(uptr(qc), uptr(kc), uptr(vc), uptr_mut(out), tq,
+                             tk, (int)kv_offset, (float)scale, hq, kh, hd,
+                             causal ? 1 : 0, cur_stream());
+  return out;
+}
+
+// MFMA fragment-map probe: D[16,16] = A[16,32] @ B[32,16].
+torch::Tensor mfma_probe16(torch::Tensor a, torch::Tensor b) {
+  CHECK_BF16_CUDA(a);
+  CHECK_BF16_CUDA(b);
+  auto ac = a.contiguous();
+  auto bc = b.contiguous();
+  TORCH_CHECK(ac.size(0) == 16 && ac.size(1) == 32);
+  TORCH_CHECK(bc.size(0) == 32 && bc.size(1) == 16);
+  auto d = torch::empty({16, 16}, torch::TensorOptions()
+                                      .dtype(at::kFloat)
+                                      .device(a.device()));
+  mfma_probe_16x16x32<<<1, 64, 0, cur_stream()>>>(uptr(ac), uptr(bc),
+                                                  d.data_ptr<float>());
+  return d;
+}
+
+torch::Tensor attn_decode_paged(torch::Tensor q, torch::Tensor kc,
+                                torch::Tensor vc, torch::Tensor page_table,
+                                int64_t seq_len, double scale) {
+  CHECK_BF16_CUDA(q);
+  CHECK_BF16_CUDA(kc);
+  auto qc = q.contiguous();
+  TORCH_CHECK(kc.is_contiguous() && vc.is_contiguous());
+  TORCH_CHECK(page_table.scalar_type() == at::kInt);
+  const int hq = qc.size(0), hd = qc.size(1);
+  const int page = kc.size(1), kh = kc.size(2);
+  const int group = hq / kh;
+  TORCH_CHECK(group <= 8, "GQA group <= 8");
+  TORCH_CHECK(hd == 32 || hd == 64 || hd == 128);
+
+  const int split_len = 256;
+  const int n_splits = (int)((seq_len + split_len - 1) / split_len);
+  auto wopt = torch::TensorOptions().dtype(at::kFloat).device(q.device());
+  auto ws_m = torch::empty({(long)kh * n_splits * group}, wopt);
+  auto ws_l = torch::empty({(long)kh * n_splits * group}, wopt);
+  auto ws_acc = torch::empty({(long)kh * n_splits * group * hd}, wopt);
+  auto out = torch::empty({hq, hd}, qc.options());
+
+  launch_attn_decode_split(uptr(qc), uptr(kc), uptr(vc),
+                           page_table.data_ptr<int>(), (int)seq_len,
+                           (float)scale, kh, group, hd, page, split_len,
+                           n_splits, ws_m.data_ptr<float>(),
+                           ws_l.data_ptr<float>(), ws_acc.data_ptr<float>(),
+                           cur_stream());
+  launch_attn_decode_combine(ws_m.data_ptr<float>(), ws_l.data_ptr<float>(),
+                             ws_acc.data_ptr<float>(), uptr_mut(out), hq,
+                             n_splits, group, hd, cur_stream());
+  return out;
+}
+
+int64_t sample(torch::Tensor logits, double temp, double top_p, int64_t seed) {
+  CHECK_BF16_CUDA(logits);
+  TORCH_CHECK(top_p >= 1.0, "kernel sample handles top_p == 1 (nucleus is a cold path)");
+  auto lc = logits.contiguous();
+  const int vocab = lc.numel();
+  auto out = torch::empty({1}, torch::TensorOptions()
+                                   .dtype(at::kInt)
+                                   .device(logits.device()));
+  sample_kernel<<<1, 256, 0, cur_stream()>>>(uptr(lc), vocab, (float)temp,
+                                             (uint32_t)seed,
+                                             out.data_ptr<int>());
+  return out.cpu().item<int>();
+}
+
+PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+  m.def("rmsnorm", &rmsnorm, "fused RMSNorm (bf16, gfx950)");
+  m.def("add_rmsnorm", &add_rmsnorm, "fused residual add + RMSNorm");
+  m.def("rope_inplace", &rope_inplace, "RoPE (interleaved pairs, table-driven)");
+  m.def("swiglu", &swiglu, "fused silu(gate)*up");
+  m.def("kv_write", &kv_write, "paged KV scatter");
+  m.def("attn_prefill", &attn_prefill, "causal prefill attention");
+  m.def("attn_prefill_simple", &attn_prefill_simple, "non-MFMA prefill (anchor)");
+  m.def("mfma_probe16", &mfma_probe16, "MFMA 16x16x32 fragment-map probe");
+  m.def("attn_decode_paged", &attn_decode_paged, "paged decode attention");
+  m.def("sample", &sample, "fused temperature softmax sample");
+}

@@ -1,0 +1,184 @@
+#!/usr/bin/env python3
+"""Flagship benchmark: debate-round wall-clock + critiques/sec on MI355X.
+
+Measures BASELINE.json's headline metric — one adversarial debate round of
+N_opp opponents critiquing an 8k-token synthetic spec — on random-init
+Llama-3-8B opponents (bf16), exactly the configuration BASELINE.json names
+(config 3 at world_size 1: "3 opponents, 8k-token spec").
+
+Scaling is WEAK: every GPU hosts `--opponents-per-gpu` (default 3)
+co-resident opponents in its 288 GB of HBM3E; adding GPUs adds opponents.
+A round = every opponent prefills the spec prompt and decodes its critique,
+then one fused RCCL all-gather over xGMI collects all critique token
+buffers + agreed flags for the consensus check (parallel/consensus.py).
+
+Launch (driver contract):
+  python bench.py --gpus 1 --steps K --warmup W
+  python -m torch.distributed.run --nnodes=1 --nproc-per-node N \
+      --master-addr 127.0.0.1 bench.py --gpus N --steps K --warmup W
+
+Rank 0 prints ONE JSON line with the whole-job aggregate.
+"""
+
+from __future__ import annotations
+
+import argparse
+import json
+import os
+import sys
+import time
+
+import torch
+
+sys.path.insert(0, os.path.dirname(os.path.abspath(__file__)))
+
+from adversarial_spec_amd.engine.local import LocalEngine  # noqa: E402
+from adversarial_spec_amd.engine.scheduler import build_user_message  # noqa: E402
+from adversarial_spec_amd.parallel.consensus import (  # noqa: E402
+    HDR,
+    pack_result,
+    unpack_results,
+)
+from adversarial_spec_amd.prompts import get_system_prompt  # noqa: E402
+from adversarial_spec_amd.protocol import detect_agreement  # noqa: E402
+from adversarial_spec_amd.utils.synth import synthetic_spec  # noqa: E402
+
+
+def parse_args():
+    p = argparse.ArgumentParser()
+    p.add_argument("--gpus", type=int, default=1)
+    p.add_argument("--steps", type=int, default=3, help="timed debate rounds")
+    p.add_argument("--warmup", type=int, default=1, help="untimed rounds")
+    p.add_argument("--opponents-per-gpu", type=int, default=3)
+    p.add_argument("--spec-tokens", type=int, default=8192)
+    p.add_argument("--decode-tokens", type=int, default=128,
+                   help="critique length decoded per opponent per round")
+    p.add_argument("--model", default="llama-3-8b")
+    p.add_argument("--temperature", type=float, default=0.7)
+    return p.parse_args()
+
+
+def main() -> int:
+    args = parse_args()
+    world = int(os.environ.get("WORLD_SIZE", "1"))
+    rank = int(os.environ.get("RANK", "0"))
+    local_rank = int(os.environ.get("LOCAL_RANK", "0"))
+
+    use_gpu = torch.cuda.is_available()
+    if not use_gpu:
+        print("bench.py requires an MI355X GPU", file=sys.stderr)
+        return 1
+    torch.cuda.set_device(local_rank)
+    device = f"cuda:{local_rank}"
+
+    dist = None
+    if world > 1:
+        import torch.distributed as dist_mod
+
+        dist = dist_mod
+        dist.init_process_group("nccl")
+
+    n_opp = args.opponents_per_gpu
+    engines = []
+    for i in range(n_opp):
+        # distinct seeds per (rank, opponent): heterogeneous random-init pool
+        engines.append(
+            LocalEngine(
+                {"name": f"{args.model}-r{rank}o{i}", "arch": args.model},
+                device=device,
+            )
+        )
+
+    spec = synthetic_spec(args.spec_tokens, seed=17)
+    system_prompt = get_system_prompt("tech")
+    user_message = build_user_message(spec, 1, "tech")
+    max_gather_tokens = args.decode_tokens
+
+    def one_round():
+        """One debate round for this rank's opponents + consensus gather."""
+        packed = []
+        for eng in engines:
+            text, _in, _out, _tm = eng.generate(
+                system_prompt,
+                user_message,
+                max_tokens=args.decode_tokens,
+                temperature=args.temperature,
+                timeout=600.0,
+            )
+            ids = eng.tokenizer.encode(text)[: max_gather_tokens]
+            packed.append(
+                pack_result(ids, detect_agreement(text), False,
+                            max_gather_tokens, torch.device(device))
+            )
+        mine = torch.cat(packed)  # [n_opp * (HDR+max)]
+        if dist is not None:
+            out = torch.zeros(world * mine.numel(), dtype=torch.int32,
+                              device=device)
+            dist.all_gather_into_tensor(out, mine)
+        else:
+            out = mine
+        rows = out.view(world * n_opp, HDR + max_gather_tokens)
+        results = unpack_results(rows)
+        ok = [r for r in results if not r.error]
+        return bool(ok) and all(r.agreed for r in ok)
+
+    # warmup
+    for _ in range(args.warmup):
+        one_round()
+
+    if dist is not None:
+        dist.barrier()
+    torch.cuda.synchronize()
+    t0 = time.perf_counter()
+    for _ in range(args.steps):
+        one_round()
+    if dist is not None:
+        dist.barrier()
+    torch.cuda.synchronize()
+    elapsed = time.perf_counter() - t0
+
+    # max over ranks
+    if dist is not None:
+        e = torch.tensor([elapsed], device=device)
+        dist.all_reduce(e, op=dist.ReduceOp.MAX)
+        elapsed = float(e.item())
+
+    total_critiques = world * n_opp * args.steps
+    value = total_critiques / elapsed
+    ms_per_step = elapsed / args.steps * 1000.0
+
+    if rank == 0:
+        print(json.dumps({
+            "metric": "critiques/sec (debate round: 8k-token spec prefill + "
+                      f"{args.decode_tokens}-token critique decode + RCCL consensus)",
+            "value": value,
+            "unit": "critiques/s",
+            "n_gpus": world,
+            "steps": args.steps,
+            "warmup": args.warmup,
+            "ms_per_step": ms_per_step,
+            "higher_is_better": True,
+            "scaling": "weak",
+            "vs_baseline": None,
+            "dtype": "bf16",
+            "data": "synthetic",
+            "config": {
+                "model": args.model,
+                "opponents": world * n_opp,
+                "opponents_per_gpu": n_opp,
+                "spec_tokens": args.spec_tokens,
+                "decode_tokens": args.decode_tokens,
+                "global_batch": world * n_opp,
+                "seq_len": args.spec_tokens,
+                "parallelism": f"opponent-parallel dp{world}",
+                "temperature": args.temperature,
+            },
+        }))
+
+    if dist is not None:
+        dist.destroy_process_group()
+    return 0
+
+
+if __name__ == "__main__":
+    sys.exit(main())

@@ -1,0 +1,229 @@
+"""GPU numerics tests: every HIP kernel vs the fp32 torch reference.
+
+Run on MI355X via:  gpurun -- 'python -m pytest tests -m gpu -x -q'
+Inputs are bf16 (the kernels' wire dtype); the reference computes in fp32
+on the same bf16-rounded values, so tolerances cover bf16 output rounding
+plus fp32-accumulation-order differences only.
+"""
+
+import math
+
+import pytest
+import torch
+
+pytestmark = pytest.mark.gpu
+
+if torch.cuda.is_available():
+    from adversarial_spec_amd import ops
+    from adversarial_spec_amd.ops import torch_ref
+else:  # collected but deselected on CPU
+    ops = torch_ref = None
+
+DEV = "cuda:0"
+
+
+def _bf(x):
+    return x.to(torch.bfloat16)
+
+
+def _assert_close(got, want, atol, rtol=2e-2, name=""):
+    g = got.float().cpu()
+    w = want.float().cpu()
+    err = (g - w).abs()
+    bound = atol + rtol * w.abs()
+    frac_bad = (err > bound).float().mean().item()
+    assert frac_bad == 0.0, (
+        f"{name}: max err {err.max():.4g} vs bound; {frac_bad*100:.2f}% out of tol"
+    )
+
+
+@pytest.fixture(autouse=True)
+def _seed():
+    torch.manual_seed(1234)
+
+
+class TestMFMAProbe:
+    def test_fragment_map(self):
+        """Pins the 16x16x32 A/B/C fragment maps vs torch.matmul."""
+        from adversarial_spec_amd.ops import _advspec_hip
+
+        a = _bf(torch.randn(16, 32)).to(DEV)
+        b = _bf(torch.randn(32, 16)).to(DEV)
+        d = _advspec_hip.mfma_probe16(a, b)
+        want = a.float() @ b.float()
+        _assert_close(d, want, atol=1e-2, name="mfma_probe")
+
+    def test_fragment_map_asymmetric(self):
+        """Asymmetric structured B catches transposed outputs (guide G9)."""
+        from adversarial_spec_amd.ops import _advspec_hip
+
+        a = torch.zeros(16, 32)
+        a[3, 5] = 1.0
+        b = torch.zeros(32, 16)
+        b[5, 11] = 2.0
+        d = _advspec_hip.mfma_probe16(_bf(a).to(DEV), _bf(b).to(DEV)).cpu()
+        assert d[3, 11].item() == pytest.approx(2.0, abs=1e-2)
+        assert d.abs().sum().item() == pytest.approx(2.0, abs=1e-2)
+
+
+class TestNorms:
+    def test_rmsnorm(self):
+        x = _bf(torch.randn(33, 4096)).to(DEV)
+        w = _bf(torch.randn(4096)).to(DEV)
+        got = ops.rmsnorm(x, w, 1e-5)
+        want = torch_ref.rmsnorm(x.cpu(), w.cpu(), 1e-5)
+        _assert_close(got, want, atol=2e-2, name="rmsnorm")
+
+    def test_add_rmsnorm(self):
+        r = _bf(torch.randn(17, 2048)).to(DEV)
+        d = _bf(torch.randn(17, 2048)).to(DEV)
+        w = _bf(torch.rand(2048) + 0.5).to(DEV)
+        r2, y = ops.add_rmsnorm(r, d, w, 1e-5)
+        r2_ref, y_ref = torch_ref.add_rmsnorm(r.cpu(), d.cpu(), w.cpu(), 1e-5)
+        _assert_close(r2, r2_ref, atol=2e-2, name="add_rmsnorm resid")
+        _assert_close(y, y_ref, atol=2e-2, name="add_rmsnorm y")
+
+    def test_rmsnorm_row_independence(self):
+        x = _bf(torch.randn(4, 256)).to(DEV)
+        w = _bf(torch.ones(256)).to(DEV)
+        full = ops.rmsnorm(x, w, 1e-5)
+        row = ops.rmsnorm(x[2:3].contiguous(), w, 1e-5)
+        assert torch.equal(full[2:3], row)
+
+
+class TestRoPE:
+    @pytest.mark.parametrize("hd", [32, 64, 128])
+    def test_vs_ref(self, hd):
+        t, hq, hk = 9, 4, 2
+        cos, sin = torch_ref.rope_tables(hd, 64, 10000.0, DEV)
+        q = _bf(torch.randn(t, hq, hd)).to(DEV)
+        k = _bf(torch.randn(t, hk, hd)).to(DEV)
+        q_ref, k_ref = torch_ref.rope(q.cpu(), k.cpu(), cos.cpu(), sin.cpu(), 3)
+        q2, k2 = ops.rope(q, k, cos, sin, 3)
+        _assert_close(q2, q_ref, atol=2e-2, name=f"rope q hd={hd}")
+        _assert_close(k2, k_ref, atol=2e-2, name=f"rope k hd={hd}")
+
+
+class TestSwiGLU:
+    def test_fused_halves(self):
+        t, f = 13, 1024
+        gu = _bf(torch.randn(t, 2 * f)).to(DEV)
+        got = ops.swiglu(gu[:, :f], gu[:, f:])
+        want = torch_ref.swiglu(gu[:, :f].cpu(), gu[:, f:].cpu())
+        _assert_close(got, want, atol=2e-2, name="swiglu")
+
+
+class TestKVWrite:
+    def test_paged_scatter(self):
+        kh, hd, ps, npg = 2, 128, 16, 8
+        kc = torch.zeros(npg, ps, kh, hd, dtype=torch.bfloat16, device=DEV)
+        vc = torch.zeros_like(kc)
+        table = torch.tensor([5, 2, 7, 0, 1, 3, 4, 6], dtype=torch.int32, device=DEV)
+        k = _bf(torch.randn(40, kh, hd)).to(DEV)
+        v = _bf(torch.randn(40, kh, hd)).to(DEV)
+        ops.kv_write(kc, vc, table, 3, k, v)
+        kc_ref = torch.zeros(npg, ps, kh, hd)
+        vc_ref = torch.zeros(npg, ps, kh, hd)
+        torch_ref.kv_write(kc_ref, vc_ref, table.cpu(), 3, k.float().cpu(), v.float().cpu())
+        assert torch.equal(kc.float().cpu(), kc_ref)
+        assert torch.equal(vc.float().cpu(), vc_ref)
+
+
+class TestDecodeAttention:
+    @pytest.mark.parametrize("seq_len", [1, 7, 255, 256, 1000])
+    @pytest.mark.parametrize("hd,group", [(128, 4), (64, 8), (32, 2)])
+    def test_vs_ref(self, seq_len, hd, group):
+        kh = 2
+        hq = kh * group
+        ps = 64
+        npg = (seq_len + ps - 1) // ps + 1
+        torch.manual_seed(seq_len * hd)
+        q = _bf(torch.randn(hq, hd)).to(DEV)
+        kc = _bf(torch.randn(npg, ps, kh, hd)).to(DEV)
+        vc = _bf(torch.randn(npg, ps, kh, hd)).to(DEV)
+        perm = torch.randperm(npg, dtype=torch.int32).to(DEV)
+        got = ops.attn_decode_paged(q, kc, vc, perm, seq_len)
+        want = torch_ref.attn_decode_paged(
+            q.cpu(), kc.cpu(), vc.cpu(), perm.cpu(), seq_len
+        )
+        _assert_close(got, want, atol=2e-2, name=f"decode s={seq_len} hd={hd}")
+
+
+class TestPrefillAttention:
+    @pytest.mark.parametrize("tq", [1, 16, 100, 129, 512])
+    def test_mfma_vs_ref_hd128(self, tq):
+        hq, kh, hd = 8, 2, 128
+        torch.manual_seed(tq)
+        q = _bf(torch.randn(tq, hq, hd)).to(DEV)
+        k = _bf(torch.randn(tq, kh, hd)).to(DEV)
+        v = _bf(torch.randn(tq, kh, hd)).to(DEV)
+        got = ops.attn_prefill(q, k, v)
+        want = torch_ref.attn_prefill(q.cpu(), k.cpu(), v.cpu())
+        _assert_close(got, want, atol=2.5e-2, name=f"prefill mfma tq={tq}")
+
+    @pytest.mark.parametrize("hd", [32, 64])
+    def test_simple_vs_ref(self, hd):
+        tq, hq, kh = 77, 4, 2
+        q = _bf(torch.randn(tq, hq, hd)).to(DEV)
+        k = _bf(torch.randn(tq, kh, hd)).to(DEV)
+        v = _bf(torch.randn(tq, kh, hd)).to(DEV)
+        got = ops.attn_prefill(q, k, v)
+        want = torch_ref.attn_prefill(q.cpu(), k.cpu(), v.cpu())
+        _assert_close(got, want, atol=2.5e-2, name=f"prefill simple hd={hd}")
+
+    def test_mfma_equals_simple(self):
+        from adversarial_spec_amd.ops import _advspec_hip
+
+        tq, hq, kh, hd = 200, 4, 2, 128
+        q = _bf(torch.randn(tq, hq, hd)).to(DEV)
+        k = _bf(torch.randn(tq, kh, hd)).to(DEV)
+        v = _bf(torch.randn(tq, kh, hd)).to(DEV)
+        scale = 1.0 / math.sqrt(hd)
+        a = _advspec_hip.attn_prefill(q, k, v, scale, True, 0)
+        b = _advspec_hip.attn_prefill_simple(q, k, v, scale, True, 0)
+        _assert_close(a, b, atol=2e-2, name="mfma vs simple")
+
+    def test_spiked_key_forces_rescale(self):
+        """Online-softmax rescale path (guide §5.4 rule 26): one huge key
+        late in the sequence forces the running max to jump."""
+        tq, hq, kh, hd = 128, 2, 1, 128
+        q = _bf(torch.randn(tq, hq, hd) * 0.1).to(DEV)
+        k = _bf(torch.randn(tq, kh, hd) * 0.1).to(DEV)
+        v = _bf(torch.randn(tq, kh, hd)).to(DEV)
+        kf = k.float()
+        kf[100] = q.float()[120, 0] * 3.0  # spike aligned with a late query
+        k = _bf(kf).to(DEV)
+        got = ops.attn_prefill(q, k, v)
+        want = torch_ref.attn_prefill(q.cpu(), k.cpu(), v.cpu())
+        _assert_close(got, want, atol=2.5e-2, name="prefill spiked")
+
+
+class TestSampling:
+    def test_greedy_is_argmax(self):
+        logits = _bf(torch.randn(128256)).to(DEV)
+        tok = ops.sample(logits, temperature=0.0)
+        assert tok == int(logits.float().argmax().item())
+
+    def test_seeded_deterministic(self):
+        logits = _bf(torch.randn(1000)).to(DEV)
+        a = ops.sample(logits, temperature=0.7, seed=42)
+        b = ops.sample(logits, temperature=0.7, seed=42)
+        assert a == b
+
+    def test_distribution_follows_logits(self):
+        logits = torch.full((512,), -10.0)
+        logits[7] = 5.0
+        logits[11] = 4.0
+        lb = _bf(logits).to(DEV)
+        counts = {}
+        for s in range(200):
+            t = ops.sample(lb, temperature=1.0, seed=s)
+            counts[t] = counts.get(t, 0) + 1
+        assert set(counts) <= {7, 11}
+        assert counts.get(7, 0) > counts.get(11, 0)
+
+    def test_valid_token_range(self):
+        logits = _bf(torch.randn(128256)).to(DEV)
+        for s in (1, 2, 3):
+            t = ops.sample(logits, temperature=0.7, seed=s)
+            assert 0 <= t < 128256
