@@ -5,11 +5,12 @@ import torch
 
 pytestmark = pytest.mark.gpu
 
+from adversarial_spec_amd.models.config import LlamaConfig
+
 if torch.cuda.is_available():
     from adversarial_spec_amd import ops
     from adversarial_spec_amd.engine.local import LocalEngine
     from adversarial_spec_amd.models import LlamaModel
-    from adversarial_spec_amd.models.config import LlamaConfig
 
 DEV = "cuda:0"
 
