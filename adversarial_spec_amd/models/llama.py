@@ -245,9 +245,16 @@ class LlamaModel:
         cache.seq_len = tokens.shape[0]
         return logits
 
-    def decode_one(self, token: int, cache: PagedKVCache) -> torch.Tensor:
-        """Append one token; returns next-token logits [vocab]."""
-        tok = torch.tensor([token], device=self.device, dtype=torch.long)
+    def decode_one(self, token, cache: PagedKVCache) -> torch.Tensor:
+        """Append one token; returns next-token logits [vocab].
+
+        `token` may be a Python int or a device scalar tensor (the async
+        decode loop feeds the sampled id back without a host round-trip).
+        """
+        if isinstance(token, torch.Tensor):
+            tok = token.reshape(1).to(torch.long)
+        else:
+            tok = torch.tensor([token], device=self.device, dtype=torch.long)
         logits = self._forward(tok, cache, cache.seq_len)
         cache.seq_len += 1
         return logits

@@ -152,9 +152,9 @@ __global__ void __launch_bounds__(256) attn_decode_split_kernel(
   }
   __syncthreads();
 
-  // threads [0, group*hd) combine the 4 wave-states for one (gi, dim)
-  const int tid = threadIdx.x;
-  if (tid < group * hd) {
+  // threads combine the 4 wave-states, one (gi, dim) pair each; group*hd
+  // can exceed the block size (e.g. 8*128), so stride the block over it.
+  for (int tid = threadIdx.x; tid < group * hd; tid += blockDim.x) {
     const int gi = tid / hd, dd = tid % hd;
     float M = -INFINITY;
 #pragma unroll

@@ -89,7 +89,11 @@ void rope_inplace(torch::Tensor q, torch::Tensor k, torch::Tensor cost,
                   torch::Tensor sint, int64_t pos0) {
   CHECK_BF16_CUDA(q);
   CHECK_BF16_CUDA(k);
-  TORCH_CHECK(q.is_contiguous() && k.is_contiguous(), "rope: q/k contiguous");
+  // q/k may be strided VIEWS of the fused QKV output: [t, h, hd] with an
+  // arbitrary row stride but contiguous head/dim dims.
+  TORCH_CHECK(q.stride(2) == 1 && k.stride(2) == 1, "rope: dim contiguous");
+  TORCH_CHECK(q.stride(1) == q.size(2) && k.stride(1) == k.size(2),
+              "rope: head dim contiguous");
   TORCH_CHECK(cost.scalar_type() == at::kFloat, "rope: cos table f32");
   const int t = q.size(0), hq = q.size(1), hd = q.size(2);
   const int hk = k.size(1);
@@ -98,7 +102,7 @@ void rope_inplace(torch::Tensor q, torch::Tensor k, torch::Tensor cost,
   const int blocks = (waves * 64 + 255) / 256;
   rope_kernel<<<blocks, 256, 0, cur_stream()>>>(
       uptr_mut(q), uptr_mut(k), cost.data_ptr<float>(), sint.data_ptr<float>(),
-      t, hq, hk, hd, (int)pos0);
+      t, hq, hk, hd, (int)pos0, q.stride(0), k.stride(0));
 }
 
 torch::Tensor swiglu(torch::Tensor gate, torch::Tensor up) {
@@ -232,6 +236,19 @@ torch::Tensor attn_decode_paged(torch::Tensor q, torch::Tensor kc,
   return out;
 }
 
+// Async variant: write the sampled token id into out[idx] (int32, on
+// device) without any host synchronization — the decode loop stays on the
+// GPU and the host checks stop conditions every N tokens.
+void sample_to(torch::Tensor logits, double temp, int64_t seed,
+               torch::Tensor out, int64_t idx) {
+  CHECK_BF16_CUDA(logits);
+  TORCH_CHECK(out.scalar_type() == at::kInt && out.is_cuda());
+  auto lc = logits.contiguous();
+  sample_kernel<<<1, 256, 0, cur_stream()>>>(
+      uptr(lc), (int)lc.numel(), (float)temp, (uint32_t)seed,
+      out.data_ptr<int>() + idx);
+}
+
 int64_t sample(torch::Tensor logits, double temp, double top_p, int64_t seed) {
   CHECK_BF16_CUDA(logits);
   TORCH_CHECK(top_p >= 1.0, "kernel sample handles top_p == 1 (nucleus is a cold path)");
@@ -257,4 +274,5 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("mfma_probe16", &mfma_probe16, "MFMA 16x16x32 fragment-map probe");
   m.def("attn_decode_paged", &attn_decode_paged, "paged decode attention");
   m.def("sample", &sample, "fused temperature softmax sample");
+  m.def("sample_to", &sample_to, "async on-device sample into out[idx]");
 }

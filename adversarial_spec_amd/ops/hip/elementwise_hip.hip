@@ -102,7 +102,10 @@ add_rmsnorm_kernel(const ushort_t *__restrict__ resid,
 extern "C" __global__ void __launch_bounds__(256)
 rope_kernel(ushort_t *__restrict__ q, ushort_t *__restrict__ k,
             const float *__restrict__ cost, const float *__restrict__ sint,
-            int t, int hq, int hk, int hd, int pos0) {
+            int t, int hq, int hk, int hd, int pos0,
+            long q_rstride, long k_rstride) {
+  // q/k may be strided views into the fused QKV GEMM output (row strides in
+  // elements); head/dim dims are contiguous.
   const int wave_id = (blockIdx.x * blockDim.x + threadIdx.x) / WAVE;
   const int lane = threadIdx.x & (WAVE - 1);
   const int heads = hq + hk;
@@ -113,8 +116,8 @@ rope_kernel(ushort_t *__restrict__ q, ushort_t *__restrict__ k,
   const int half = hd / 2;
 
   ushort_t *base = (head < hq)
-                       ? q + ((size_t)tok * hq + head) * hd
-                       : k + ((size_t)tok * hk + (head - hq)) * hd;
+                       ? q + (size_t)tok * q_rstride + (size_t)head * hd
+                       : k + (size_t)tok * k_rstride + (size_t)(head - hq) * hd;
   const float *crow = cost + (size_t)pos * half;
   const float *srow = sint + (size_t)pos * half;
 

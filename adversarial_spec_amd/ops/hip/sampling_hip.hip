@@ -66,8 +66,8 @@ sample_kernel(const ushort_t *__restrict__ logits, int vocab, float temp,
   }
   z = block_reduce_sum(z, scratch);
 
-  // draw u in (0, Z]
-  const float target = uniform01(seed) * z;
+  // draw u in (0, Z]; golden-ratio offset avoids the degenerate hash(0)=0
+  const float target = uniform01(seed ^ 0x9e3779b9u) * z;
 
   // pass 3: find the crossing chunk, then scan inside it
   if (tid == 0) result = amax;  // fallback: rounding may exhaust the loop
