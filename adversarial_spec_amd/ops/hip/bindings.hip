@@ -27,7 +27,7 @@ __global__ void rmsnorm_kernel(const ushort_t*, const ushort_t*, ushort_t*, int,
 __global__ void add_rmsnorm_kernel(const ushort_t*, const ushort_t*, const ushort_t*,
                                    ushort_t*, ushort_t*, int, float);
 __global__ void rope_kernel(ushort_t*, ushort_t*, const float*, const float*,
-                            int, int, int, int, int);
+                            int, int, int, int, int, long, long);
 __global__ void swiglu_kernel(const ushort_t*, const ushort_t*, ushort_t*, int, int, int);
 __global__ void kv_write_kernel(const ushort_t*, const ushort_t*, ushort_t*, ushort_t*,
                                 const int*, int, int, int, int, int);
