@@ -23,10 +23,13 @@ class LlamaConfig:
     max_seq_len: int = 8192
     rope_theta: float = 500000.0
     norm_eps: float = 1e-5
+    # explicit head_dim survives tensor-parallel sharding (where n_heads
+    # shrinks but the per-head width does not); None derives dim/n_heads.
+    head_dim_override: int = 0
 
     @property
     def head_dim(self) -> int:
-        return self.dim // self.n_heads
+        return self.head_dim_override or self.dim // self.n_heads
 
     def param_count(self) -> int:
         """Approximate parameter count (embeddings untied)."""

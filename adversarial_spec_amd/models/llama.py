@@ -70,7 +70,17 @@ class LlamaModel:
     """Decoder-only transformer (RMSNorm / RoPE / GQA / SwiGLU)."""
 
     def __init__(self, config: LlamaConfig, device="cpu",
-                 dtype: Optional[torch.dtype] = None, seed: int = 0) -> None:
+                 dtype: Optional[torch.dtype] = None, seed: int = 0,
+                 tp=None) -> None:
+        # tp: Optional[parallel.tp.TPContext] — head-sharded tensor
+        # parallelism; self.config becomes the per-rank LOCAL config while
+        # full_config keeps the replicated dims (embed/norm/lm_head).
+        self.full_config = config
+        self.tp = tp
+        if tp is not None and tp.size > 1:
+            from ..parallel.tp import shard_config
+
+            config = shard_config(config, tp.size)
         self.config = config
         self.device = torch.device(device)
         if dtype is None:
@@ -95,8 +105,11 @@ class LlamaModel:
         BASELINE.json mandates random-init opponent weights for the
         benchmark (no network for checkpoints); std is scaled so logits
         stay finite through deep stacks.
+
+        Under TP every rank generates the FULL weights from the shared seed
+        and keeps its shard, so TP=N is numerically identical to TP=1.
         """
-        c = self.config
+        c = self.full_config
         g = torch.Generator(device="cpu").manual_seed(self.seed)
 
         def t(*shape, std=0.02):
@@ -116,16 +129,33 @@ class LlamaModel:
         self.embed = t(c.vocab_size, d)
         self.final_norm = torch.ones(d, device=self.device, dtype=self.dtype)
         self.lm_head = t(d, c.vocab_size)
+        tp = self.tp
         self.layers = []
         for _ in range(c.n_layers):
+            wqkv = t(d, (c.n_heads + 2 * c.n_kv_heads) * hd)
+            wo = t(c.n_heads * hd, d, std=proj_std)
+            w_gate_up = t(d, 2 * c.ffn_dim)
+            w_down = t(c.ffn_dim, d, std=proj_std)
+            if tp is not None and tp.size > 1:
+                from ..parallel.tp import (
+                    shard_down,
+                    shard_gate_up,
+                    shard_o,
+                    shard_qkv,
+                )
+
+                wqkv = shard_qkv(wqkv, c, tp.size, tp.rank)
+                wo = shard_o(wo, c, tp.size, tp.rank)
+                w_gate_up = shard_gate_up(w_gate_up, c, tp.size, tp.rank)
+                w_down = shard_down(w_down, c, tp.size, tp.rank)
             self.layers.append(
                 LayerWeights(
                     attn_norm=torch.ones(d, device=self.device, dtype=self.dtype),
-                    wqkv=t(d, (c.n_heads + 2 * c.n_kv_heads) * hd),
-                    wo=t(c.n_heads * hd, d, std=proj_std),
+                    wqkv=wqkv,
+                    wo=wo,
                     mlp_norm=torch.ones(d, device=self.device, dtype=self.dtype),
-                    w_gate_up=t(d, 2 * c.ffn_dim),
-                    w_down=t(c.ffn_dim, d, std=proj_std),
+                    w_gate_up=w_gate_up,
+                    w_down=w_down,
                 )
             )
         return self
@@ -209,9 +239,14 @@ class LlamaModel:
         t = tokens.shape[0]
         h, kh, hd = c.n_heads, c.n_kv_heads, c.head_dim
         resid = self.embed[tokens]  # [t, d]
+
+        # batch-1 decode projections stream weights through the hand-written
+        # GEMV kernel; prefill (t>1) goes through hipBLASLt.
+        mm = ops.gemv if t == 1 else (lambda a, b: a @ b)
+
         normed = ops.rmsnorm(resid, self.layers[0].attn_norm, c.norm_eps)
         for i, L in enumerate(self.layers):
-            qkv = normed @ L.wqkv  # [t, (h+2kh)*hd]
+            qkv = mm(normed, L.wqkv)  # [t, (h+2kh)*hd]
             q = qkv[:, : h * hd].view(t, h, hd)
             k = qkv[:, h * hd : (h + kh) * hd].view(t, kh, hd)
             v = qkv[:, (h + kh) * hd :].view(t, kh, hd)
@@ -225,16 +260,20 @@ class LlamaModel:
                 attn = ops.attn_decode_paged(
                     q[0], cache.k[i], cache.v[i], cache.page_table, pos0 + 1, self.scale
                 ).unsqueeze(0)
-            attn_out = attn.reshape(t, h * hd) @ L.wo
+            attn_out = mm(attn.reshape(t, h * hd), L.wo)
+            if self.tp is not None and self.tp.size > 1:
+                self.tp.all_reduce_(attn_out)  # row-parallel wo partial sums
             resid, normed = ops.add_rmsnorm(resid, attn_out, L.mlp_norm, c.norm_eps)
-            gu = normed @ L.w_gate_up
+            gu = mm(normed, L.w_gate_up)
             act = ops.swiglu(gu[:, : c.ffn_dim], gu[:, c.ffn_dim :])
-            mlp_out = act @ L.w_down
+            mlp_out = mm(act, L.w_down)
+            if self.tp is not None and self.tp.size > 1:
+                self.tp.all_reduce_(mlp_out)  # row-parallel down partial sums
             next_norm = (
                 self.layers[i + 1].attn_norm if i + 1 < c.n_layers else self.final_norm
             )
             resid, normed = ops.add_rmsnorm(resid, mlp_out, next_norm, c.norm_eps)
-        logits = normed[-1:] @ self.lm_head  # [1, vocab]
+        logits = ops.gemv(normed[-1:].contiguous(), self.lm_head)  # [1, vocab]
         return logits[0]
 
     def prefill(self, tokens: torch.Tensor, cache: PagedKVCache) -> torch.Tensor:

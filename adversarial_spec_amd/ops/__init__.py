@@ -134,6 +134,15 @@ def kv_write(
     torch_ref.kv_write(k_cache, v_cache, page_table, pos0, k, v)
 
 
+def gemv(x: torch.Tensor, w: torch.Tensor) -> torch.Tensor:
+    """Batch-1 matmul y = x @ w. On GPU, a hand-written weight-streaming
+    kernel (decode's dominant cost is reading w once from HBM3E); falls
+    back to torch.matmul when the column count is not 64-aligned."""
+    if _on_gpu(x) and w.size(1) % 64 == 0 and hip_available():
+        return _require_hip().gemv(x, w)
+    return x @ w
+
+
 def sample(
     logits: torch.Tensor,
     temperature: float = 0.7,

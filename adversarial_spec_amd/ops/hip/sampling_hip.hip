@@ -2,45 +2,52 @@
 // On-device sampling: fused temperature scale + softmax + inverse-CDF draw
 // (and greedy argmax), one block over the vocab row.
 //
-// Decode emits one token at a time; doing the sample on-device avoids a
-// logits round-trip to the host (vocab 128256 x 2 B each step) — only the
-// 4-byte token id crosses PCIe. The random draw is a counter-based hash of
-// the host-provided seed, so replays (HIP graphs) stay deterministic.
+// Decode emits one token at a time; sampling on-device avoids a logits
+// round-trip (vocab 128256 x 2 B per step) — only a 4-byte token id ever
+// reaches the host, and only every N tokens (engine/local.py async loop).
+// The random draw is a counter-based hash of the host-provided seed, so
+// replays (HIP graphs) stay deterministic.
 //
-// Exact nucleus (top-p < 1) requires a sorted vocab and runs on the cold
-// path in Python (ops/__init__.py); the kernel handles the default
-// temperature/greedy paths (reference defaults: temp 0.7 / 0.3,
-// BASELINE.md sampling row).
+// All passes read the logits as bf16x8 (16 B/lane): the row was just
+// written by the lm_head GEMV so it is L2-resident; vectorized reads keep
+// the single-workgroup scan latency-bound rather than issue-bound.
+//
+// Exact nucleus (top-p < 1) needs a sorted vocab and runs on the cold path
+// in Python (ops/__init__.py); this kernel is the default temperature /
+// greedy path (reference defaults 0.7 / 0.3: models.py:626).
 
 #include "common.h"
 
-// logits: [vocab] bf16. out: int32[1]. temp <= 0 -> greedy.
+// logits: [vocab] bf16, vocab % 8 == 0. out: int32. temp <= 0 -> greedy.
 extern "C" __global__ void __launch_bounds__(256)
 sample_kernel(const ushort_t *__restrict__ logits, int vocab, float temp,
               uint32_t seed, int *__restrict__ out) {
   __shared__ float scratch[16];
-  __shared__ float chunk_vals[256];
+  __shared__ float tsum[256];
   __shared__ int result;
   const int tid = threadIdx.x;
+  const int nvec = vocab / 8;
+  const bf16x8 *lv = (const bf16x8 *)logits;
 
-  // pass 1: max (argmax for greedy)
+  // ---- pass 1: max (+argmax for greedy) ----
   float vmax = -INFINITY;
   int amax = 0;
-  for (int i = tid; i < vocab; i += blockDim.x) {
-    const float v = bf16_to_f32(logits[i]);
-    if (v > vmax) { vmax = v; amax = i; }
+  for (int i = tid; i < nvec; i += blockDim.x) {
+    const f32x8 v = unpack8(lv[i]);
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      if (v.v[j] > vmax) { vmax = v.v[j]; amax = i * 8 + j; }
+    }
   }
-  // block argmax via LDS pairs
-  __shared__ float mv[16];
-  __shared__ int mi[16];
   {
-    // wave-level argmax
 #pragma unroll
     for (int off = 32; off > 0; off >>= 1) {
       const float ov = __shfl_xor(vmax, off, WAVE);
       const int oi = __shfl_xor(amax, off, WAVE);
       if (ov > vmax || (ov == vmax && oi < amax)) { vmax = ov; amax = oi; }
     }
+    __shared__ float mv[16];
+    __shared__ int mi[16];
     const int wid = tid / WAVE;
     if ((tid & (WAVE - 1)) == 0) { mv[wid] = vmax; mi[wid] = amax; }
     __syncthreads();
@@ -56,39 +63,68 @@ sample_kernel(const ushort_t *__restrict__ logits, int vocab, float temp,
     if (tid == 0) *out = amax;
     return;
   }
-
   const float inv_t = 1.0f / temp;
 
-  // pass 2: sum of exp((v - vmax)/temp)
+  // ---- pass 2: Z = sum exp((v - max)/temp) ----
   float z = 0.f;
-  for (int i = tid; i < vocab; i += blockDim.x) {
-    z += __expf((bf16_to_f32(logits[i]) - vmax) * inv_t);
+  for (int i = tid; i < nvec; i += blockDim.x) {
+    const f32x8 v = unpack8(lv[i]);
+#pragma unroll
+    for (int j = 0; j < 8; ++j) z += __expf((v.v[j] - vmax) * inv_t);
   }
   z = block_reduce_sum(z, scratch);
 
   // draw u in (0, Z]; golden-ratio offset avoids the degenerate hash(0)=0
   const float target = uniform01(seed ^ 0x9e3779b9u) * z;
 
-  // pass 3: find the crossing chunk, then scan inside it
-  if (tid == 0) result = amax;  // fallback: rounding may exhaust the loop
+  // ---- pass 3: find the crossing element ----
+  if (tid == 0) result = amax;  // fallback if rounding exhausts the scan
   __syncthreads();
   float running = 0.f;
-  for (int base = 0; base < vocab; base += blockDim.x) {
-    const int i = base + tid;
-    const float e =
-        (i < vocab) ? __expf((bf16_to_f32(logits[i]) - vmax) * inv_t) : 0.f;
-    chunk_vals[tid] = e;
-    float csum = block_reduce_sum(e, scratch);
+  const int per_iter = blockDim.x * 8;  // elements per block iteration
+  for (int base = 0; base < vocab; base += per_iter) {
+    const int i = base / 8 + tid;
+    float my = 0.f;
+    f32x8 ev;
+    if (i < nvec) {
+      const f32x8 v = unpack8(lv[i]);
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        ev.v[j] = __expf((v.v[j] - vmax) * inv_t);
+        my += ev.v[j];
+      }
+    } else {
+#pragma unroll
+      for (int j = 0; j < 8; ++j) ev.v[j] = 0.f;
+    }
+    tsum[tid] = my;
+    const float csum = block_reduce_sum(my, scratch);
     __syncthreads();
     if (running + csum >= target) {
+      // crossing chunk: thread 0 scans the 256 per-thread sums, the owner
+      // thread then pinpoints its element.
+      __shared__ int owner;
+      __shared__ float owner_base;
       if (tid == 0) {
         float acc = running;
-        int pick = -1;
-        for (int j = 0; j < (int)blockDim.x && base + j < vocab; ++j) {
-          acc += chunk_vals[j];
-          if (acc >= target) { pick = base + j; break; }
+        int who = blockDim.x - 1;
+        for (int s = 0; s < (int)blockDim.x; ++s) {
+          if (acc + tsum[s] >= target) { who = s; break; }
+          acc += tsum[s];
         }
-        result = (pick >= 0) ? pick : amax;
+        owner = who;
+        owner_base = acc;
+      }
+      __syncthreads();
+      if (tid == owner) {
+        float acc = owner_base;
+        int pick = -1;
+#pragma unroll
+        for (int j = 0; j < 8; ++j) {
+          acc += ev.v[j];
+          if (acc >= target) { pick = base + tid * 8 + j; break; }
+        }
+        if (pick >= 0 && pick < vocab) result = pick;
       }
       break;
     }
