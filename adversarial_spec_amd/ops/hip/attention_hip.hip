@@ -42,7 +42,7 @@ DEVINL float group_reduce_sum(float v) {
 //                   ws_acc:     [kh, n_splits, group, hd]
 // ---------------------------------------------------------------------------
 
-template <int LPP>
+template <int LPP, int MG>
 __global__ void __launch_bounds__(256) attn_decode_split_kernel(
     const ushort_t *__restrict__ q,   // [hq, hd]
     const ushort_t *__restrict__ kc, const ushort_t *__restrict__ vc,
@@ -65,9 +65,9 @@ __global__ void __launch_bounds__(256) attn_decode_split_kernel(
                                      (split + 1) * split_len));
 
   // q fragments: [group][8] for this lane's dim slice
-  float qf[MAXG][8];
+  float qf[MG][8];
 #pragma unroll
-  for (int gi = 0; gi < MAXG; ++gi) {
+  for (int gi = 0; gi < MG; ++gi) {
     if (gi < group) {
       const bf16x8 qv =
           ((const bf16x8 *)(q + ((size_t)(g * group + gi)) * hd))[sl];
@@ -77,9 +77,9 @@ __global__ void __launch_bounds__(256) attn_decode_split_kernel(
     }
   }
 
-  float m[MAXG], l[MAXG], acc[MAXG][8];
+  float m[MG], l[MG], acc[MG][8];
 #pragma unroll
-  for (int gi = 0; gi < MAXG; ++gi) {
+  for (int gi = 0; gi < MG; ++gi) {
     m[gi] = -INFINITY;
     l[gi] = 0.f;
 #pragma unroll
@@ -93,7 +93,7 @@ __global__ void __launch_bounds__(256) attn_decode_split_kernel(
     const f32x8 vd = unpack8(((const bf16x8 *)(vc + row + (size_t)g * hd))[sl]);
 
 #pragma unroll
-    for (int gi = 0; gi < MAXG; ++gi) {
+    for (int gi = 0; gi < MG; ++gi) {
       if (gi >= group) break;
       float dot = 0.f;
 #pragma unroll
@@ -114,7 +114,7 @@ __global__ void __launch_bounds__(256) attn_decode_split_kernel(
 #pragma unroll
   for (int off = LPP; off < WAVE; off <<= 1) {
 #pragma unroll
-    for (int gi = 0; gi < MAXG; ++gi) {
+    for (int gi = 0; gi < MG; ++gi) {
       if (gi >= group) break;
       const float mo = __shfl_xor(m[gi], off, WAVE);
       const float lo = __shfl_xor(l[gi], off, WAVE);
@@ -140,7 +140,7 @@ __global__ void __launch_bounds__(256) attn_decode_split_kernel(
 
   if (lane < LPP) {  // one lane per dim slice (sub==0 lanes)
 #pragma unroll
-    for (int gi = 0; gi < MAXG; ++gi) {
+    for (int gi = 0; gi < MG; ++gi) {
       if (gi >= group) break;
       if (lane == 0) {
         lm[wid * MAXG + gi] = m[gi];
@@ -193,17 +193,42 @@ attn_decode_combine_kernel(const float *__restrict__ ws_m,
   const int dd = threadIdx.x;
   if (dd >= hd) return;
 
+  // 4-way unrolled split walk keeps >=4 independent loads in flight
+  // (the serial version was latency-bound at large split counts).
   float M = -INFINITY;
-  for (int s = 0; s < n_splits; ++s)
-    M = fmaxf(M, ws_m[((size_t)g * n_splits + s) * group + gi]);
-  float L = 0.f, A = 0.f;
-  for (int s = 0; s < n_splits; ++s) {
+  {
+    float m4[4] = {-INFINITY, -INFINITY, -INFINITY, -INFINITY};
+    int s = 0;
+    for (; s + 4 <= n_splits; s += 4) {
+#pragma unroll
+      for (int u = 0; u < 4; ++u)
+        m4[u] = fmaxf(m4[u], ws_m[((size_t)g * n_splits + s + u) * group + gi]);
+    }
+    for (; s < n_splits; ++s)
+      m4[0] = fmaxf(m4[0], ws_m[((size_t)g * n_splits + s) * group + gi]);
+    M = fmaxf(fmaxf(m4[0], m4[1]), fmaxf(m4[2], m4[3]));
+  }
+  float L4[4] = {0.f, 0.f, 0.f, 0.f}, A4[4] = {0.f, 0.f, 0.f, 0.f};
+  int s = 0;
+  for (; s + 4 <= n_splits; s += 4) {
+#pragma unroll
+    for (int u = 0; u < 4; ++u) {
+      const size_t base = ((size_t)g * n_splits + s + u) * group + gi;
+      const float mw = ws_m[base];
+      const float sc = (mw == -INFINITY) ? 0.f : __expf(mw - M);
+      L4[u] += ws_l[base] * sc;
+      A4[u] += ws_acc[base * hd + dd] * sc;
+    }
+  }
+  for (; s < n_splits; ++s) {
     const size_t base = ((size_t)g * n_splits + s) * group + gi;
     const float mw = ws_m[base];
     const float sc = (mw == -INFINITY) ? 0.f : __expf(mw - M);
-    L += ws_l[base] * sc;
-    A += ws_acc[base * hd + dd] * sc;
+    L4[0] += ws_l[base] * sc;
+    A4[0] += ws_acc[base * hd + dd] * sc;
   }
+  const float L = L4[0] + L4[1] + L4[2] + L4[3];
+  const float A = A4[0] + A4[1] + A4[2] + A4[3];
   out[(size_t)h * hd + dd] = f32_to_bf16(A / L);
 }
 
@@ -277,23 +302,31 @@ extern "C" void launch_attn_decode_split(
     float *ws_acc, hipStream_t stream) {
   dim3 grid(kh, n_splits);
   const int lds = (4 * MAXG * 2 + 4 * MAXG * hd) * sizeof(float);
+  // MG = smallest supported bound >= group keeps the per-head state arrays
+  // (q fragments + online-softmax accumulators) sized to the real GQA
+  // group: MG=8 cost 194 VGPR (2 waves/SIMD); MG=4 fits 4 waves/SIMD.
+#define DISPATCH_MG(LPP)                                                       \
+  do {                                                                         \
+    if (group <= 2)                                                            \
+     hipLaunchKernelGGL(( attn_decode_split_kernel<LPP, 2>), dim3(grid), dim3(256), lds, stream,             \
+          q, kc, vc, page_table, seq_len, scale, kh, group, hd, page,          \
+          split_len, ws_m, ws_l, ws_acc);                                      \
+    else if (group <= 4)                                                       \
+     hipLaunchKernelGGL(( attn_decode_split_kernel<LPP, 4>), dim3(grid), dim3(256), lds, stream,             \
+          q, kc, vc, page_table, seq_len, scale, kh, group, hd, page,          \
+          split_len, ws_m, ws_l, ws_acc);                                      \
+    else                                                                       \
+     hipLaunchKernelGGL(( attn_decode_split_kernel<LPP, 8>), dim3(grid), dim3(256), lds, stream,             \
+          q, kc, vc, page_table, seq_len, scale, kh, group, hd, page,          \
+          split_len, ws_m, ws_l, ws_acc);                                      \
+  } while (0)
+
   switch (hd / 8) {
-    case 4:
-     hipLaunchKernelGGL(( attn_decode_split_kernel<4>), dim3(grid), dim3(256), lds, stream, 
-          q, kc, vc, page_table, seq_len, scale, kh, group, hd, page,
-          split_len, ws_m, ws_l, ws_acc);
-      break;
-    case 8:
-     hipLaunchKernelGGL(( attn_decode_split_kernel<8>), dim3(grid), dim3(256), lds, stream, 
-          q, kc, vc, page_table, seq_len, scale, kh, group, hd, page,
-          split_len, ws_m, ws_l, ws_acc);
-      break;
-    case 16:
-     hipLaunchKernelGGL(( attn_decode_split_kernel<16>), dim3(grid), dim3(256), lds, stream, 
-          q, kc, vc, page_table, seq_len, scale, kh, group, hd, page,
-          split_len, ws_m, ws_l, ws_acc);
-      break;
+    case 4: DISPATCH_MG(4); break;
+    case 8: DISPATCH_MG(8); break;
+    case 16: DISPATCH_MG(16); break;
   }
+#undef DISPATCH_MG
 }
 
 extern "C" void launch_attn_decode_combine(

@@ -30,7 +30,25 @@ gemv_partial_kernel(const ushort_t *__restrict__ x,
 #pragma unroll
   for (int j = 0; j < 8; ++j) acc[j] = 0.f;
 
-  for (int k = k0 + kl; k < k1; k += 32) {
+  // 4-deep k unroll: 4 independent 16 B weight loads in flight per thread
+  // (single-load version measured latency-bound at ~2 TB/s).
+  int k = k0 + kl;
+  for (; k + 96 < k1; k += 128) {
+    bf16x8 wv[4];
+    float xv[4];
+#pragma unroll
+    for (int u = 0; u < 4; ++u) {
+      const int kk = k + 32 * u;
+      xv[u] = bf16_to_f32(x[kk]);
+      wv[u] = ((const bf16x8 *)(w + (size_t)kk * N + c0))[vc];
+    }
+#pragma unroll
+    for (int u = 0; u < 4; ++u) {
+#pragma unroll
+      for (int j = 0; j < 8; ++j) acc[j] += xv[u] * bf16_to_f32(wv[u].u[j]);
+    }
+  }
+  for (; k < k1; k += 32) {
     const float xv = bf16_to_f32(x[k]);
     const bf16x8 wv = ((const bf16x8 *)(w + (size_t)k * N + c0))[vc];
 #pragma unroll

@@ -36,6 +36,7 @@ from adversarial_spec_amd.engine.local import LocalEngine  # noqa: E402
 from adversarial_spec_amd.engine.scheduler import build_user_message  # noqa: E402
 from adversarial_spec_amd.parallel.consensus import (  # noqa: E402
     HDR,
+    AsyncRoundGather,
     pack_result,
     unpack_results,
 )
@@ -55,6 +56,9 @@ def parse_args():
                    help="critique length decoded per opponent per round")
     p.add_argument("--model", default="llama-3-8b")
     p.add_argument("--temperature", type=float, default=0.7)
+    p.add_argument("--tp", type=int, default=1,
+                   help="tensor-parallel degree: all ranks form ONE sharded "
+                        "opponent (BASELINE config 5: llama-3-70b --tp 8)")
     return p.parse_args()
 
 
@@ -78,26 +82,51 @@ def main() -> int:
         dist = dist_mod
         dist.init_process_group("nccl")
 
-    n_opp = args.opponents_per_gpu
-    engines = []
-    for i in range(n_opp):
-        # distinct seeds per (rank, opponent): heterogeneous random-init pool
-        engines.append(
+    tp_mode = args.tp > 1
+    if tp_mode:
+        if dist is None or world != args.tp:
+            print("--tp N needs a torchrun world of exactly N ranks", file=sys.stderr)
+            return 1
+        from adversarial_spec_amd.parallel.tp import TPContext
+
+        n_opp = 1
+        engines = [
             LocalEngine(
-                {"name": f"{args.model}-r{rank}o{i}", "arch": args.model},
-                device=device,
+                {"name": f"{args.model}-tp", "arch": args.model},
+                device=device, tp=TPContext.from_default_group(),
             )
-        )
+        ]
+    else:
+        n_opp = args.opponents_per_gpu
+        engines = []
+        for i in range(n_opp):
+            # distinct seeds per (rank, opponent): heterogeneous random-init pool
+            engines.append(
+                LocalEngine(
+                    {"name": f"{args.model}-r{rank}o{i}", "arch": args.model},
+                    device=device,
+                )
+            )
 
     spec = synthetic_spec(args.spec_tokens, seed=17)
     system_prompt = get_system_prompt("tech")
     user_message = build_user_message(spec, 1, "tech")
     max_gather_tokens = args.decode_tokens
 
+    # One fused all-gather per OPPONENT, launched on a dedicated comm
+    # stream as soon as that opponent's decode finishes: opponent i's
+    # consensus traffic overlaps opponent i+1's prefill/decode (the
+    # BASELINE north-star overlap; collectives are issued in the same
+    # opponent order on every rank).
+    gathers = (
+        [AsyncRoundGather(max_tokens=max_gather_tokens) for _ in range(n_opp)]
+        if (dist is not None and not tp_mode) else None
+    )
+
     def one_round():
         """One debate round for this rank's opponents + consensus gather."""
-        packed = []
-        for eng in engines:
+        local = []
+        for i, eng in enumerate(engines):
             text, _in, _out, _tm = eng.generate(
                 system_prompt,
                 user_message,
@@ -106,18 +135,23 @@ def main() -> int:
                 timeout=600.0,
             )
             ids = eng.tokenizer.encode(text)[: max_gather_tokens]
-            packed.append(
-                pack_result(ids, detect_agreement(text), False,
-                            max_gather_tokens, torch.device(device))
-            )
-        mine = torch.cat(packed)  # [n_opp * (HDR+max)]
-        if dist is not None:
-            out = torch.zeros(world * mine.numel(), dtype=torch.int32,
-                              device=device)
-            dist.all_gather_into_tensor(out, mine)
-        else:
-            out = mine
-        rows = out.view(world * n_opp, HDR + max_gather_tokens)
+            agreed = detect_agreement(text)
+            if gathers is not None:
+                gathers[i].launch(ids, agreed, False)
+            else:
+                local.append(
+                    pack_result(ids, agreed, False, max_gather_tokens,
+                                torch.device(device))
+                )
+        if gathers is not None:
+            results = []
+            all_agreed = True
+            for g in gathers:
+                r, ok = g.wait()
+                results.extend(r)
+                all_agreed = all_agreed and ok
+            return all_agreed
+        rows = torch.cat(local).view(n_opp, HDR + max_gather_tokens)
         results = unpack_results(rows)
         ok = [r for r in results if not r.error]
         return bool(ok) and all(r.agreed for r in ok)
@@ -143,7 +177,8 @@ def main() -> int:
         dist.all_reduce(e, op=dist.ReduceOp.MAX)
         elapsed = float(e.item())
 
-    total_critiques = world * n_opp * args.steps
+    n_job_opponents = n_opp if tp_mode else world * n_opp
+    total_critiques = n_job_opponents * args.steps
     value = total_critiques / elapsed
     ms_per_step = elapsed / args.steps * 1000.0
 
@@ -164,13 +199,14 @@ def main() -> int:
             "data": "synthetic",
             "config": {
                 "model": args.model,
-                "opponents": world * n_opp,
-                "opponents_per_gpu": n_opp,
+                "opponents": n_job_opponents,
+                "opponents_per_gpu": n_opp if not tp_mode else 1.0 / world,
                 "spec_tokens": args.spec_tokens,
                 "decode_tokens": args.decode_tokens,
-                "global_batch": world * n_opp,
+                "global_batch": n_job_opponents,
                 "seq_len": args.spec_tokens,
-                "parallelism": f"opponent-parallel dp{world}",
+                "parallelism": (f"tp{args.tp}" if tp_mode
+                                else f"opponent-parallel dp{world}"),
                 "temperature": args.temperature,
             },
         }))
