@@ -12,6 +12,7 @@ in HBM3E between rounds, so round k+1 pays only prefill+decode.
 
 from __future__ import annotations
 
+import os
 import threading
 import time
 from typing import Any, Optional
@@ -112,7 +113,11 @@ class LocalEngine:
                     and top_p >= 1.0
                     and ops.hip_available()
                 )
-                if use_async:
+                if use_async and max_new >= 8 and not os.environ.get("ADVSPEC_NO_GRAPH"):
+                    out_ids = self._decode_graphed(
+                        logits, cache, max_new, temperature, stop_ids, deadline
+                    )
+                elif use_async:
                     out_ids = self._decode_async(
                         logits, cache, max_new, temperature, stop_ids, deadline
                     )
@@ -156,6 +161,92 @@ class LocalEngine:
                 # the opponent (a partial decode is still a usable critique).
                 break
             logits = self.model.decode_one(tok, cache)
+        return out_ids
+
+    def _decode_graphed(self, logits, cache, max_new, temperature, stop_ids,
+                        deadline) -> list[int]:
+        """HIP-graph decode: the whole per-token step (sample -> 32-layer
+        forward -> logits copy -> position bump) is captured once and
+        replayed per token.
+
+        Eager decode is launch-gap bound (~420 kernel launches/token,
+        rocprof: dispatch span ~2x kernel busy); graph replay collapses that
+        to one hipGraphLaunch. All dynamic state (position, RNG, token ids)
+        lives in device words read in-kernel, so the captured launch
+        geometry is replay-invariant.
+        """
+        from ..ops import _load_hip
+
+        hip = _load_hip()
+        dev = self.device
+        prompt_len = cache.seq_len
+        max_total = min(cache.max_seq, prompt_len + max_new + 1)
+
+        pos_state = torch.tensor([prompt_len], dtype=torch.int32, device=dev)
+        step_state = torch.zeros(1, dtype=torch.int32, device=dev)
+        rng_state = torch.tensor([self._next_seed() | 1], dtype=torch.int32,
+                                 device=dev)
+        tok_hist = torch.full((max_new + 2,), -1, dtype=torch.int32, device=dev)
+        tok_slot = torch.zeros(1, dtype=torch.int32, device=dev)
+        logits_buf = logits.reshape(-1).contiguous().clone()
+
+        def step():
+            hip.sample_state(logits_buf, temperature, rng_state, tok_hist,
+                             step_state, tok_slot)
+            lg = self.model.decode_one_graph(tok_slot, cache, pos_state, max_total)
+            logits_buf.copy_(lg.reshape(-1))
+            hip.bump(pos_state, step_state)
+
+        # warm up on a side stream (torch graph-capture requirement); these
+        # are REAL decode steps — their tokens land in tok_hist[0..1].
+        warm = min(2, max_new)
+        s = torch.cuda.Stream(device=dev)
+        s.wait_stream(torch.cuda.current_stream(dev))
+        with torch.cuda.stream(s):
+            for _ in range(warm):
+                step()
+        torch.cuda.current_stream(dev).wait_stream(s)
+
+        graph = torch.cuda.CUDAGraph()
+        with torch.cuda.graph(graph):
+            step()
+
+        CHECK = 32
+        done = False
+        i = warm
+        n_keep = None
+        out_ids: list[int] = []
+        tail = ""
+        scanned = 0
+
+        def scan_until(upto: int) -> bool:
+            """Pull tokens [scanned, upto) to host, extend out_ids; True if a
+            stop condition fired."""
+            nonlocal scanned, tail
+            chunk = tok_hist[scanned:upto].cpu().tolist()
+            scanned = upto
+            for tok in chunk:
+                if tok in stop_ids or tok < 0:
+                    return True
+                out_ids.append(tok)
+                if 0 <= tok < 256:
+                    tail = (tail + chr(tok))[-16:]
+                    if tail.endswith(_STOP_SUBSTR):
+                        return True
+            return False
+
+        while i < max_new and not done:
+            n = min(CHECK, max_new - i)
+            for _ in range(n):
+                graph.replay()
+            i += n
+            done = scan_until(i)
+            if time.monotonic() > deadline:
+                done = True
+        if not done:
+            scan_until(i)
+
+        cache.seq_len = prompt_len + i  # device-side pos advanced i times
         return out_ids
 
     def _decode_async(self, logits, cache, max_new, temperature, stop_ids,

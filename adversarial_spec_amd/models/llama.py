@@ -233,8 +233,16 @@ class LlamaModel:
             self.config, max_seq or self.config.max_seq_len, self.device, self.dtype
         )
 
-    def _forward(self, tokens: torch.Tensor, cache: PagedKVCache, pos0: int) -> torch.Tensor:
-        """Run t tokens at positions pos0..pos0+t-1; return last-token logits."""
+    def _forward(self, tokens: torch.Tensor, cache: PagedKVCache, pos0: int,
+                 pos_state: Optional[torch.Tensor] = None,
+                 max_seq_bound: int = 0) -> torch.Tensor:
+        """Run t tokens at positions pos0..pos0+t-1; return last-token logits.
+
+        Graph mode (t==1, GPU): pos_state is a device int32[1] holding the
+        position — rope/kv_write/attention read it in-kernel so one captured
+        HIP graph replays for every decode token; max_seq_bound sizes the
+        attention split geometry once for the whole generation.
+        """
         c = self.config
         t = tokens.shape[0]
         h, kh, hd = c.n_heads, c.n_kv_heads, c.head_dim
@@ -250,15 +258,18 @@ class LlamaModel:
             q = qkv[:, : h * hd].view(t, h, hd)
             k = qkv[:, h * hd : (h + kh) * hd].view(t, kh, hd)
             v = qkv[:, (h + kh) * hd :].view(t, kh, hd)
-            q, k = ops.rope(q, k, self.cos, self.sin, pos0)
-            ops.kv_write(cache.k[i], cache.v[i], cache.page_table, pos0, k, v)
+            q, k = ops.rope(q, k, self.cos, self.sin, pos0, pos_state=pos_state)
+            ops.kv_write(cache.k[i], cache.v[i], cache.page_table, pos0, k, v,
+                         pos_state=pos_state)
             if t > 1:
                 if pos0 != 0:
                     raise NotImplementedError("chunked prefill lands with the 32k path")
                 attn = ops.attn_prefill(q, k, v, self.scale, causal=True)
             else:
+                seq = max_seq_bound if pos_state is not None else pos0 + 1
                 attn = ops.attn_decode_paged(
-                    q[0], cache.k[i], cache.v[i], cache.page_table, pos0 + 1, self.scale
+                    q[0], cache.k[i], cache.v[i], cache.page_table, seq,
+                    self.scale, pos_state=pos_state,
                 ).unsqueeze(0)
             attn_out = mm(attn.reshape(t, h * hd), L.wo)
             if self.tp is not None and self.tp.size > 1:
@@ -297,3 +308,12 @@ class LlamaModel:
         logits = self._forward(tok, cache, cache.seq_len)
         cache.seq_len += 1
         return logits
+
+    def decode_one_graph(self, tok_slot: torch.Tensor, cache: PagedKVCache,
+                         pos_state: torch.Tensor, max_seq_bound: int) -> torch.Tensor:
+        """One graph-capturable decode step: position from pos_state, token
+        from tok_slot (int32[1] on device). Does NOT advance cache.seq_len —
+        the engine reconciles it after the replay loop."""
+        tok = tok_slot.to(torch.long)
+        return self._forward(tok, cache, 0, pos_state=pos_state,
+                             max_seq_bound=max_seq_bound)

@@ -74,10 +74,14 @@ def add_rmsnorm(
 
 
 def rope(
-    q: torch.Tensor, k: torch.Tensor, cos: torch.Tensor, sin: torch.Tensor, pos0: int
+    q: torch.Tensor, k: torch.Tensor, cos: torch.Tensor, sin: torch.Tensor, pos0: int,
+    pos_state: Optional[torch.Tensor] = None,
 ) -> Tuple[torch.Tensor, torch.Tensor]:
     if _on_gpu(q):
-        _require_hip().rope_inplace(q, k, cos, sin, pos0)
+        if pos_state is not None:
+            _require_hip().rope_inplace_ds(q, k, cos, sin, pos_state)
+        else:
+            _require_hip().rope_inplace(q, k, cos, sin, pos0)
         return q, k
     return torch_ref.rope(q, k, cos, sin, pos0)
 
@@ -111,11 +115,18 @@ def attn_decode_paged(
     page_table: torch.Tensor,
     seq_len: int,
     scale: Optional[float] = None,
+    pos_state: Optional[torch.Tensor] = None,
 ) -> torch.Tensor:
     if _on_gpu(q):
         import math
 
         s = scale if scale is not None else 1.0 / math.sqrt(q.shape[-1])
+        if pos_state is not None:
+            # graph mode: true length = *pos_state + 1 in-kernel; seq_len is
+            # the static max bound sizing the split geometry.
+            return _require_hip().attn_decode_paged_ds(
+                q, k_cache, v_cache, page_table, pos_state, seq_len, s
+            )
         return _require_hip().attn_decode_paged(q, k_cache, v_cache, page_table, seq_len, s)
     return torch_ref.attn_decode_paged(q, k_cache, v_cache, page_table, seq_len, scale)
 
@@ -127,9 +138,13 @@ def kv_write(
     pos0: int,
     k: torch.Tensor,
     v: torch.Tensor,
+    pos_state: Optional[torch.Tensor] = None,
 ) -> None:
     if _on_gpu(k_cache):
-        _require_hip().kv_write(k_cache, v_cache, page_table, pos0, k, v)
+        if pos_state is not None:
+            _require_hip().kv_write_ds(k_cache, v_cache, page_table, pos_state, k, v)
+        else:
+            _require_hip().kv_write(k_cache, v_cache, page_table, pos0, k, v)
         return
     torch_ref.kv_write(k_cache, v_cache, page_table, pos0, k, v)
 

@@ -48,7 +48,10 @@ __global__ void __launch_bounds__(256) attn_decode_split_kernel(
     const int *__restrict__ page_table, int seq_len, float scale,
     int kh, int group, int hd, int page, int split_len,
     float *__restrict__ ws_m, float *__restrict__ ws_l,
-    float *__restrict__ ws_acc) {
+    float *__restrict__ ws_acc, const int *__restrict__ pos_ptr) {
+  // graph mode: seq_len = *pos_ptr + 1 (attend up to and incl. the token
+  // kv_write just appended at position *pos_ptr)
+  if (pos_ptr) seq_len = *pos_ptr + 1;
   const int g = blockIdx.x;          // kv head
   const int split = blockIdx.y;
   const int n_splits = gridDim.y;
@@ -298,7 +301,7 @@ extern "C" void launch_attn_decode_split(
     const ushort_t *q, const ushort_t *kc, const ushort_t *vc,
     const int *page_table, int seq_len, float scale, int kh, int group,
     int hd, int page, int split_len, int n_splits, float *ws_m, float *ws_l,
-    float *ws_acc, hipStream_t stream) {
+    float *ws_acc, const int *pos_ptr, hipStream_t stream) {
   dim3 grid(kh, n_splits);
   const int lds = (4 * MAXG * 2 + 4 * MAXG * hd) * sizeof(float);
   // MG = smallest supported bound >= group keeps the per-head state arrays
@@ -309,15 +312,15 @@ extern "C" void launch_attn_decode_split(
     if (group <= 2)                                                            \
       attn_decode_split_kernel<LPP, 2><<<grid, 256, lds, stream>>>(            \
           q, kc, vc, page_table, seq_len, scale, kh, group, hd, page,          \
-          split_len, ws_m, ws_l, ws_acc);                                      \
+          split_len, ws_m, ws_l, ws_acc, pos_ptr);                             \
     else if (group <= 4)                                                       \
       attn_decode_split_kernel<LPP, 4><<<grid, 256, lds, stream>>>(            \
           q, kc, vc, page_table, seq_len, scale, kh, group, hd, page,          \
-          split_len, ws_m, ws_l, ws_acc);                                      \
+          split_len, ws_m, ws_l, ws_acc, pos_ptr);                             \
     else                                                                       \
       attn_decode_split_kernel<LPP, 8><<<grid, 256, lds, stream>>>(            \
           q, kc, vc, page_table, seq_len, scale, kh, group, hd, page,          \
-          split_len, ws_m, ws_l, ws_acc);                                      \
+          split_len, ws_m, ws_l, ws_acc, pos_ptr);                             \
   } while (0)
 
   switch (hd / 8) {

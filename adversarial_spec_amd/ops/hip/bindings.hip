@@ -13,7 +13,7 @@ using ushort_t = unsigned short;
 extern "C" {
 void launch_attn_decode_split(const ushort_t*, const ushort_t*, const ushort_t*,
                               const int*, int, float, int, int, int, int, int,
-                              int, float*, float*, float*, hipStream_t);
+                              int, float*, float*, float*, const int*, hipStream_t);
 void launch_attn_decode_combine(const float*, const float*, const float*,
                                 ushort_t*, int, int, int, int, hipStream_t);
 void launch_attn_prefill_simple(const ushort_t*, const ushort_t*, const ushort_t*,
@@ -27,11 +27,14 @@ __global__ void rmsnorm_kernel(const ushort_t*, const ushort_t*, ushort_t*, int,
 __global__ void add_rmsnorm_kernel(const ushort_t*, const ushort_t*, const ushort_t*,
                                    ushort_t*, ushort_t*, int, float);
 __global__ void rope_kernel(ushort_t*, ushort_t*, const float*, const float*,
-                            int, int, int, int, int, long, long);
+                            int, int, int, int, int, long, long, const int*);
 __global__ void swiglu_kernel(const ushort_t*, const ushort_t*, ushort_t*, int, int, int);
 __global__ void kv_write_kernel(const ushort_t*, const ushort_t*, ushort_t*, ushort_t*,
-                                const int*, int, int, int, int, int);
+                                const int*, int, int, int, int, int, const int*);
 __global__ void sample_kernel(const ushort_t*, int, float, uint32_t, int*);
+__global__ void sample_state_kernel(const ushort_t*, int, float, uint32_t*, int*,
+                                    const int*, int*);
+__global__ void bump_kernel(int*, int*);
 void launch_gemv(const ushort_t*, const ushort_t*, float*, ushort_t*, int, int,
                  int, hipStream_t);
 }
@@ -104,7 +107,20 @@ void rope_inplace(torch::Tensor q, torch::Tensor k, torch::Tensor cost,
   const int blocks = (waves * 64 + 255) / 256;
   rope_kernel<<<blocks, 256, 0, cur_stream()>>>(
       uptr_mut(q), uptr_mut(k), cost.data_ptr<float>(), sint.data_ptr<float>(),
-      t, hq, hk, hd, (int)pos0, q.stride(0), k.stride(0));
+      t, hq, hk, hd, (int)pos0, q.stride(0), k.stride(0), nullptr);
+}
+
+// graph-mode RoPE: position read from a device int32 word
+void rope_inplace_ds(torch::Tensor q, torch::Tensor k, torch::Tensor cost,
+                     torch::Tensor sint, torch::Tensor pos_state) {
+  CHECK_BF16_CUDA(q);
+  const int t = q.size(0), hq = q.size(1), hd = q.size(2);
+  const int hk = k.size(1);
+  const int waves = t * (hq + hk);
+  const int blocks = (waves * 64 + 255) / 256;
+  rope_kernel<<<blocks, 256, 0, cur_stream()>>>(
+      uptr_mut(q), uptr_mut(k), cost.data_ptr<float>(), sint.data_ptr<float>(),
+      t, hq, hk, hd, 0, q.stride(0), k.stride(0), pos_state.data_ptr<int>());
 }
 
 torch::Tensor swiglu(torch::Tensor gate, torch::Tensor up) {
@@ -139,7 +155,19 @@ void kv_write(torch::Tensor kc, torch::Tensor vc, torch::Tensor page_table,
   TORCH_CHECK((kh * hd) % 8 == 0);
   kv_write_kernel<<<t, 256, 0, cur_stream()>>>(
       uptr(kq), uptr(vq), uptr_mut(kcc), uptr_mut(vc),
-      page_table.data_ptr<int>(), (int)pos0, t, kh, hd, page);
+      page_table.data_ptr<int>(), (int)pos0, t, kh, hd, page, nullptr);
+}
+
+void kv_write_ds(torch::Tensor kc, torch::Tensor vc, torch::Tensor page_table,
+                 torch::Tensor pos_state, torch::Tensor k, torch::Tensor v) {
+  auto kq = k.contiguous();
+  auto vq = v.contiguous();
+  const int t = kq.size(0), kh = kq.size(1), hd = kq.size(2);
+  const int page = kc.size(1);
+  kv_write_kernel<<<t, 256, 0, cur_stream()>>>(
+      uptr(kq), uptr(vq), uptr_mut(kc), uptr_mut(vc),
+      page_table.data_ptr<int>(), 0, t, kh, hd, page,
+      pos_state.data_ptr<int>());
 }
 
 torch::Tensor attn_prefill(torch::Tensor q, torch::Tensor k, torch::Tensor v,
@@ -235,7 +263,7 @@ torch::Tensor attn_decode_paged(torch::Tensor q, torch::Tensor kc,
                            (float)scale, kh, group, hd, page, split_len,
                            n_splits, ws_m.data_ptr<float>(),
                            ws_l.data_ptr<float>(), ws_acc.data_ptr<float>(),
-                           cur_stream());
+                           nullptr, cur_stream());
   launch_attn_decode_combine(ws_m.data_ptr<float>(), ws_l.data_ptr<float>(),
                              ws_acc.data_ptr<float>(), uptr_mut(out), hq,
                              n_splits, group, hd, cur_stream());
@@ -292,6 +320,55 @@ int64_t sample(torch::Tensor logits, double temp, double top_p, int64_t seed) {
   return out.cpu().item<int>();
 }
 
+// graph-mode decode attention: seq_len = *pos_state + 1 in-kernel; split
+// count is sized once for the WHOLE generation (max_seq bound) so the
+// launch geometry is replay-stable.
+torch::Tensor attn_decode_paged_ds(torch::Tensor q, torch::Tensor kc,
+                                   torch::Tensor vc, torch::Tensor page_table,
+                                   torch::Tensor pos_state, int64_t max_seq,
+                                   double scale) {
+  CHECK_BF16_CUDA(q);
+  auto qc = q.contiguous();
+  const int hq = qc.size(0), hd = qc.size(1);
+  const int page = kc.size(1), kh = kc.size(2);
+  const int group = hq / kh;
+  int n_splits = std::max(1, std::min((int)((max_seq + 63) / 64), 1024 / kh));
+  int split_len = (int)((max_seq + n_splits - 1) / n_splits + 63) / 64 * 64;
+  n_splits = (int)((max_seq + split_len - 1) / split_len);
+  auto wopt = torch::TensorOptions().dtype(at::kFloat).device(q.device());
+  auto ws_m = torch::empty({(long)kh * n_splits * group}, wopt);
+  auto ws_l = torch::empty({(long)kh * n_splits * group}, wopt);
+  auto ws_acc = torch::empty({(long)kh * n_splits * group * hd}, wopt);
+  auto out = torch::empty({hq, hd}, qc.options());
+  launch_attn_decode_split(uptr(qc), uptr(kc), uptr(vc),
+                           page_table.data_ptr<int>(), (int)max_seq,
+                           (float)scale, kh, group, hd, page, split_len,
+                           n_splits, ws_m.data_ptr<float>(),
+                           ws_l.data_ptr<float>(), ws_acc.data_ptr<float>(),
+                           pos_state.data_ptr<int>(), cur_stream());
+  launch_attn_decode_combine(ws_m.data_ptr<float>(), ws_l.data_ptr<float>(),
+                             ws_acc.data_ptr<float>(), uptr_mut(out), hq,
+                             n_splits, group, hd, cur_stream());
+  return out;
+}
+
+void sample_state(torch::Tensor logits, double temp, torch::Tensor rng_state,
+                  torch::Tensor tok_hist, torch::Tensor step_state,
+                  torch::Tensor tok_slot) {
+  CHECK_BF16_CUDA(logits);
+  auto lc = logits.contiguous();
+  sample_state_kernel<<<1, 256, 0, cur_stream()>>>(
+      uptr(lc), (int)lc.numel(), (float)temp,
+      reinterpret_cast<uint32_t*>(rng_state.data_ptr<int>()),
+      tok_hist.data_ptr<int>(), step_state.data_ptr<int>(),
+      tok_slot.data_ptr<int>());
+}
+
+void bump(torch::Tensor pos_state, torch::Tensor step_state) {
+  bump_kernel<<<1, 64, 0, cur_stream()>>>(pos_state.data_ptr<int>(),
+                                          step_state.data_ptr<int>());
+}
+
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("rmsnorm", &rmsnorm, "fused RMSNorm (bf16, gfx950)");
   m.def("add_rmsnorm", &add_rmsnorm, "fused residual add + RMSNorm");
@@ -305,4 +382,10 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("sample", &sample, "fused temperature softmax sample");
   m.def("sample_to", &sample_to, "async on-device sample into out[idx]");
   m.def("gemv", &gemv, "batch-1 decode GEMV (weight streaming)");
+  m.def("rope_inplace_ds", &rope_inplace_ds, "graph-mode RoPE (device pos)");
+  m.def("kv_write_ds", &kv_write_ds, "graph-mode KV scatter (device pos)");
+  m.def("attn_decode_paged_ds", &attn_decode_paged_ds,
+        "graph-mode paged decode attention (device pos)");
+  m.def("sample_state", &sample_state, "graph-mode on-device sampling");
+  m.def("bump", &bump, "graph-mode pos/step bump");
 }

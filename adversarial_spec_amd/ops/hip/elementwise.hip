@@ -102,7 +102,11 @@ extern "C" __global__ void __launch_bounds__(256)
 rope_kernel(ushort_t *__restrict__ q, ushort_t *__restrict__ k,
             const float *__restrict__ cost, const float *__restrict__ sint,
             int t, int hq, int hk, int hd, int pos0,
-            long q_rstride, long k_rstride) {
+            long q_rstride, long k_rstride,
+            const int *__restrict__ pos_ptr) {
+  // HIP-graph decode: the position lives in a device word so one captured
+  // graph replays for every token (pos_ptr == nullptr -> use pos0).
+  if (pos_ptr) pos0 = *pos_ptr;
   // q/k may be strided views into the fused QKV GEMM output (row strides in
   // elements); head/dim dims are contiguous.
   const int wave_id = (blockIdx.x * blockDim.x + threadIdx.x) / WAVE;
@@ -173,7 +177,9 @@ extern "C" __global__ void __launch_bounds__(256)
 kv_write_kernel(const ushort_t *__restrict__ k, const ushort_t *__restrict__ v,
                 ushort_t *__restrict__ kc, ushort_t *__restrict__ vc,
                 const int *__restrict__ page_table, int pos0, int t,
-                int kh, int hd, int page) {
+                int kh, int hd, int page,
+                const int *__restrict__ pos_ptr) {
+  if (pos_ptr) pos0 = *pos_ptr;
   const int tok = blockIdx.x;
   if (tok >= t) return;
   const int pos = pos0 + tok;

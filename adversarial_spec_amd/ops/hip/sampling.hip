@@ -17,10 +17,10 @@
 
 #include "common.h"
 
-// logits: [vocab] bf16, vocab % 8 == 0. out: int32. temp <= 0 -> greedy.
-extern "C" __global__ void __launch_bounds__(256)
-sample_kernel(const ushort_t *__restrict__ logits, int vocab, float temp,
-              uint32_t seed, int *__restrict__ out) {
+// logits: [vocab] bf16, vocab % 8 == 0. temp <= 0 -> greedy. Returns the
+// sampled id in *out; shared body for the plain and graph-state kernels.
+DEVINL void sample_body(const ushort_t *__restrict__ logits, int vocab,
+                        float temp, uint32_t seed, int *__restrict__ out) {
   __shared__ float scratch[16];
   __shared__ float tsum[256];
   __shared__ int result;
@@ -132,4 +132,43 @@ sample_kernel(const ushort_t *__restrict__ logits, int vocab, float temp,
   }
   __syncthreads();
   if (tid == 0) *out = result;
+}
+
+extern "C" __global__ void __launch_bounds__(256)
+sample_kernel(const ushort_t *__restrict__ logits, int vocab, float temp,
+              uint32_t seed, int *__restrict__ out) {
+  sample_body(logits, vocab, temp, seed, out);
+}
+
+// HIP-graph decode variant: the RNG seed, step index and token outputs all
+// live in device words so one captured graph replays per token.
+//   rng_state:  uint32[1], advanced each call
+//   step_state: int32[1], the decode step index (bumped by bump_kernel)
+//   tok_hist:   int32[max_new] history the host polls every N tokens
+//   tok_slot:   int32[1] fixed slot feeding the next embedding lookup
+extern "C" __global__ void __launch_bounds__(256)
+sample_state_kernel(const ushort_t *__restrict__ logits, int vocab,
+                    float temp, uint32_t *__restrict__ rng_state,
+                    int *__restrict__ tok_hist,
+                    const int *__restrict__ step_state,
+                    int *__restrict__ tok_slot) {
+  __shared__ int picked[1];
+  const uint32_t seed = *rng_state;
+  sample_body(logits, vocab, temp, seed, picked);
+  __syncthreads();
+  if (threadIdx.x == 0) {
+    const int tok = picked[0];
+    tok_slot[0] = tok;
+    tok_hist[*step_state] = tok;
+    *rng_state = hash_u32(seed ^ 0x6a09e667u) | 1u;  // never 0
+  }
+}
+
+// pos/step bump — the single tail kernel of the captured decode step.
+extern "C" __global__ void bump_kernel(int *__restrict__ pos_state,
+                                       int *__restrict__ step_state) {
+  if (threadIdx.x == 0 && blockIdx.x == 0) {
+    pos_state[0] += 1;
+    step_state[0] += 1;
+  }
 }

@@ -86,3 +86,16 @@ class TestGenerationGPU:
         a = eng.generate("s", "u", max_tokens=12, temperature=0.0, timeout=300)
         b = eng.generate("s", "u", max_tokens=12, temperature=0.0, timeout=300)
         assert a[0] == b[0]
+
+    def test_graph_decode_matches_eager_greedy(self, monkeypatch):
+        """HIP-graph decode must emit the same greedy tokens as the eager
+        async loop (same kernels, same state)."""
+        eng = LocalEngine({"name": "g3", "arch": "debug-1b"}, device=DEV)
+        monkeypatch.setenv("ADVSPEC_NO_GRAPH", "1")
+        a = eng.generate("sys", "graph parity prompt", max_tokens=24,
+                         temperature=0.0, timeout=300)
+        monkeypatch.delenv("ADVSPEC_NO_GRAPH")
+        b = eng.generate("sys", "graph parity prompt", max_tokens=24,
+                         temperature=0.0, timeout=300)
+        assert a[0] == b[0], (a[0][:80], b[0][:80])
+        assert b[2] > 0
