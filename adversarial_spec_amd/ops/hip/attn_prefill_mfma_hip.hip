@@ -3,15 +3,21 @@
 // fp32 accumulation. This is the hot prefill kernel for hd=128 models
 // (Llama-3 family); hd∈{32,64} falls back to attention.hip's simple kernel.
 //
-// Structure (cdna_hip_programming.md §5/§B: LDS-staged K/V tiles, online
-// softmax, MFMA 16x16x32):
+// Structure (cdna_hip_programming.md §5/§B):
 //   grid  = (hq, ceil(tq/128));  block = 512 threads (8 waves)
-//   each wave owns a 16-row Q tile; the block shares K/V tiles of 32 keys
-//   staged in LDS (double-barrier loop, v1).
+//   each wave owns a 16-row Q tile; the block shares LDS-staged K/V tiles
+//   of 32 keys.
 //
-// MFMA fragment maps used (verified on hardware by tests/test_ops_gpu.py's
-// mfma probe — the C/D map is documented in the CDNA4 guide §3; A/B are the
-// standard CDNA maps with K doubled for gfx950):
+// LDS images (all carved from ONE array — compiler trap §5 4a):
+//   K  [32][128] bf16, XOR-swizzled byte^=((key&7)<<4): a 256-B row puts a
+//      column read's whole 16-lane group on one bank without it
+//      (Guideline 4's D=128 hazard — 16-way, measured 5.1 ms/layer).
+//   Vt [128][40] bf16: V TRANSPOSED with rows padded 32->40 elements so the
+//      PV B-fragment is one ds_read_b128 per lane (row stride 80 B makes
+//      the 16 consecutive-row banks distinct) instead of 8 scalar reads.
+//   P  [16][40] bf16 per wave (C->A layout bounce), same 80-B row pitch.
+//
+// MFMA fragment maps (hardware-verified by tests/test_ops_gpu.py mfma probe):
 //   mfma_f32_16x16x32_bf16:
 //     A: lane l holds A[row = l%16][k = (l/16)*8 + j], j = 0..7
 //     B: lane l holds B[k = (l/16)*8 + j][col = l%16]
@@ -26,6 +32,7 @@ typedef __attribute__((__vector_size__(4 * sizeof(float)))) float f32x4v;
 #define KVBLK 32   // keys per staged tile
 #define NWAVE 8    // waves per block
 #define HD 128     // head dim (this kernel is hd=128 only)
+#define VPITCH 40  // padded row length (elements) of the Vt and P images
 
 extern "C" __global__ void __launch_bounds__(512, 1)
 attn_prefill_mfma_kernel(const ushort_t *__restrict__ q,
@@ -44,13 +51,11 @@ attn_prefill_mfma_kernel(const ushort_t *__restrict__ q,
   const int qb0 = qblock * (NWAVE * QROWS);
   const int q0 = qb0 + wid * QROWS; // this wave's first q row
 
-  // LDS: K tile + V tile (bf16 [KVBLK][HD]), plus per-wave P tiles
-  // ([QROWS][KVBLK] bf16). Single __shared__ object (compiler trap #4a).
   __shared__ __attribute__((aligned(16))) ushort_t lds[
-      2 * KVBLK * HD + NWAVE * QROWS * KVBLK];
-  ushort_t *ldsK = lds;
-  ushort_t *ldsV = lds + KVBLK * HD;
-  ushort_t *ldsP = lds + 2 * KVBLK * HD + wid * QROWS * KVBLK;
+      KVBLK * HD + HD * VPITCH + NWAVE * QROWS * VPITCH];
+  ushort_t *ldsK = lds;                       // swizzled [32][128]
+  ushort_t *ldsVt = lds + KVBLK * HD;         // [128][40]
+  ushort_t *ldsP = ldsVt + HD * VPITCH + wid * QROWS * VPITCH;
 
   const int lrow = lane & 15;       // 0..15
   const int lhi = lane >> 4;        // 0..3
@@ -83,17 +88,15 @@ attn_prefill_mfma_kernel(const ushort_t *__restrict__ q,
 #pragma unroll
   for (int d = 0; d < 8; ++d) o[d] = (f32x4v){0.f, 0.f, 0.f, 0.f};
 
-  // causal upper bound on keys for this block (kv-position space)
   int kmax = tk;
   if (causal) kmax = min(tk, kv_offset + qb0 + NWAVE * QROWS);
 
   for (int kt = 0; kt < kmax; kt += KVBLK) {
-    // ---- stage K/V tile (coalesced 16B per thread) ----
+    // ---- stage K (swizzled) and Vt (transposed, padded) ----
     {
-      // KVBLK*HD bf16 = 8 KB = 512 * 16 B
-      const int flat = tid;  // one 16B piece each
-      const int krow = flat / (HD / 8);
-      const int kcol8 = flat % (HD / 8);
+      const int flat = tid;             // one 16B K piece each
+      const int krow = flat >> 4;       // key within tile
+      const int kcol8 = flat & 15;      // 16B chunk within the 256B row
       const int key = kt + krow;
       bf16x8 kk, vv;
       if (key < tk) {
@@ -104,8 +107,18 @@ attn_prefill_mfma_kernel(const ushort_t *__restrict__ q,
         for (int j = 0; j < 8; ++j) { kk.u[j] = 0; vv.u[j] = 0; }
       }
       __syncthreads();  // previous iteration's reads done
-      ((bf16x8 *)ldsK)[flat] = kk;
-      ((bf16x8 *)ldsV)[flat] = vv;
+      {
+        // K: swizzled 16B store
+        unsigned kbyte = (unsigned)flat * 16u;
+        kbyte ^= ((unsigned)(krow & 7)) << 4;
+        *(bf16x8 *)((char *)ldsK + kbyte) = kk;
+        // Vt: 8 scalar transposed stores (staging-only cost)
+        const int d0 = kcol8 * 8;
+#pragma unroll
+        for (int j = 0; j < 8; ++j) {
+          ldsVt[(size_t)(d0 + j) * VPITCH + krow] = vv.u[j];
+        }
+      }
       __syncthreads();
     }
 
@@ -114,23 +127,24 @@ attn_prefill_mfma_kernel(const ushort_t *__restrict__ q,
 #pragma unroll
     for (int n = 0; n < 2; ++n) {
       s[n] = (f32x4v){0.f, 0.f, 0.f, 0.f};
-      // B frag: B[kd][col] = K[n*16+col][c*32+kd8]
       const int keyr = n * 16 + lrow;
 #pragma unroll
       for (int c = 0; c < 4; ++c) {
+        // B frag: B[kd][col] = K[keyr][c*32+kd8] from the swizzled image
+        unsigned kbyte = (unsigned)keyr * 256u + (unsigned)(c * 4 + lhi) * 16u;
+        kbyte ^= ((unsigned)(keyr & 7)) << 4;
+        const bf16x8 t = *(const bf16x8 *)((const char *)ldsK + kbyte);
         bf16x8v bfr;
-        const bf16x8 t = ((const bf16x8 *)(ldsK + keyr * HD))[c * 4 + lhi];
 #pragma unroll
         for (int j = 0; j < 8; ++j) bfr[j] = (short)t.u[j];
         s[n] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(qfrag[c], bfr, s[n], 0, 0, 0);
       }
     }
 
-    // ---- online softmax update (4 q rows per lane: row=(lhi)*4+r) ----
+    // ---- online softmax update (4 q rows per lane: row=lhi*4+r) ----
     float rmax[4];
 #pragma unroll
     for (int r = 0; r < 4; ++r) {
-      // scale + causal mask
       const int qrow_abs = kv_offset + q0 + lhi * 4 + r;
 #pragma unroll
       for (int n = 0; n < 2; ++n) {
@@ -163,36 +177,35 @@ attn_prefill_mfma_kernel(const ushort_t *__restrict__ q,
         psum += __shfl_xor(psum, off, WAVE);
       lsum[r] = lsum[r] * alpha + psum;
       m[r] = mn;
-      // rescale O rows
 #pragma unroll
       for (int d = 0; d < 8; ++d) o[d][r] *= alpha;
     }
 
-    // ---- P -> LDS (C layout -> A layout bounce) ----
+    // ---- P -> LDS (C layout -> A layout bounce; padded pitch) ----
 #pragma unroll
     for (int n = 0; n < 2; ++n) {
 #pragma unroll
       for (int r = 0; r < 4; ++r) {
-        ldsP[(lhi * 4 + r) * KVBLK + n * 16 + lrow] = f32_to_bf16(s[n][r]);
+        ldsP[(lhi * 4 + r) * VPITCH + n * 16 + lrow] = f32_to_bf16(s[n][r]);
       }
     }
-    // wave-local LDS dependency; compiler inserts lgkmcnt before reads.
 
     // ---- O += P V ----
     bf16x8v pfrag;
     {
-      const bf16x8 t = *(const bf16x8 *)(ldsP + lrow * KVBLK + lhi * 8);
+      const bf16x8 t = *(const bf16x8 *)(ldsP + lrow * VPITCH + lhi * 8);
 #pragma unroll
       for (int j = 0; j < 8; ++j) pfrag[j] = (short)t.u[j];
     }
 #pragma unroll
     for (int d = 0; d < 8; ++d) {
-      // B frag: B[kd][col] = V[kd8][d*16+col] (strided LDS read, v1)
+      // B frag: B[kd][col] = V[kd8][d*16+col] = Vt[d*16+col][kd8] — one
+      // contiguous 16B LDS read per lane.
+      const bf16x8 t =
+          *(const bf16x8 *)(ldsVt + (size_t)(d * 16 + lrow) * VPITCH + lhi * 8);
       bf16x8v vfr;
 #pragma unroll
-      for (int j = 0; j < 8; ++j) {
-        vfr[j] = (short)ldsV[(lhi * 8 + j) * HD + d * 16 + lrow];
-      }
+      for (int j = 0; j < 8; ++j) vfr[j] = (short)t.u[j];
       o[d] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(pfrag, vfr, o[d], 0, 0, 0);
     }
   }
