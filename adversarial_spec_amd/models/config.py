@@ -49,7 +49,7 @@ class LlamaConfig:
 PRESETS: dict[str, LlamaConfig] = {
     "llama-3-8b": LlamaConfig(
         name="llama-3-8b", dim=4096, n_layers=32, n_heads=32, n_kv_heads=8,
-        ffn_dim=14336, vocab_size=128256, max_seq_len=16384, rope_theta=500000.0,
+        ffn_dim=14336, vocab_size=128256, max_seq_len=32768, rope_theta=500000.0,
     ),
     "llama-3-70b": LlamaConfig(
         name="llama-3-70b", dim=8192, n_layers=80, n_heads=64, n_kv_heads=8,
@@ -57,7 +57,7 @@ PRESETS: dict[str, LlamaConfig] = {
     ),
     "mistral-7b": LlamaConfig(
         name="mistral-7b", dim=4096, n_layers=32, n_heads=32, n_kv_heads=8,
-        ffn_dim=14336, vocab_size=32000, max_seq_len=16384, rope_theta=10000.0,
+        ffn_dim=14336, vocab_size=32000, max_seq_len=32768, rope_theta=10000.0,
     ),
     # CPU-testable / smoke model: byte tokenizer fits in 1024 vocab.
     "tiny": LlamaConfig(

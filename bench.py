@@ -55,6 +55,9 @@ def parse_args():
     p.add_argument("--decode-tokens", type=int, default=128,
                    help="critique length decoded per opponent per round")
     p.add_argument("--model", default="llama-3-8b")
+    p.add_argument("--arch-mix", default=None,
+                   help="comma list of archs cycled over opponents for the "
+                        "heterogeneous config (e.g. llama-3-8b,mistral-7b)")
     p.add_argument("--temperature", type=float, default=0.7)
     p.add_argument("--tp", type=int, default=1,
                    help="tensor-parallel degree: all ranks form ONE sharded "
@@ -98,12 +101,14 @@ def main() -> int:
         ]
     else:
         n_opp = args.opponents_per_gpu
+        archs = (args.arch_mix.split(",") if args.arch_mix else [args.model])
         engines = []
         for i in range(n_opp):
             # distinct seeds per (rank, opponent): heterogeneous random-init pool
+            arch = archs[(rank * n_opp + i) % len(archs)]
             engines.append(
                 LocalEngine(
-                    {"name": f"{args.model}-r{rank}o{i}", "arch": args.model},
+                    {"name": f"{arch}-r{rank}o{i}", "arch": arch},
                     device=device,
                 )
             )
