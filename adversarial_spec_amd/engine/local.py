@@ -28,6 +28,10 @@ from .tokenizer import build_tokenizer
 # complete by protocol.
 _STOP_SUBSTR = "[/SPEC]"
 
+# Serializes HIP-graph captures across opponent threads (capture itself is
+# rare — once per engine — but must not run concurrently with another capture).
+_CAPTURE_LOCK = threading.Lock()
+
 
 def _seed_from_name(name: str) -> int:
     h = 2166136261
@@ -37,7 +41,8 @@ def _seed_from_name(name: str) -> int:
 
 
 class LocalEngine:
-    def __init__(self, spec: dict[str, Any], device: Optional[str] = None) -> None:
+    def __init__(self, spec: dict[str, Any], device: Optional[str] = None,
+                 tp=None) -> None:
         self.name = spec.get("name", spec["arch"])
         self.arch = spec["arch"]
         self.config = get_config(self.arch)
@@ -49,7 +54,16 @@ class LocalEngine:
                 device = "cpu"
         self.device = torch.device(device)
         self.model = LlamaModel(
-            self.config, device=self.device, seed=_seed_from_name(self.name)
+            self.config, device=self.device, seed=_seed_from_name(self.name),
+            tp=tp,
+        )
+        # Per-engine HIP stream: co-resident opponents on one GPU issue all
+        # their work (prefill GEMMs, decode GEMV/attention, graph replays)
+        # on their own stream, so concurrent rounds of N opponents overlap
+        # on-device instead of serializing through the default stream.
+        self.stream = (
+            torch.cuda.Stream(device=self.device)
+            if self.device.type == "cuda" else None
         )
         weights = spec.get("weights")
         if weights:
@@ -61,6 +75,8 @@ class LocalEngine:
         self._gen_lock = threading.Lock()
         self._sample_gen = torch.Generator().manual_seed(_seed_from_name(self.name) ^ 0x5EED)
         self._seed_counter = _seed_from_name(self.name) ^ 0x5EED
+        self._cache = None  # persistent KV cache (see _get_cache)
+        self._graph_state: Optional[dict] = None  # captured decode graph + state
 
     def _fit_prompt(self, ids: list[int], reserve: int) -> list[int]:
         """Clamp a prompt into the context window, dropping the middle."""
@@ -101,34 +117,61 @@ class LocalEngine:
             prev_device = torch.cuda.current_device()
             torch.cuda.set_device(self.device)
         try:
-            cache = self.model.new_cache(len(ids) + max_new + 8)
-            tokens = torch.tensor(ids, device=self.device, dtype=torch.long)
-            with timer.phase("prefill"):
-                logits = self.model.prefill(tokens, cache)
+            import contextlib
 
-            out_ids: list[int] = []
-            with timer.phase("decode"):
-                use_async = (
-                    self.device.type == "cuda"
-                    and top_p >= 1.0
-                    and ops.hip_available()
+            stream_ctx = (
+                torch.cuda.stream(self.stream)
+                if self.stream is not None else contextlib.nullcontext()
+            )
+            with stream_ctx:
+                return self._generate_on_stream(
+                    ids, max_new, temperature, top_p, stop_ids, deadline, timer
                 )
-                if use_async and max_new >= 8 and not os.environ.get("ADVSPEC_NO_GRAPH"):
-                    out_ids = self._decode_graphed(
-                        logits, cache, max_new, temperature, stop_ids, deadline
-                    )
-                elif use_async:
-                    out_ids = self._decode_async(
-                        logits, cache, max_new, temperature, stop_ids, deadline
-                    )
-                else:
-                    out_ids = self._decode_sync(
-                        logits, cache, max_new, temperature, top_p, stop_ids,
-                        deadline,
-                    )
         finally:
             if prev_device is not None:
                 torch.cuda.set_device(prev_device)
+
+    def _get_cache(self, need: int):
+        """Persistent KV cache, bucketed to 2k tokens: the same cache tensors
+        (and therefore the same captured decode graph) serve every round
+        whose prompt+decode fits. Growth reallocates and invalidates the
+        graph."""
+        cache = self._cache
+        if cache is None or cache.max_seq < need:
+            bucket = (need + 2047) // 2048 * 2048
+            cache = self.model.new_cache(bucket)
+            self._cache = cache
+            self._graph_state = None
+        cache.seq_len = 0
+        return cache
+
+    def _generate_on_stream(self, ids, max_new, temperature, top_p, stop_ids,
+                            deadline, timer):
+        cache = self._get_cache(len(ids) + max_new + 8)
+        tokens = torch.tensor(ids, device=self.device, dtype=torch.long)
+        with timer.phase("prefill"):
+            logits = self.model.prefill(tokens, cache)
+
+        out_ids: list[int] = []
+        with timer.phase("decode"):
+            use_async = (
+                self.device.type == "cuda"
+                and top_p >= 1.0
+                and ops.hip_available()
+            )
+            if use_async and max_new >= 8 and not os.environ.get("ADVSPEC_NO_GRAPH"):
+                out_ids = self._decode_graphed(
+                    logits, cache, max_new, temperature, stop_ids, deadline
+                )
+            elif use_async:
+                out_ids = self._decode_async(
+                    logits, cache, max_new, temperature, stop_ids, deadline
+                )
+            else:
+                out_ids = self._decode_sync(
+                    logits, cache, max_new, temperature, top_p, stop_ids,
+                    deadline,
+                )
 
         text = self.tokenizer.decode(out_ids)
         return text, len(ids), len(out_ids), timer.as_dict()
@@ -180,41 +223,76 @@ class LocalEngine:
         hip = _load_hip()
         dev = self.device
         prompt_len = cache.seq_len
-        max_total = min(cache.max_seq, prompt_len + max_new + 1)
+        # replay-invariant attention bound: the split geometry is sized once
+        # for the whole cache, so the SAME captured graph serves every round
+        # that reuses this cache (splits past the live position contribute
+        # nothing in the combine).
+        max_total = cache.max_seq
 
-        pos_state = torch.tensor([prompt_len], dtype=torch.int32, device=dev)
-        step_state = torch.zeros(1, dtype=torch.int32, device=dev)
-        rng_state = torch.tensor([self._next_seed() | 1], dtype=torch.int32,
-                                 device=dev)
-        tok_hist = torch.full((max_new + 2,), -1, dtype=torch.int32, device=dev)
-        tok_slot = torch.zeros(1, dtype=torch.int32, device=dev)
-        logits_buf = logits.reshape(-1).contiguous().clone()
+        gs = self._graph_state
+        key = (id(cache), cache.max_seq, max_new, float(temperature))
+        if gs is not None and gs["key"] == key:
+            # Round k+1 on a warm engine: reset the device-side state words
+            # and replay the graph captured in round 1 — no re-capture.
+            pos_state = gs["pos_state"]
+            step_state = gs["step_state"]
+            rng_state = gs["rng_state"]
+            tok_hist = gs["tok_hist"]
+            tok_slot = gs["tok_slot"]
+            logits_buf = gs["logits_buf"]
+            graph = gs["graph"]
+            pos_state.fill_(prompt_len)
+            step_state.zero_()
+            rng_state.fill_(self._next_seed() | 1)
+            tok_hist.fill_(-1)
+            tok_slot.zero_()
+            logits_buf.copy_(logits.reshape(-1))
+            warm = 0
+        else:
+            pos_state = torch.tensor([prompt_len], dtype=torch.int32, device=dev)
+            step_state = torch.zeros(1, dtype=torch.int32, device=dev)
+            rng_state = torch.tensor([self._next_seed() | 1], dtype=torch.int32,
+                                     device=dev)
+            tok_hist = torch.full((max_new + 2,), -1, dtype=torch.int32, device=dev)
+            tok_slot = torch.zeros(1, dtype=torch.int32, device=dev)
+            logits_buf = logits.reshape(-1).contiguous().clone()
 
-        def step():
-            hip.sample_state(logits_buf, temperature, rng_state, tok_hist,
-                             step_state, tok_slot)
-            lg = self.model.decode_one_graph(tok_slot, cache, pos_state, max_total)
-            logits_buf.copy_(lg.reshape(-1))
-            hip.bump(pos_state, step_state)
+            def step():
+                hip.sample_state(logits_buf, temperature, rng_state, tok_hist,
+                                 step_state, tok_slot)
+                lg = self.model.decode_one_graph(tok_slot, cache, pos_state,
+                                                 max_total)
+                logits_buf.copy_(lg.reshape(-1))
+                hip.bump(pos_state, step_state)
 
-        # warm up on a side stream (torch graph-capture requirement); these
-        # are REAL decode steps — their tokens land in tok_hist[0..1].
-        warm = min(2, max_new)
-        s = torch.cuda.Stream(device=dev)
-        s.wait_stream(torch.cuda.current_stream(dev))
-        with torch.cuda.stream(s):
-            for _ in range(warm):
-                step()
-        torch.cuda.current_stream(dev).wait_stream(s)
+            # warm up on a side stream (torch graph-capture requirement);
+            # these are REAL decode steps — tokens land in tok_hist[0..1].
+            warm = min(2, max_new)
+            s = torch.cuda.Stream(device=dev)
+            s.wait_stream(torch.cuda.current_stream(dev))
+            with torch.cuda.stream(s):
+                for _ in range(warm):
+                    step()
+            torch.cuda.current_stream(dev).wait_stream(s)
 
-        graph = torch.cuda.CUDAGraph()
-        with torch.cuda.graph(graph):
-            step()
+            # thread_local capture mode + a process-wide capture lock: other
+            # opponents' threads keep replaying their own graphs on their own
+            # streams while this engine captures (global mode would abort the
+            # capture on any concurrent allocator traffic).
+            graph = torch.cuda.CUDAGraph()
+            with _CAPTURE_LOCK:
+                with torch.cuda.graph(graph, capture_error_mode="thread_local"):
+                    step()
+            self._graph_state = {
+                "key": key, "graph": graph, "pos_state": pos_state,
+                "step_state": step_state, "rng_state": rng_state,
+                "tok_hist": tok_hist, "tok_slot": tok_slot,
+                "logits_buf": logits_buf,
+            }
 
         CHECK = 32
         done = False
         i = warm
-        n_keep = None
         out_ids: list[int] = []
         tail = ""
         scanned = 0

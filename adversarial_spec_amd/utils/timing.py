@@ -23,7 +23,10 @@ class PhaseTimer:
             import torch
 
             if torch.cuda.is_available():
-                torch.cuda.synchronize()
+                # Sync only this thread's stream: co-resident opponents run
+                # concurrently on per-engine streams, and a device-wide
+                # synchronize here would serialize them at phase boundaries.
+                torch.cuda.current_stream().synchronize()
 
     @contextmanager
     def phase(self, name: str) -> Iterator[None]:

@@ -27,6 +27,7 @@ import json
 import os
 import sys
 import time
+from concurrent.futures import ThreadPoolExecutor
 
 import torch
 
@@ -128,19 +129,33 @@ def main() -> int:
         if (dist is not None and not tp_mode) else None
     )
 
+    pool = ThreadPoolExecutor(max_workers=max(1, n_opp))
+
+    def run_one(eng):
+        text, _in, _out, _tm = eng.generate(
+            system_prompt,
+            user_message,
+            max_tokens=args.decode_tokens,
+            temperature=args.temperature,
+            timeout=600.0,
+        )
+        ids = eng.tokenizer.encode(text)[: max_gather_tokens]
+        agreed = detect_agreement(text)
+        return ids, agreed
+
     def one_round():
-        """One debate round for this rank's opponents + consensus gather."""
+        """One debate round for this rank's opponents + consensus gather.
+
+        Co-resident opponents run CONCURRENTLY: each engine issues its
+        prefill/decode on its own HIP stream from its own host thread, so
+        the decode GEMVs of N opponents interleave and fill HBM bandwidth
+        that a single bandwidth-bound decode leaves idle. Collectives are
+        still launched in opponent order (identical on every rank).
+        """
         local = []
-        for i, eng in enumerate(engines):
-            text, _in, _out, _tm = eng.generate(
-                system_prompt,
-                user_message,
-                max_tokens=args.decode_tokens,
-                temperature=args.temperature,
-                timeout=600.0,
-            )
-            ids = eng.tokenizer.encode(text)[: max_gather_tokens]
-            agreed = detect_agreement(text)
+        futures = [pool.submit(run_one, eng) for eng in engines]
+        for i, fut in enumerate(futures):
+            ids, agreed = fut.result()
             if gathers is not None:
                 gathers[i].launch(ids, agreed, False)
             else:
