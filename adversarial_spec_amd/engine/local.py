@@ -279,9 +279,14 @@ class LocalEngine:
             # opponents' threads keep replaying their own graphs on their own
             # streams while this engine captures (global mode would abort the
             # capture on any concurrent allocator traffic).
+            # capture on the ENGINE's stream (not torch's shared default
+            # capture stream): the fused split-K kernels use per-stream
+            # scratch, so two graphs captured on one stream would share
+            # buffers and race when replayed concurrently.
             graph = torch.cuda.CUDAGraph()
             with _CAPTURE_LOCK:
-                with torch.cuda.graph(graph, capture_error_mode="thread_local"):
+                with torch.cuda.graph(graph, stream=self.stream,
+                                      capture_error_mode="thread_local"):
                     step()
             self._graph_state = {
                 "key": key, "graph": graph, "pos_state": pos_state,

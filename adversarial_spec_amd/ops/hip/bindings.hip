@@ -7,15 +7,18 @@
 #include <c10/hip/HIPStream.h>
 #include <hip/hip_runtime.h>
 
+#include <mutex>
+#include <unordered_map>
+#include <vector>
+
 using ushort_t = unsigned short;
 
 // launch wrappers defined in the .hip translation units
 extern "C" {
 void launch_attn_decode_split(const ushort_t*, const ushort_t*, const ushort_t*,
                               const int*, int, float, int, int, int, int, int,
-                              int, float*, float*, float*, const int*, hipStream_t);
-void launch_attn_decode_combine(const float*, const float*, const float*,
-                                ushort_t*, int, int, int, int, hipStream_t);
+                              int, float*, float*, float*, const int*,
+                              unsigned int*, ushort_t*, hipStream_t);
 void launch_attn_prefill_simple(const ushort_t*, const ushort_t*, const ushort_t*,
                                 ushort_t*, int, int, int, float, int, int, int,
                                 int, hipStream_t);
@@ -35,12 +38,53 @@ __global__ void sample_kernel(const ushort_t*, int, float, uint32_t, int*);
 __global__ void sample_state_kernel(const ushort_t*, int, float, uint32_t*, int*,
                                     const int*, int*);
 __global__ void bump_kernel(int*, int*);
-void launch_gemv(const ushort_t*, const ushort_t*, float*, ushort_t*, int, int,
-                 int, hipStream_t);
+void launch_gemv(const ushort_t*, const ushort_t*, float*, unsigned int*,
+                 ushort_t*, int, int, int, hipStream_t);
 }
 
 static hipStream_t cur_stream() {
   return c10::hip::getCurrentHIPStream().stream();
+}
+
+// --------------------------------------------------------------------------
+// Per-stream persistent workspaces for the fused split-K kernels (GEMV and
+// decode attention). Kernels on one stream serialize, so one scratch set
+// per stream is race-free: distinct engines run on distinct streams, and a
+// HIP graph captured on a stream bakes THAT stream's buffers (torch's
+// stream pool is 32/device and engines are few, so capture streams are not
+// shared between concurrently-replaying graphs). The atomic counters are
+// self-resetting (the last block zeroes them), so they are allocated zeroed
+// once and never re-initialized. Grown buffers retire the old tensor into a
+// keep-alive list because a previously captured graph may still replay with
+// the old pointer.
+struct StreamWS {
+  torch::Tensor part;      // fp32 scratch: gemv partials
+  torch::Tensor gemv_ctr;  // u32 per-column-block arrival counters
+  torch::Tensor attn_ws;   // fp32 scratch: decode-attention m/l/acc
+  torch::Tensor attn_ctr;  // u32 per-kv-head arrival counters
+  std::vector<torch::Tensor> retired;
+};
+static std::mutex g_ws_mu;
+static std::unordered_map<void*, StreamWS> g_ws;
+
+static float* ws_f32(torch::Tensor& t, std::vector<torch::Tensor>& retired,
+                     int64_t elems, const torch::Device& dev) {
+  if (!t.defined() || t.numel() < elems) {
+    if (t.defined()) retired.push_back(t);
+    t = torch::empty({std::max<int64_t>(elems, 64)},
+                     torch::TensorOptions().dtype(at::kFloat).device(dev));
+  }
+  return t.data_ptr<float>();
+}
+static unsigned int* ws_ctr(torch::Tensor& t,
+                            std::vector<torch::Tensor>& retired,
+                            int64_t elems, const torch::Device& dev) {
+  if (!t.defined() || t.numel() < elems) {
+    if (t.defined()) retired.push_back(t);
+    t = torch::zeros({std::max<int64_t>(elems, 64)},
+                     torch::TensorOptions().dtype(at::kInt).device(dev));
+  }
+  return reinterpret_cast<unsigned int*>(t.data_ptr<int>());
 }
 
 #define CHECK_BF16_CUDA(t)                                                 \
@@ -252,21 +296,22 @@ torch::Tensor attn_decode_paged(torch::Tensor q, torch::Tensor kc,
   int n_splits = std::max(1, std::min((int)((seq_len + 63) / 64), 1024 / kh));
   int split_len = (int)((seq_len + n_splits - 1) / n_splits + 63) / 64 * 64;
   n_splits = (int)((seq_len + split_len - 1) / split_len);
-  auto wopt = torch::TensorOptions().dtype(at::kFloat).device(q.device());
-  auto ws_m = torch::empty({(long)kh * n_splits * group}, wopt);
-  auto ws_l = torch::empty({(long)kh * n_splits * group}, wopt);
-  auto ws_acc = torch::empty({(long)kh * n_splits * group * hd}, wopt);
   auto out = torch::empty({hq, hd}, qc.options());
-
+  const long khnsg = (long)kh * n_splits * group;
+  auto stream = cur_stream();
+  float* ws;
+  unsigned int* ctr;
+  {
+    std::lock_guard<std::mutex> lk(g_ws_mu);
+    auto& w = g_ws[(void*)stream];
+    ws = ws_f32(w.attn_ws, w.retired, khnsg * (2 + hd), q.device());
+    ctr = ws_ctr(w.attn_ctr, w.retired, kh, q.device());
+  }
   launch_attn_decode_split(uptr(qc), uptr(kc), uptr(vc),
                            page_table.data_ptr<int>(), (int)seq_len,
                            (float)scale, kh, group, hd, page, split_len,
-                           n_splits, ws_m.data_ptr<float>(),
-                           ws_l.data_ptr<float>(), ws_acc.data_ptr<float>(),
-                           nullptr, cur_stream());
-  launch_attn_decode_combine(ws_m.data_ptr<float>(), ws_l.data_ptr<float>(),
-                             ws_acc.data_ptr<float>(), uptr_mut(out), hq,
-                             n_splits, group, hd, cur_stream());
+                           n_splits, ws, ws + khnsg, ws + 2 * khnsg,
+                           nullptr, ctr, uptr_mut(out), stream);
   return out;
 }
 
@@ -296,13 +341,20 @@ torch::Tensor gemv(torch::Tensor x, torch::Tensor w) {
   const int ncb = N / 64;
   int ksplit = std::max(1, std::min(16, 1024 / std::max(1, ncb)));
   ksplit = std::min(ksplit, std::max(1, K / 64));
-  auto part = torch::empty({(long)ksplit * N},
-                           torch::TensorOptions().dtype(at::kFloat).device(x.device()));
   auto y = torch::empty(x.dim() == 2 ? std::vector<int64_t>{1, N}
                                      : std::vector<int64_t>{N},
                         x.options());
-  launch_gemv(uptr(xc), uptr(w), part.data_ptr<float>(), uptr_mut(y), K, N,
-              ksplit, cur_stream());
+  auto stream = cur_stream();
+  float* part = nullptr;
+  unsigned int* ctr = nullptr;
+  if (ksplit > 1) {
+    std::lock_guard<std::mutex> lk(g_ws_mu);
+    auto& w = g_ws[(void*)stream];
+    part = ws_f32(w.part, w.retired, (int64_t)ksplit * N, x.device());
+    ctr = ws_ctr(w.gemv_ctr, w.retired, ncb, x.device());
+  }
+  launch_gemv(uptr(xc), uptr(w), part, ctr, uptr_mut(y), K, N, ksplit,
+              stream);
   return y;
 }
 
@@ -335,20 +387,23 @@ torch::Tensor attn_decode_paged_ds(torch::Tensor q, torch::Tensor kc,
   int n_splits = std::max(1, std::min((int)((max_seq + 63) / 64), 1024 / kh));
   int split_len = (int)((max_seq + n_splits - 1) / n_splits + 63) / 64 * 64;
   n_splits = (int)((max_seq + split_len - 1) / split_len);
-  auto wopt = torch::TensorOptions().dtype(at::kFloat).device(q.device());
-  auto ws_m = torch::empty({(long)kh * n_splits * group}, wopt);
-  auto ws_l = torch::empty({(long)kh * n_splits * group}, wopt);
-  auto ws_acc = torch::empty({(long)kh * n_splits * group * hd}, wopt);
   auto out = torch::empty({hq, hd}, qc.options());
+  const long khnsg = (long)kh * n_splits * group;
+  auto stream = cur_stream();
+  float* ws;
+  unsigned int* ctr;
+  {
+    std::lock_guard<std::mutex> lk(g_ws_mu);
+    auto& w = g_ws[(void*)stream];
+    ws = ws_f32(w.attn_ws, w.retired, khnsg * (2 + hd), q.device());
+    ctr = ws_ctr(w.attn_ctr, w.retired, kh, q.device());
+  }
   launch_attn_decode_split(uptr(qc), uptr(kc), uptr(vc),
                            page_table.data_ptr<int>(), (int)max_seq,
                            (float)scale, kh, group, hd, page, split_len,
-                           n_splits, ws_m.data_ptr<float>(),
-                           ws_l.data_ptr<float>(), ws_acc.data_ptr<float>(),
-                           pos_state.data_ptr<int>(), cur_stream());
-  launch_attn_decode_combine(ws_m.data_ptr<float>(), ws_l.data_ptr<float>(),
-                             ws_acc.data_ptr<float>(), uptr_mut(out), hq,
-                             n_splits, group, hd, cur_stream());
+                           n_splits, ws, ws + khnsg, ws + 2 * khnsg,
+                           pos_state.data_ptr<int>(), ctr, uptr_mut(out),
+                           stream);
   return out;
 }
 
