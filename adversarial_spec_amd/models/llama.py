@@ -258,9 +258,9 @@ class LlamaModel:
             q = qkv[:, : h * hd].view(t, h, hd)
             k = qkv[:, h * hd : (h + kh) * hd].view(t, kh, hd)
             v = qkv[:, (h + kh) * hd :].view(t, kh, hd)
-            q, k = ops.rope(q, k, self.cos, self.sin, pos0, pos_state=pos_state)
-            ops.kv_write(cache.k[i], cache.v[i], cache.page_table, pos0, k, v,
-                         pos_state=pos_state)
+            q, k = ops.rope_kv(q, k, v, self.cos, self.sin, cache.k[i],
+                               cache.v[i], cache.page_table, pos0,
+                               pos_state=pos_state)
             if t > 1:
                 if pos0 != 0:
                     raise NotImplementedError("chunked prefill lands with the 32k path")

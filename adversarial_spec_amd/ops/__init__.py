@@ -149,6 +149,30 @@ def kv_write(
     torch_ref.kv_write(k_cache, v_cache, page_table, pos0, k, v)
 
 
+def rope_kv(
+    q: torch.Tensor,
+    k: torch.Tensor,
+    v: torch.Tensor,
+    cos: torch.Tensor,
+    sin: torch.Tensor,
+    k_cache: torch.Tensor,
+    v_cache: torch.Tensor,
+    page_table: torch.Tensor,
+    pos0: int,
+    pos_state: Optional[torch.Tensor] = None,
+) -> Tuple[torch.Tensor, torch.Tensor]:
+    """Fused RoPE (q,k in place) + paged KV scatter of rotated k and v.
+    One launch replaces rope + kv_write on the kernel-count-bound decode
+    path. Returns (q, k) rotated."""
+    if _on_gpu(q):
+        _require_hip().rope_kv(q, k, v, cos, sin, k_cache, v_cache,
+                               page_table, pos0, pos_state)
+        return q, k
+    q2, k2 = torch_ref.rope(q, k, cos, sin, pos0)
+    torch_ref.kv_write(k_cache, v_cache, page_table, pos0, k2, v)
+    return q2, k2
+
+
 def gemv(x: torch.Tensor, w: torch.Tensor) -> torch.Tensor:
     """Batch-1 matmul y = x @ w. On GPU, a hand-written weight-streaming
     kernel (decode's dominant cost is reading w once from HBM3E); falls

@@ -34,6 +34,10 @@ __global__ void rope_kernel(ushort_t*, ushort_t*, const float*, const float*,
 __global__ void swiglu_kernel(const ushort_t*, const ushort_t*, ushort_t*, int, int, int);
 __global__ void kv_write_kernel(const ushort_t*, const ushort_t*, ushort_t*, ushort_t*,
                                 const int*, int, int, int, int, int, const int*);
+__global__ void rope_kv_kernel(ushort_t*, ushort_t*, const ushort_t*,
+                               const float*, const float*, ushort_t*, ushort_t*,
+                               const int*, int, int, int, int, int, long, long,
+                               long, int, const int*);
 __global__ void sample_kernel(const ushort_t*, int, float, uint32_t, int*);
 __global__ void sample_state_kernel(const ushort_t*, int, float, uint32_t*, int*,
                                     const int*, int*);
@@ -165,6 +169,39 @@ void rope_inplace_ds(torch::Tensor q, torch::Tensor k, torch::Tensor cost,
   rope_kernel<<<blocks, 256, 0, cur_stream()>>>(
       uptr_mut(q), uptr_mut(k), cost.data_ptr<float>(), sint.data_ptr<float>(),
       t, hq, hk, hd, 0, q.stride(0), k.stride(0), pos_state.data_ptr<int>());
+}
+
+// Fused RoPE + paged KV scatter: rotate q/k in place AND write the rotated
+// k plus v into the page pool in one launch (saves a kernel + a K re-read
+// per layer on the kernel-count-bound decode path). pos_state (int32[1] on
+// device) replaces pos0 in graph mode.
+void rope_kv(torch::Tensor q, torch::Tensor k, torch::Tensor v,
+             torch::Tensor cost, torch::Tensor sint, torch::Tensor kc,
+             torch::Tensor vc, torch::Tensor page_table, int64_t pos0,
+             const c10::optional<torch::Tensor>& pos_state) {
+  CHECK_BF16_CUDA(q);
+  CHECK_BF16_CUDA(k);
+  CHECK_BF16_CUDA(kc);
+  TORCH_CHECK(q.stride(2) == 1 && k.stride(2) == 1 && v.stride(2) == 1,
+              "rope_kv: dim contiguous");
+  TORCH_CHECK(q.stride(1) == q.size(2) && k.stride(1) == k.size(2) &&
+                  v.stride(1) == v.size(2),
+              "rope_kv: head dim contiguous");
+  TORCH_CHECK(cost.scalar_type() == at::kFloat, "rope_kv: cos table f32");
+  TORCH_CHECK(kc.is_contiguous() && vc.is_contiguous(), "cache contiguous");
+  TORCH_CHECK(page_table.scalar_type() == at::kInt);
+  const int t = q.size(0), hq = q.size(1), hd = q.size(2);
+  const int hk = k.size(1);
+  const int page = kc.size(1);
+  TORCH_CHECK(hd % 16 == 0, "rope_kv: hd % 16 == 0");
+  const int waves = t * (hq + 2 * hk);
+  const int blocks = (waves * 64 + 255) / 256;
+  const int* pp = pos_state.has_value() ? pos_state->data_ptr<int>() : nullptr;
+  rope_kv_kernel<<<blocks, 256, 0, cur_stream()>>>(
+      uptr_mut(q), uptr_mut(k), uptr(v), cost.data_ptr<float>(),
+      sint.data_ptr<float>(), uptr_mut(kc), uptr_mut(vc),
+      page_table.data_ptr<int>(), t, hq, hk, hd, (int)pos0, q.stride(0),
+      k.stride(0), v.stride(0), page, pp);
 }
 
 torch::Tensor swiglu(torch::Tensor gate, torch::Tensor up) {
@@ -323,7 +360,7 @@ void sample_to(torch::Tensor logits, double temp, int64_t seed,
   CHECK_BF16_CUDA(logits);
   TORCH_CHECK(out.scalar_type() == at::kInt && out.is_cuda());
   auto lc = logits.contiguous();
-  sample_kernel<<<1, 256, 0, cur_stream()>>>(
+  sample_kernel<<<1, 1024, 0, cur_stream()>>>(
       uptr(lc), (int)lc.numel(), (float)temp, (uint32_t)seed,
       out.data_ptr<int>() + idx);
 }
@@ -366,7 +403,7 @@ int64_t sample(torch::Tensor logits, double temp, double top_p, int64_t seed) {
   auto out = torch::empty({1}, torch::TensorOptions()
                                    .dtype(at::kInt)
                                    .device(logits.device()));
-  sample_kernel<<<1, 256, 0, cur_stream()>>>(uptr(lc), vocab, (float)temp,
+  sample_kernel<<<1, 1024, 0, cur_stream()>>>(uptr(lc), vocab, (float)temp,
                                              (uint32_t)seed,
                                              out.data_ptr<int>());
   return out.cpu().item<int>();
@@ -412,7 +449,7 @@ void sample_state(torch::Tensor logits, double temp, torch::Tensor rng_state,
                   torch::Tensor tok_slot) {
   CHECK_BF16_CUDA(logits);
   auto lc = logits.contiguous();
-  sample_state_kernel<<<1, 256, 0, cur_stream()>>>(
+  sample_state_kernel<<<1, 1024, 0, cur_stream()>>>(
       uptr(lc), (int)lc.numel(), (float)temp,
       reinterpret_cast<uint32_t*>(rng_state.data_ptr<int>()),
       tok_hist.data_ptr<int>(), step_state.data_ptr<int>(),
@@ -438,6 +475,10 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("sample_to", &sample_to, "async on-device sample into out[idx]");
   m.def("gemv", &gemv, "batch-1 decode GEMV (weight streaming)");
   m.def("rope_inplace_ds", &rope_inplace_ds, "graph-mode RoPE (device pos)");
+  m.def("rope_kv", &rope_kv, "fused RoPE + paged KV scatter",
+        py::arg("q"), py::arg("k"), py::arg("v"), py::arg("cost"),
+        py::arg("sint"), py::arg("kc"), py::arg("vc"), py::arg("page_table"),
+        py::arg("pos0"), py::arg("pos_state") = py::none());
   m.def("kv_write_ds", &kv_write_ds, "graph-mode KV scatter (device pos)");
   m.def("attn_decode_paged_ds", &attn_decode_paged_ds,
         "graph-mode paged decode attention (device pos)");

@@ -137,6 +137,69 @@ rope_kernel(ushort_t *__restrict__ q, ushort_t *__restrict__ k,
 }
 
 // ---------------------------------------------------------------------------
+// Fused RoPE + paged KV scatter. Decode is kernel-count bound (~450
+// launches/token); this replaces the rope_kernel + kv_write_kernel pair and
+// also saves one full re-read of K (rope read+wrote it, kv_write re-read it).
+//   waves map over t * (hq + 2*hk) "slots":
+//     slot < hq:            rotate q in place
+//     hq <= slot < hq+hk:   rotate k in place AND scatter the rotated row
+//                           into the cache page pool
+//     slot >= hq+hk:        copy v into the cache page pool
+// ---------------------------------------------------------------------------
+
+extern "C" __global__ void __launch_bounds__(256)
+rope_kv_kernel(ushort_t *__restrict__ q, ushort_t *__restrict__ k,
+               const ushort_t *__restrict__ v,
+               const float *__restrict__ cost, const float *__restrict__ sint,
+               ushort_t *__restrict__ kc, ushort_t *__restrict__ vc,
+               const int *__restrict__ page_table, int t, int hq, int hk,
+               int hd, int pos0, long q_rstride, long k_rstride,
+               long v_rstride, int page, const int *__restrict__ pos_ptr) {
+  if (pos_ptr) pos0 = *pos_ptr;
+  const int wave_id = (blockIdx.x * blockDim.x + threadIdx.x) / WAVE;
+  const int lane = threadIdx.x & (WAVE - 1);
+  const int slots = hq + 2 * hk;
+  const int tok = wave_id / slots;
+  if (tok >= t) return;
+  const int slot = wave_id % slots;
+  const int pos = pos0 + tok;
+  const int half = hd / 2;
+
+  const int phys = page_table[pos / page];
+  const size_t cache_row = ((size_t)phys * page + (pos % page)) * hk * hd;
+
+  if (slot >= hq + hk) {  // v copy: no rotation
+    const int vh = slot - hq - hk;
+    const bf16x8 *src =
+        (const bf16x8 *)(v + (size_t)tok * v_rstride + (size_t)vh * hd);
+    bf16x8 *dst = (bf16x8 *)(vc + cache_row + (size_t)vh * hd);
+    for (int i = lane; i < hd / 8; i += WAVE) dst[i] = src[i];
+    return;
+  }
+
+  const bool is_q = slot < hq;
+  ushort_t *base = is_q
+                       ? q + (size_t)tok * q_rstride + (size_t)slot * hd
+                       : k + (size_t)tok * k_rstride + (size_t)(slot - hq) * hd;
+  uint32_t *cache_dst =
+      is_q ? nullptr
+           : (uint32_t *)(kc + cache_row + (size_t)(slot - hq) * hd);
+  const float *crow = cost + (size_t)pos * half;
+  const float *srow = sint + (size_t)pos * half;
+
+  for (int p = lane; p < half; p += WAVE) {
+    uint32_t pair = ((uint32_t *)base)[p];  // 2 bf16: (even, odd)
+    float ev = bf16_to_f32((ushort_t)(pair & 0xffffu));
+    float od = bf16_to_f32((ushort_t)(pair >> 16));
+    float c = crow[p], s = srow[p];
+    const uint32_t rot = (uint32_t)f32_to_bf16(ev * c - od * s) |
+                         ((uint32_t)f32_to_bf16(ev * s + od * c) << 16);
+    ((uint32_t *)base)[p] = rot;
+    if (cache_dst) cache_dst[p] = rot;
+  }
+}
+
+// ---------------------------------------------------------------------------
 // SwiGLU: out = silu(gate) * up, elementwise over [n] (n = t * ffn).
 // gate/up are the two contiguous halves of the fused gate_up GEMM output,
 // passed as separate base pointers with a row stride.

@@ -19,122 +19,60 @@
 
 // logits: [vocab] bf16, vocab % 8 == 0. temp <= 0 -> greedy. Returns the
 // sampled id in *out; shared body for the plain and graph-state kernels.
+//
+// Softmax sampling via the GUMBEL-MAX trick: argmax_i(logit_i/T + g_i) with
+// g_i = -log(-log(u_i)) is an exact draw from softmax(logits/T). That
+// collapses the old 3-pass (max, Z, CDF-scan) sampler into ONE pass over
+// the vocab — the decode step's tail latency (the lm_head logits row is
+// L2-resident when this runs). Per-element RNG is a counter-based hash of
+// (seed, index), so HIP-graph replays stay deterministic for a given seed.
 DEVINL void sample_body(const ushort_t *__restrict__ logits, int vocab,
                         float temp, uint32_t seed, int *__restrict__ out) {
-  __shared__ float scratch[16];
-  __shared__ float tsum[256];
-  __shared__ int result;
   const int tid = threadIdx.x;
   const int nvec = vocab / 8;
   const bf16x8 *lv = (const bf16x8 *)logits;
+  const bool greedy = temp <= 0.f;
+  const float inv_t = greedy ? 1.0f : 1.0f / temp;
 
-  // ---- pass 1: max (+argmax for greedy) ----
-  float vmax = -INFINITY;
-  int amax = 0;
+  float best = -INFINITY;
+  int besti = 0;
   for (int i = tid; i < nvec; i += blockDim.x) {
     const f32x8 v = unpack8(lv[i]);
 #pragma unroll
     for (int j = 0; j < 8; ++j) {
-      if (v.v[j] > vmax) { vmax = v.v[j]; amax = i * 8 + j; }
-    }
-  }
-  {
-#pragma unroll
-    for (int off = 32; off > 0; off >>= 1) {
-      const float ov = __shfl_xor(vmax, off, WAVE);
-      const int oi = __shfl_xor(amax, off, WAVE);
-      if (ov > vmax || (ov == vmax && oi < amax)) { vmax = ov; amax = oi; }
-    }
-    __shared__ float mv[16];
-    __shared__ int mi[16];
-    const int wid = tid / WAVE;
-    if ((tid & (WAVE - 1)) == 0) { mv[wid] = vmax; mi[wid] = amax; }
-    __syncthreads();
-    vmax = -INFINITY; amax = 0;
-    for (int w = 0; w < (int)(blockDim.x / WAVE); ++w) {
-      if (mv[w] > vmax || (mv[w] == vmax && mi[w] < amax)) {
-        vmax = mv[w]; amax = mi[w];
+      const int idx = i * 8 + j;
+      float val = v.v[j] * inv_t;
+      if (!greedy) {
+        const uint32_t h = hash_u32(seed ^ ((uint32_t)idx * 2654435761u));
+        const float u = ((float)h + 1.0f) * 2.3283064e-10f;  // (0, 1]
+        val -= __logf(-__logf(u) + 1e-30f);  // + Gumbel(0,1)
       }
+      if (val > best || (val == best && idx < besti)) { best = val; besti = idx; }
     }
   }
-
-  if (temp <= 0.f) {
-    if (tid == 0) *out = amax;
-    return;
-  }
-  const float inv_t = 1.0f / temp;
-
-  // ---- pass 2: Z = sum exp((v - max)/temp) ----
-  float z = 0.f;
-  for (int i = tid; i < nvec; i += blockDim.x) {
-    const f32x8 v = unpack8(lv[i]);
+  // wave argmax reduce (ties -> lowest index, replay-deterministic)
 #pragma unroll
-    for (int j = 0; j < 8; ++j) z += __expf((v.v[j] - vmax) * inv_t);
+  for (int off = 32; off > 0; off >>= 1) {
+    const float ov = __shfl_xor(best, off, WAVE);
+    const int oi = __shfl_xor(besti, off, WAVE);
+    if (ov > best || (ov == best && oi < besti)) { best = ov; besti = oi; }
   }
-  z = block_reduce_sum(z, scratch);
-
-  // draw u in (0, Z]; golden-ratio offset avoids the degenerate hash(0)=0
-  const float target = uniform01(seed ^ 0x9e3779b9u) * z;
-
-  // ---- pass 3: find the crossing element ----
-  if (tid == 0) result = amax;  // fallback if rounding exhausts the scan
+  __shared__ float mv[16];
+  __shared__ int mi[16];
+  const int wid = tid / WAVE;
+  if ((tid & (WAVE - 1)) == 0) { mv[wid] = best; mi[wid] = besti; }
   __syncthreads();
-  float running = 0.f;
-  const int per_iter = blockDim.x * 8;  // elements per block iteration
-  for (int base = 0; base < vocab; base += per_iter) {
-    const int i = base / 8 + tid;
-    float my = 0.f;
-    f32x8 ev;
-    if (i < nvec) {
-      const f32x8 v = unpack8(lv[i]);
-#pragma unroll
-      for (int j = 0; j < 8; ++j) {
-        ev.v[j] = __expf((v.v[j] - vmax) * inv_t);
-        my += ev.v[j];
+  if (tid == 0) {
+    for (int w = 1; w < (int)(blockDim.x / WAVE); ++w) {
+      if (mv[w] > best || (mv[w] == best && mi[w] < besti)) {
+        best = mv[w]; besti = mi[w];
       }
-    } else {
-#pragma unroll
-      for (int j = 0; j < 8; ++j) ev.v[j] = 0.f;
     }
-    tsum[tid] = my;
-    const float csum = block_reduce_sum(my, scratch);
-    __syncthreads();
-    if (running + csum >= target) {
-      // crossing chunk: thread 0 scans the 256 per-thread sums, the owner
-      // thread then pinpoints its element.
-      __shared__ int owner;
-      __shared__ float owner_base;
-      if (tid == 0) {
-        float acc = running;
-        int who = blockDim.x - 1;
-        for (int s = 0; s < (int)blockDim.x; ++s) {
-          if (acc + tsum[s] >= target) { who = s; break; }
-          acc += tsum[s];
-        }
-        owner = who;
-        owner_base = acc;
-      }
-      __syncthreads();
-      if (tid == owner) {
-        float acc = owner_base;
-        int pick = -1;
-#pragma unroll
-        for (int j = 0; j < 8; ++j) {
-          acc += ev.v[j];
-          if (acc >= target) { pick = base + tid * 8 + j; break; }
-        }
-        if (pick >= 0 && pick < vocab) result = pick;
-      }
-      break;
-    }
-    running += csum;
-    __syncthreads();
+    *out = besti;
   }
-  __syncthreads();
-  if (tid == 0) *out = result;
 }
 
-extern "C" __global__ void __launch_bounds__(256)
+extern "C" __global__ void __launch_bounds__(1024)
 sample_kernel(const ushort_t *__restrict__ logits, int vocab, float temp,
               uint32_t seed, int *__restrict__ out) {
   sample_body(logits, vocab, temp, seed, out);
@@ -146,7 +84,7 @@ sample_kernel(const ushort_t *__restrict__ logits, int vocab, float temp,
 //   step_state: int32[1], the decode step index (bumped by bump_kernel)
 //   tok_hist:   int32[max_new] history the host polls every N tokens
 //   tok_slot:   int32[1] fixed slot feeding the next embedding lookup
-extern "C" __global__ void __launch_bounds__(256)
+extern "C" __global__ void __launch_bounds__(1024)
 sample_state_kernel(const ushort_t *__restrict__ logits, int vocab,
                     float temp, uint32_t *__restrict__ rng_state,
                     int *__restrict__ tok_hist,

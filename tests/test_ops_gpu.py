@@ -227,3 +227,59 @@ class TestSampling:
         for s in (1, 2, 3):
             t = ops.sample(logits, temperature=0.7, seed=s)
             assert 0 <= t < 128256
+
+
+class TestRopeKV:
+    def test_fused_matches_rope_then_write(self):
+        """rope_kv == rope (q,k) + kv_write(rotated k, v), incl. strided
+        qkv views (the fused-QKV slice layout the model passes)."""
+        t, hq, kh, hd, ps, npg = 5, 8, 2, 128, 16, 8
+        qkv = _bf(torch.randn(t, (hq + 2 * kh) * hd)).to(DEV)
+        q = qkv[:, : hq * hd].view(t, hq, hd)
+        k = qkv[:, hq * hd : (hq + kh) * hd].view(t, kh, hd)
+        v = qkv[:, (hq + kh) * hd :].view(t, kh, hd)
+        half = hd // 2
+        cos = torch.randn(64, half, device=DEV)
+        sin = torch.randn(64, half, device=DEV)
+        kc = torch.zeros(npg, ps, kh, hd, dtype=torch.bfloat16, device=DEV)
+        vc = torch.zeros_like(kc)
+        table = torch.tensor([5, 2, 7, 0, 1, 3, 4, 6], dtype=torch.int32, device=DEV)
+        pos0 = 11
+
+        q_ref, k_ref = torch_ref.rope(
+            q.float().cpu(), k.float().cpu(), cos.cpu(), sin.cpu(), pos0
+        )
+        kc_ref = torch.zeros(npg, ps, kh, hd)
+        vc_ref = torch.zeros(npg, ps, kh, hd)
+        torch_ref.kv_write(
+            kc_ref, vc_ref, table.cpu(), pos0,
+            k_ref.to(torch.bfloat16).float(), v.float().cpu(),
+        )
+
+        q2, k2 = ops.rope_kv(q, k, v, cos, sin, kc, vc, table, pos0)
+        _assert_close(q2, q_ref, atol=2e-2, name="rope_kv q")
+        _assert_close(k2, k_ref, atol=2e-2, name="rope_kv k")
+        _assert_close(kc, kc_ref, atol=2e-2, name="rope_kv k cache")
+        assert torch.equal(vc.float().cpu(), vc_ref)
+
+    def test_graph_mode_pos_state(self):
+        t, hq, kh, hd, ps, npg = 1, 4, 2, 64, 8, 4
+        q = _bf(torch.randn(t, hq, hd)).to(DEV)
+        k = _bf(torch.randn(t, kh, hd)).to(DEV)
+        v = _bf(torch.randn(t, kh, hd)).to(DEV)
+        cos = torch.randn(32, hd // 2, device=DEV)
+        sin = torch.randn(32, hd // 2, device=DEV)
+        kc = torch.zeros(npg, ps, kh, hd, dtype=torch.bfloat16, device=DEV)
+        vc = torch.zeros_like(kc)
+        table = torch.arange(npg, dtype=torch.int32, device=DEV)
+        pos = 13
+        pos_state = torch.tensor([pos], dtype=torch.int32, device=DEV)
+
+        q_want = q.clone(); k_want = k.clone()
+        kc2 = kc.clone(); vc2 = vc.clone()
+        ops.rope_kv(q_want, k_want, v, cos, sin, kc2, vc2, table, pos)
+        q2, k2 = ops.rope_kv(q, k, v, cos, sin, kc, vc, table, 0,
+                             pos_state=pos_state)
+        assert torch.equal(q2.cpu(), q_want.cpu())
+        assert torch.equal(kc.cpu(), kc2.cpu())
+        assert torch.equal(vc.cpu(), vc2.cpu())
