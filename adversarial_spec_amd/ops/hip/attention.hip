@@ -49,7 +49,7 @@ __global__ void __launch_bounds__(256) attn_decode_split_kernel(
     int kh, int group, int hd, int page, int split_len,
     float *__restrict__ ws_m, float *__restrict__ ws_l,
     float *__restrict__ ws_acc, const int *__restrict__ pos_ptr,
-    unsigned int *__restrict__ counters, ushort_t *__restrict__ out) {
+    ushort_t *__restrict__ out) {
   // graph mode: seq_len = *pos_ptr + 1 (attend up to and incl. the token
   // kv_write just appended at position *pos_ptr)
   if (pos_ptr) seq_len = *pos_ptr + 1;
@@ -185,27 +185,31 @@ __global__ void __launch_bounds__(256) attn_decode_split_kernel(
     }
     ws_acc[base * hd + dd] = A;
   }
-  if (n_splits == 1) return;
+}
 
-  // FUSED cross-split combine: the last split block to finish head g sums
-  // all splits (decode is kernel-count bound; the separate combine kernel
-  // cost ~15.7 us execution per layer). The per-head atomic counters
-  // self-reset to zero so the buffer needs no zeroing between launches or
-  // HIP-graph replays.
-  __threadfence();  // publish this block's ws entries (per-wave stores)
-  __syncthreads();
-  __shared__ unsigned int arrival;
-  if (threadIdx.x == 0) {
-    arrival = atomicAdd(&counters[g], 1u);
-  }
-  __syncthreads();
-  if (arrival != (unsigned)(n_splits - 1)) return;
-  if (threadIdx.x == 0) counters[g] = 0;  // reset for next launch/replay
+// ---------------------------------------------------------------------------
+// Decode combine kernel: grid.x = hq, block = hd threads.
+// out[h, :] = sum_splits(acc * exp(m - M)) / L_total
+// A SEPARATE kernel on purpose: a fused last-block-arrives combine needs
+// device-scope fences, which on the 8-XCD MI355X triggered cross-L2 traffic
+// that slowed the whole device ~5x (round-1 profiles).
+// ---------------------------------------------------------------------------
 
-  for (int tid = threadIdx.x; tid < group * hd; tid += blockDim.x) {
-    const int gi = tid / hd, dd = tid % hd;
-    // 4-way unrolled split walk (serial version is latency-bound at large
-    // split counts).
+extern "C" __global__ void __launch_bounds__(128)
+attn_decode_combine_kernel(const float *__restrict__ ws_m,
+                           const float *__restrict__ ws_l,
+                           const float *__restrict__ ws_acc,
+                           ushort_t *__restrict__ out, int n_splits,
+                           int group, int hd) {
+  const int h = blockIdx.x;
+  const int g = h / group, gi = h % group;
+  const int dd = threadIdx.x;
+  if (dd >= hd) return;
+
+  // 4-way unrolled split walk keeps >=4 independent loads in flight
+  // (the serial version was latency-bound at large split counts).
+  float M = -INFINITY;
+  {
     float m4[4] = {-INFINITY, -INFINITY, -INFINITY, -INFINITY};
     int s = 0;
     for (; s + 4 <= n_splits; s += 4) {
@@ -215,31 +219,30 @@ __global__ void __launch_bounds__(256) attn_decode_split_kernel(
     }
     for (; s < n_splits; ++s)
       m4[0] = fmaxf(m4[0], ws_m[((size_t)g * n_splits + s) * group + gi]);
-    const float M = fmaxf(fmaxf(m4[0], m4[1]), fmaxf(m4[2], m4[3]));
-    float L4[4] = {0.f, 0.f, 0.f, 0.f}, A4[4] = {0.f, 0.f, 0.f, 0.f};
-    s = 0;
-    for (; s + 4 <= n_splits; s += 4) {
+    M = fmaxf(fmaxf(m4[0], m4[1]), fmaxf(m4[2], m4[3]));
+  }
+  float L4[4] = {0.f, 0.f, 0.f, 0.f}, A4[4] = {0.f, 0.f, 0.f, 0.f};
+  int s = 0;
+  for (; s + 4 <= n_splits; s += 4) {
 #pragma unroll
-      for (int u = 0; u < 4; ++u) {
-        const size_t base = ((size_t)g * n_splits + s + u) * group + gi;
-        const float mw = ws_m[base];
-        const float sc = (mw == -INFINITY) ? 0.f : __expf(mw - M);
-        L4[u] += ws_l[base] * sc;
-        A4[u] += ws_acc[base * hd + dd] * sc;
-      }
-    }
-    for (; s < n_splits; ++s) {
-      const size_t base = ((size_t)g * n_splits + s) * group + gi;
+    for (int u = 0; u < 4; ++u) {
+      const size_t base = ((size_t)g * n_splits + s + u) * group + gi;
       const float mw = ws_m[base];
       const float sc = (mw == -INFINITY) ? 0.f : __expf(mw - M);
-      L4[0] += ws_l[base] * sc;
-      A4[0] += ws_acc[base * hd + dd] * sc;
+      L4[u] += ws_l[base] * sc;
+      A4[u] += ws_acc[base * hd + dd] * sc;
     }
-    const float L = L4[0] + L4[1] + L4[2] + L4[3];
-    const float A = A4[0] + A4[1] + A4[2] + A4[3];
-    out[(size_t)(g * group + gi) * hd + dd] =
-        f32_to_bf16(L > 0.f ? A / L : 0.f);
   }
+  for (; s < n_splits; ++s) {
+    const size_t base = ((size_t)g * n_splits + s) * group + gi;
+    const float mw = ws_m[base];
+    const float sc = (mw == -INFINITY) ? 0.f : __expf(mw - M);
+    L4[0] += ws_l[base] * sc;
+    A4[0] += ws_acc[base * hd + dd] * sc;
+  }
+  const float L = L4[0] + L4[1] + L4[2] + L4[3];
+  const float A = A4[0] + A4[1] + A4[2] + A4[3];
+  out[(size_t)h * hd + dd] = f32_to_bf16(L > 0.f ? A / L : 0.f);
 }
 
 // ---------------------------------------------------------------------------
@@ -309,8 +312,7 @@ extern "C" void launch_attn_decode_split(
     const ushort_t *q, const ushort_t *kc, const ushort_t *vc,
     const int *page_table, int seq_len, float scale, int kh, int group,
     int hd, int page, int split_len, int n_splits, float *ws_m, float *ws_l,
-    float *ws_acc, const int *pos_ptr, unsigned int *counters, ushort_t *out,
-    hipStream_t stream) {
+    float *ws_acc, const int *pos_ptr, ushort_t *out, hipStream_t stream) {
   dim3 grid(kh, n_splits);
   const int lds = (4 * MAXG * 2 + 4 * MAXG * hd) * sizeof(float);
   // MG = smallest supported bound >= group keeps the per-head state arrays
@@ -321,15 +323,15 @@ extern "C" void launch_attn_decode_split(
     if (group <= 2)                                                            \
       attn_decode_split_kernel<LPP, 2><<<grid, 256, lds, stream>>>(            \
           q, kc, vc, page_table, seq_len, scale, kh, group, hd, page,          \
-          split_len, ws_m, ws_l, ws_acc, pos_ptr, counters, out);              \
+          split_len, ws_m, ws_l, ws_acc, pos_ptr, out);                        \
     else if (group <= 4)                                                       \
       attn_decode_split_kernel<LPP, 4><<<grid, 256, lds, stream>>>(            \
           q, kc, vc, page_table, seq_len, scale, kh, group, hd, page,          \
-          split_len, ws_m, ws_l, ws_acc, pos_ptr, counters, out);              \
+          split_len, ws_m, ws_l, ws_acc, pos_ptr, out);                        \
     else                                                                       \
       attn_decode_split_kernel<LPP, 8><<<grid, 256, lds, stream>>>(            \
           q, kc, vc, page_table, seq_len, scale, kh, group, hd, page,          \
-          split_len, ws_m, ws_l, ws_acc, pos_ptr, counters, out);              \
+          split_len, ws_m, ws_l, ws_acc, pos_ptr, out);                        \
   } while (0)
 
   switch (hd / 8) {
@@ -338,6 +340,10 @@ extern "C" void launch_attn_decode_split(
     case 16: DISPATCH_MG(16); break;
   }
 #undef DISPATCH_MG
+  if (n_splits > 1) {
+    attn_decode_combine_kernel<<<dim3(kh * group), dim3(hd), 0, stream>>>(
+        ws_m, ws_l, ws_acc, out, n_splits, group, hd);
+  }
 }
 
 extern "C" void launch_attn_prefill_simple(

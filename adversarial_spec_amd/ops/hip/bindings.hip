@@ -18,7 +18,7 @@ extern "C" {
 void launch_attn_decode_split(const ushort_t*, const ushort_t*, const ushort_t*,
                               const int*, int, float, int, int, int, int, int,
                               int, float*, float*, float*, const int*,
-                              unsigned int*, ushort_t*, hipStream_t);
+                              ushort_t*, hipStream_t);
 void launch_attn_prefill_simple(const ushort_t*, const ushort_t*, const ushort_t*,
                                 ushort_t*, int, int, int, float, int, int, int,
                                 int, hipStream_t);
@@ -42,8 +42,8 @@ __global__ void sample_kernel(const ushort_t*, int, float, uint32_t, int*);
 __global__ void sample_state_kernel(const ushort_t*, int, float, uint32_t*, int*,
                                     const int*, int*);
 __global__ void bump_kernel(int*, int*);
-void launch_gemv(const ushort_t*, const ushort_t*, float*, unsigned int*,
-                 ushort_t*, int, int, int, hipStream_t);
+void launch_gemv(const ushort_t*, const ushort_t*, ushort_t*, int, int,
+                 hipStream_t);
 }
 
 static hipStream_t cur_stream() {
@@ -51,21 +51,16 @@ static hipStream_t cur_stream() {
 }
 
 // --------------------------------------------------------------------------
-// Per-stream persistent workspaces for the fused split-K kernels (GEMV and
-// decode attention). Kernels on one stream serialize, so one scratch set
-// per stream is race-free: distinct engines run on distinct streams, and a
-// HIP graph captured on a stream bakes THAT stream's buffers (torch's
-// stream pool is 32/device and engines are few, so capture streams are not
-// shared between concurrently-replaying graphs). The atomic counters are
-// self-resetting (the last block zeroes them), so they are allocated zeroed
-// once and never re-initialized. Grown buffers retire the old tensor into a
-// keep-alive list because a previously captured graph may still replay with
-// the old pointer.
+// Per-stream persistent workspace for the split-KV decode attention.
+// Kernels on one stream serialize, so one scratch set per stream is
+// race-free: distinct engines run on distinct streams, and a HIP graph
+// captured on a stream bakes THAT stream's buffers (torch's stream pool is
+// 32/device and engines are few, so capture streams are not shared between
+// concurrently-replaying graphs). Grown buffers retire the old tensor into
+// a keep-alive list because a previously captured graph may still replay
+// with the old pointer.
 struct StreamWS {
-  torch::Tensor part;      // fp32 scratch: gemv partials
-  torch::Tensor gemv_ctr;  // u32 per-column-block arrival counters
   torch::Tensor attn_ws;   // fp32 scratch: decode-attention m/l/acc
-  torch::Tensor attn_ctr;  // u32 per-kv-head arrival counters
   std::vector<torch::Tensor> retired;
 };
 static std::mutex g_ws_mu;
@@ -79,16 +74,6 @@ static float* ws_f32(torch::Tensor& t, std::vector<torch::Tensor>& retired,
                      torch::TensorOptions().dtype(at::kFloat).device(dev));
   }
   return t.data_ptr<float>();
-}
-static unsigned int* ws_ctr(torch::Tensor& t,
-                            std::vector<torch::Tensor>& retired,
-                            int64_t elems, const torch::Device& dev) {
-  if (!t.defined() || t.numel() < elems) {
-    if (t.defined()) retired.push_back(t);
-    t = torch::zeros({std::max<int64_t>(elems, 64)},
-                     torch::TensorOptions().dtype(at::kInt).device(dev));
-  }
-  return reinterpret_cast<unsigned int*>(t.data_ptr<int>());
 }
 
 #define CHECK_BF16_CUDA(t)                                                 \
@@ -337,18 +322,16 @@ torch::Tensor attn_decode_paged(torch::Tensor q, torch::Tensor kc,
   const long khnsg = (long)kh * n_splits * group;
   auto stream = cur_stream();
   float* ws;
-  unsigned int* ctr;
   {
     std::lock_guard<std::mutex> lk(g_ws_mu);
     auto& w = g_ws[(void*)stream];
     ws = ws_f32(w.attn_ws, w.retired, khnsg * (2 + hd), q.device());
-    ctr = ws_ctr(w.attn_ctr, w.retired, kh, q.device());
   }
   launch_attn_decode_split(uptr(qc), uptr(kc), uptr(vc),
                            page_table.data_ptr<int>(), (int)seq_len,
                            (float)scale, kh, group, hd, page, split_len,
                            n_splits, ws, ws + khnsg, ws + 2 * khnsg,
-                           nullptr, ctr, uptr_mut(out), stream);
+                           nullptr, uptr_mut(out), stream);
   return out;
 }
 
@@ -375,23 +358,10 @@ torch::Tensor gemv(torch::Tensor x, torch::Tensor w) {
   TORCH_CHECK((long)x.numel() == (long)K, "gemv: x numel == K");
   TORCH_CHECK(N % 64 == 0, "gemv: N % 64 == 0");
   auto xc = x.contiguous();
-  const int ncb = N / 64;
-  int ksplit = std::max(1, std::min(16, 1024 / std::max(1, ncb)));
-  ksplit = std::min(ksplit, std::max(1, K / 64));
   auto y = torch::empty(x.dim() == 2 ? std::vector<int64_t>{1, N}
                                      : std::vector<int64_t>{N},
                         x.options());
-  auto stream = cur_stream();
-  float* part = nullptr;
-  unsigned int* ctr = nullptr;
-  if (ksplit > 1) {
-    std::lock_guard<std::mutex> lk(g_ws_mu);
-    auto& w = g_ws[(void*)stream];
-    part = ws_f32(w.part, w.retired, (int64_t)ksplit * N, x.device());
-    ctr = ws_ctr(w.gemv_ctr, w.retired, ncb, x.device());
-  }
-  launch_gemv(uptr(xc), uptr(w), part, ctr, uptr_mut(y), K, N, ksplit,
-              stream);
+  launch_gemv(uptr(xc), uptr(w), uptr_mut(y), K, N, cur_stream());
   return y;
 }
 
@@ -428,18 +398,16 @@ torch::Tensor attn_decode_paged_ds(torch::Tensor q, torch::Tensor kc,
   const long khnsg = (long)kh * n_splits * group;
   auto stream = cur_stream();
   float* ws;
-  unsigned int* ctr;
   {
     std::lock_guard<std::mutex> lk(g_ws_mu);
     auto& w = g_ws[(void*)stream];
     ws = ws_f32(w.attn_ws, w.retired, khnsg * (2 + hd), q.device());
-    ctr = ws_ctr(w.attn_ctr, w.retired, kh, q.device());
   }
   launch_attn_decode_split(uptr(qc), uptr(kc), uptr(vc),
                            page_table.data_ptr<int>(), (int)max_seq,
                            (float)scale, kh, group, hd, page, split_len,
                            n_splits, ws, ws + khnsg, ws + 2 * khnsg,
-                           pos_state.data_ptr<int>(), ctr, uptr_mut(out),
+                           pos_state.data_ptr<int>(), uptr_mut(out),
                            stream);
   return out;
 }
