@@ -159,13 +159,15 @@ class LocalEngine:
                 and top_p >= 1.0
                 and ops.hip_available()
             )
-            if use_async and max_new >= 8 and not os.environ.get("ADVSPEC_NO_GRAPH"):
+            if use_async:
+                # same device-state step either way; capture only pays off
+                # for longer generations (ADVSPEC_NO_GRAPH forces eager —
+                # same kernels, bitwise-identical tokens)
+                use_graph = (max_new >= 8
+                             and not os.environ.get("ADVSPEC_NO_GRAPH"))
                 out_ids = self._decode_graphed(
-                    logits, cache, max_new, temperature, stop_ids, deadline
-                )
-            elif use_async:
-                out_ids = self._decode_async(
-                    logits, cache, max_new, temperature, stop_ids, deadline
+                    logits, cache, max_new, temperature, stop_ids, deadline,
+                    use_graph=use_graph,
                 )
             else:
                 out_ids = self._decode_sync(
@@ -207,16 +209,18 @@ class LocalEngine:
         return out_ids
 
     def _decode_graphed(self, logits, cache, max_new, temperature, stop_ids,
-                        deadline) -> list[int]:
-        """HIP-graph decode: the whole per-token step (sample -> 32-layer
-        forward -> logits copy -> position bump) is captured once and
-        replayed per token.
+                        deadline, use_graph: bool = True) -> list[int]:
+        """Device-state decode loop; HIP-graph capture optional.
 
-        Eager decode is launch-gap bound (~420 kernel launches/token,
-        rocprof: dispatch span ~2x kernel busy); graph replay collapses that
-        to one hipGraphLaunch. All dynamic state (position, RNG, token ids)
-        lives in device words read in-kernel, so the captured launch
-        geometry is replay-invariant.
+        The whole per-token step (sample -> n-layer forward -> logits copy
+        -> position bump) keeps all dynamic state (position, RNG, token ids)
+        in device words read in-kernel. With use_graph the step is captured
+        once and replayed per token — eager decode is launch-gap bound
+        (~420 kernel launches/token; rocprof: dispatch span ~2x kernel
+        busy), graph replay collapses that to one hipGraphLaunch. Without
+        use_graph (ADVSPEC_NO_GRAPH, short generations) the SAME step runs
+        eagerly, so both modes execute identical kernels with identical
+        split geometry and produce bitwise-identical tokens.
         """
         from ..ops import _load_hip
 
@@ -229,7 +233,7 @@ class LocalEngine:
         # nothing in the combine).
         max_total = cache.max_seq
 
-        gs = self._graph_state
+        gs = self._graph_state if use_graph else None
         key = (id(cache), cache.max_seq, max_new, float(temperature))
         if gs is not None and gs["key"] == key:
             # Round k+1 on a warm engine: reset the device-side state words
@@ -265,35 +269,39 @@ class LocalEngine:
                 logits_buf.copy_(lg.reshape(-1))
                 hip.bump(pos_state, step_state)
 
-            # warm up on a side stream (torch graph-capture requirement);
-            # these are REAL decode steps — tokens land in tok_hist[0..1].
-            warm = min(2, max_new)
-            s = torch.cuda.Stream(device=dev)
-            s.wait_stream(torch.cuda.current_stream(dev))
-            with torch.cuda.stream(s):
-                for _ in range(warm):
-                    step()
-            torch.cuda.current_stream(dev).wait_stream(s)
+            warm = 0
+            graph = None
+            if use_graph:
+                # warm up on a side stream (torch graph-capture requirement);
+                # these are REAL decode steps — tokens land in tok_hist[0..1].
+                warm = min(2, max_new)
+                s = torch.cuda.Stream(device=dev)
+                s.wait_stream(torch.cuda.current_stream(dev))
+                with torch.cuda.stream(s):
+                    for _ in range(warm):
+                        step()
+                torch.cuda.current_stream(dev).wait_stream(s)
 
-            # thread_local capture mode + a process-wide capture lock: other
-            # opponents' threads keep replaying their own graphs on their own
-            # streams while this engine captures (global mode would abort the
-            # capture on any concurrent allocator traffic).
-            # capture on the ENGINE's stream (not torch's shared default
-            # capture stream): the fused split-K kernels use per-stream
-            # scratch, so two graphs captured on one stream would share
-            # buffers and race when replayed concurrently.
-            graph = torch.cuda.CUDAGraph()
-            with _CAPTURE_LOCK:
-                with torch.cuda.graph(graph, stream=self.stream,
-                                      capture_error_mode="thread_local"):
-                    step()
-            self._graph_state = {
-                "key": key, "graph": graph, "pos_state": pos_state,
-                "step_state": step_state, "rng_state": rng_state,
-                "tok_hist": tok_hist, "tok_slot": tok_slot,
-                "logits_buf": logits_buf,
-            }
+                # thread_local capture mode + a process-wide capture lock:
+                # other opponents' threads keep replaying their own graphs on
+                # their own streams while this engine captures (global mode
+                # would abort the capture on any concurrent allocator
+                # traffic). Capture on the ENGINE's stream (not torch's
+                # shared default capture stream): the decode-attention kernel
+                # uses per-stream scratch, so two graphs captured on one
+                # stream would share buffers and race when replayed
+                # concurrently.
+                graph = torch.cuda.CUDAGraph()
+                with _CAPTURE_LOCK:
+                    with torch.cuda.graph(graph, stream=self.stream,
+                                          capture_error_mode="thread_local"):
+                        step()
+                self._graph_state = {
+                    "key": key, "graph": graph, "pos_state": pos_state,
+                    "step_state": step_state, "rng_state": rng_state,
+                    "tok_hist": tok_hist, "tok_slot": tok_slot,
+                    "logits_buf": logits_buf,
+                }
 
         CHECK = 32
         done = False
@@ -320,8 +328,12 @@ class LocalEngine:
 
         while i < max_new and not done:
             n = min(CHECK, max_new - i)
-            for _ in range(n):
-                graph.replay()
+            if graph is not None:
+                for _ in range(n):
+                    graph.replay()
+            else:
+                for _ in range(n):
+                    step()
             i += n
             done = scan_until(i)
             if time.monotonic() > deadline:
@@ -331,47 +343,6 @@ class LocalEngine:
 
         cache.seq_len = prompt_len + i  # device-side pos advanced i times
         return out_ids
-
-    def _decode_async(self, logits, cache, max_new, temperature, stop_ids,
-                      deadline) -> list[int]:
-        """GPU decode loop with on-device sampling.
-
-        The sampled token id never round-trips to the host inside the loop:
-        the fused sampling kernel writes into a device buffer that feeds the
-        next step's embedding lookup. Stop conditions (eos/eot ids, the
-        [/SPEC] close tag, the deadline) are checked every CHECK tokens with
-        one small D2H copy — ~16x fewer synchronizations than per-token.
-        """
-        from ..ops import _load_hip
-
-        hip = _load_hip()
-        CHECK = 16
-        tok_buf = torch.empty(max_new, dtype=torch.int32, device=self.device)
-        out_ids: list[int] = []
-        tail = ""
-        done = False
-        i = 0
-        while i < max_new and not done:
-            hip.sample_to(logits, temperature, self._next_seed(), tok_buf, i)
-            if i < max_new - 1:
-                logits = self.model.decode_one(tok_buf[i], cache)
-            i += 1
-            if i % CHECK == 0 or i == max_new:
-                chunk = tok_buf[i - ((i - 1) % CHECK + 1) : i].cpu().tolist()
-                for tok in chunk:
-                    if tok in stop_ids:
-                        done = True
-                        break
-                    out_ids.append(tok)
-                    if 0 <= tok < 256:
-                        tail = (tail + chr(tok))[-16:]
-                        if tail.endswith(_STOP_SUBSTR):
-                            done = True
-                            break
-                if time.monotonic() > deadline:
-                    done = True
-        return out_ids
-
 
 _ENGINES: dict[tuple, LocalEngine] = {}
 _ENGINES_LOCK = threading.Lock()
