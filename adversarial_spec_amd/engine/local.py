@@ -291,9 +291,18 @@ class LocalEngine:
                 # uses per-stream scratch, so two graphs captured on one
                 # stream would share buffers and race when replayed
                 # concurrently.
+                # Capture on a FRESH dedicated stream, never the engine's
+                # live eager stream: capturing on a stream that also carries
+                # eager work corrupts the caching allocator's bookkeeping
+                # (measured: later engines' prefills read aliased memory ->
+                # NaN logits, flaky by allocation layout). The decode
+                # attention scratch is keyed by the KV-cache pointer, not the
+                # stream, so the capture-stream choice carries no aliasing
+                # risk.
                 graph = torch.cuda.CUDAGraph()
                 with _CAPTURE_LOCK:
-                    with torch.cuda.graph(graph, stream=self.stream,
+                    with torch.cuda.graph(graph,
+                                          stream=torch.cuda.Stream(device=dev),
                                           capture_error_mode="thread_local"):
                         step()
                 self._graph_state = {

@@ -51,14 +51,14 @@ static hipStream_t cur_stream() {
 }
 
 // --------------------------------------------------------------------------
-// Per-stream persistent workspace for the split-KV decode attention.
-// Kernels on one stream serialize, so one scratch set per stream is
-// race-free: distinct engines run on distinct streams, and a HIP graph
-// captured on a stream bakes THAT stream's buffers (torch's stream pool is
-// 32/device and engines are few, so capture streams are not shared between
-// concurrently-replaying graphs). Grown buffers retire the old tensor into
-// a keep-alive list because a previously captured graph may still replay
-// with the old pointer.
+// Persistent workspace for the split-KV decode attention, keyed by the KV
+// CACHE pointer: one engine owns one cache, and the engine's generate lock
+// serializes all work touching it (eager kernels and graph replays alike),
+// so a per-cache scratch set is race-free — and unlike stream-keyed
+// scratch it cannot alias when torch's 32-entry stream pool hands two
+// engines the same underlying stream. Grown buffers retire the old tensor
+// into a keep-alive list because a previously captured graph may still
+// replay with the old pointer.
 struct StreamWS {
   torch::Tensor attn_ws;   // fp32 scratch: decode-attention m/l/acc
   std::vector<torch::Tensor> retired;
@@ -324,7 +324,7 @@ torch::Tensor attn_decode_paged(torch::Tensor q, torch::Tensor kc,
   float* ws;
   {
     std::lock_guard<std::mutex> lk(g_ws_mu);
-    auto& w = g_ws[(void*)stream];
+    auto& w = g_ws[(void*)kc.data_ptr()];
     ws = ws_f32(w.attn_ws, w.retired, khnsg * (2 + hd), q.device());
   }
   launch_attn_decode_split(uptr(qc), uptr(kc), uptr(vc),
@@ -400,7 +400,7 @@ torch::Tensor attn_decode_paged_ds(torch::Tensor q, torch::Tensor kc,
   float* ws;
   {
     std::lock_guard<std::mutex> lk(g_ws_mu);
-    auto& w = g_ws[(void*)stream];
+    auto& w = g_ws[(void*)kc.data_ptr()];
     ws = ws_f32(w.attn_ws, w.retired, khnsg * (2 + hd), q.device());
   }
   launch_attn_decode_split(uptr(qc), uptr(kc), uptr(vc),

@@ -88,14 +88,24 @@ class TestGenerationGPU:
         assert a[0] == b[0]
 
     def test_graph_decode_matches_eager_greedy(self, monkeypatch):
-        """HIP-graph decode must emit the same greedy tokens as the eager
-        async loop (same kernels, same state)."""
+        """HIP-graph decode must emit the same greedy TOKEN IDS as the eager
+        device-state loop (same kernels, same geometry, same state)."""
         eng = LocalEngine({"name": "g3", "arch": "debug-1b"}, device=DEV)
+        captured = []
+        orig = eng.tokenizer.decode
+
+        def capture(ids):
+            captured.append(list(ids))
+            return orig(ids)
+
+        monkeypatch.setattr(eng.tokenizer, "decode", capture)
         monkeypatch.setenv("ADVSPEC_NO_GRAPH", "1")
-        a = eng.generate("sys", "graph parity prompt", max_tokens=24,
-                         temperature=0.0, timeout=300)
+        eng.generate("sys", "graph parity prompt", max_tokens=24,
+                     temperature=0.0, timeout=300)
         monkeypatch.delenv("ADVSPEC_NO_GRAPH")
-        b = eng.generate("sys", "graph parity prompt", max_tokens=24,
-                         temperature=0.0, timeout=300)
-        assert a[0] == b[0], (a[0][:80], b[0][:80])
+        eng.generate("sys", "graph parity prompt", max_tokens=24,
+                     temperature=0.0, timeout=300)
+        a, b = captured
+        div = next((i for i, (p, q) in enumerate(zip(a, b)) if p != q), None)
+        assert a == b, f"diverge at {div}: eager={a} graph={b}"
         assert b[2] > 0
