@@ -151,6 +151,10 @@ class LocalEngine:
         tokens = torch.tensor(ids, device=self.device, dtype=torch.long)
         with timer.phase("prefill"):
             logits = self.model.prefill(tokens, cache)
+        if os.environ.get("ADVSPEC_DEBUG_NAN"):
+            n = int(torch.isnan(logits.float()).sum().item())
+            print(f"[nan-debug] {self.name} prefill({len(ids)}): {n} NaNs",
+                  flush=True)
 
         out_ids: list[int] = []
         with timer.phase("decode"):
@@ -281,6 +285,11 @@ class LocalEngine:
                     for _ in range(warm):
                         step()
                 torch.cuda.current_stream(dev).wait_stream(s)
+                if os.environ.get("ADVSPEC_DEBUG_NAN"):
+                    print(f"[nan-debug] {self.name} warm toks:",
+                          tok_hist[:warm].cpu().tolist(), "logits_buf NaNs:",
+                          int(torch.isnan(logits_buf.float()).sum().item()),
+                          flush=True)
 
                 # thread_local capture mode + a process-wide capture lock:
                 # other opponents' threads keep replaying their own graphs on

@@ -25,6 +25,17 @@ from typing import Optional
 
 import torch
 
+# hipBLASLt's skinny-m bf16 GEMMs intermittently return garbage (~1e33
+# finite values) on first use in recycled-memory states (round-1 debugging:
+# L0 qkv amax 4.6e33 from clean inputs, flaky by allocation layout, both
+# prompt-sized m=36 and m=56). Route torch matmuls through rocBLAS until
+# the in-tree MFMA GEMM replaces library GEMMs on the prefill path.
+if torch.cuda.is_available():  # pragma: no cover - GPU only
+    try:
+        torch.backends.cuda.preferred_blas_library("cublas")
+    except Exception:
+        pass
+
 from .. import ops
 from ..ops import torch_ref
 from .config import LlamaConfig
@@ -249,8 +260,8 @@ class LlamaModel:
         resid = self.embed[tokens]  # [t, d]
 
         # batch-1 decode projections stream weights through the hand-written
-        # GEMV kernel; prefill (t>1) goes through hipBLASLt.
-        mm = ops.gemv if t == 1 else (lambda a, b: a @ b)
+        # GEMV kernel; prefill (t>1) uses the in-tree tiled MFMA GEMM.
+        mm = ops.gemv if t == 1 else ops.gemm
 
         normed = ops.rmsnorm(resid, self.layers[0].attn_norm, c.norm_eps)
         for i, L in enumerate(self.layers):

@@ -182,6 +182,16 @@ def gemv(x: torch.Tensor, w: torch.Tensor) -> torch.Tensor:
     return x @ w
 
 
+def gemm(x: torch.Tensor, w: torch.Tensor) -> torch.Tensor:
+    """Prefill matmul C = x @ w. On GPU the in-tree tiled MFMA kernel —
+    library GEMMs (hipBLASLt/rocBLAS Tensile kernels) intermittently return
+    garbage on skinny-m bf16 shapes in recycled-memory states, and the
+    in-tree kernel is deterministic and workspace-free. CPU uses torch."""
+    if _on_gpu(x) and w.size(1) % 16 == 0:
+        return _require_hip().gemm(x, w)
+    return x @ w
+
+
 def sample(
     logits: torch.Tensor,
     temperature: float = 0.7,

@@ -44,6 +44,8 @@ __global__ void sample_state_kernel(const ushort_t*, int, float, uint32_t*, int*
 __global__ void bump_kernel(int*, int*);
 void launch_gemv(const ushort_t*, const ushort_t*, ushort_t*, int, int,
                  hipStream_t);
+void launch_gemm(const ushort_t*, const ushort_t*, ushort_t*, int, int, int,
+                 hipStream_t);
 }
 
 static hipStream_t cur_stream() {
@@ -365,6 +367,22 @@ torch::Tensor gemv(torch::Tensor x, torch::Tensor w) {
   return y;
 }
 
+// Tiled MFMA GEMM: C = A @ B, bf16, fp32 accumulation. Replaces library
+// GEMMs on the prefill path (deterministic, workspace-free; see gemm.hip).
+torch::Tensor gemm(torch::Tensor a, torch::Tensor b) {
+  CHECK_BF16_CUDA(a);
+  CHECK_BF16_CUDA(b);
+  TORCH_CHECK(a.dim() == 2 && b.dim() == 2, "gemm: 2-D inputs");
+  TORCH_CHECK(a.size(1) == b.size(0), "gemm: K mismatch");
+  auto ac = a.contiguous();
+  auto bc = b.contiguous();
+  const int M = ac.size(0), K = ac.size(1), N = bc.size(1);
+  TORCH_CHECK(N % 16 == 0, "gemm: N % 16 == 0");
+  auto c = torch::empty({M, N}, ac.options());
+  launch_gemm(uptr(ac), uptr(bc), uptr_mut(c), M, N, K, cur_stream());
+  return c;
+}
+
 int64_t sample(torch::Tensor logits, double temp, double top_p, int64_t seed) {
   CHECK_BF16_CUDA(logits);
   TORCH_CHECK(top_p >= 1.0, "kernel sample handles top_p == 1 (nucleus is a cold path)");
@@ -442,6 +460,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("sample", &sample, "fused temperature softmax sample");
   m.def("sample_to", &sample_to, "async on-device sample into out[idx]");
   m.def("gemv", &gemv, "batch-1 decode GEMV (weight streaming)");
+  m.def("gemm", &gemm, "tiled MFMA GEMM (bf16, fp32 accum)");
   m.def("rope_inplace_ds", &rope_inplace_ds, "graph-mode RoPE (device pos)");
   m.def("rope_kv", &rope_kv, "fused RoPE + paged KV scatter",
         py::arg("q"), py::arg("k"), py::arg("v"), py::arg("cost"),

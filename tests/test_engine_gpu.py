@@ -81,11 +81,21 @@ class TestGenerationGPU:
         assert in_tok > 0 and out_tok > 0
         assert tm["prefill"] > 0 and tm["decode"] > 0
 
-    def test_greedy_deterministic_gpu(self):
+    def test_greedy_deterministic_gpu(self, monkeypatch):
         eng = LocalEngine({"name": "g2", "arch": "debug-1b"}, device=DEV)
-        a = eng.generate("s", "u", max_tokens=12, temperature=0.0, timeout=300)
-        b = eng.generate("s", "u", max_tokens=12, temperature=0.0, timeout=300)
-        assert a[0] == b[0]
+        captured = []
+        orig = eng.tokenizer.decode
+
+        def capture(ids):
+            captured.append(list(ids))
+            return orig(ids)
+
+        monkeypatch.setattr(eng.tokenizer, "decode", capture)
+        eng.generate("s", "u", max_tokens=12, temperature=0.0, timeout=300)
+        eng.generate("s", "u", max_tokens=12, temperature=0.0, timeout=300)
+        a, b = captured
+        div = next((i for i, (p, q) in enumerate(zip(a, b)) if p != q), None)
+        assert a == b, f"diverge at {div}: first={a} second={b}"
 
     def test_graph_decode_matches_eager_greedy(self, monkeypatch):
         """HIP-graph decode must emit the same greedy TOKEN IDS as the eager

@@ -283,3 +283,29 @@ class TestRopeKV:
         assert torch.equal(q2.cpu(), q_want.cpu())
         assert torch.equal(kc.cpu(), kc2.cpu())
         assert torch.equal(vc.cpu(), vc2.cpu())
+
+
+class TestGemm:
+    @pytest.mark.parametrize("m", [3, 36, 56, 128, 300])
+    def test_vs_torch_fp32(self, m):
+        """In-tree MFMA GEMM vs fp32 torch on model-shaped projections."""
+        k, n = 512, 1024
+        a = _bf(torch.randn(m, k)).to(DEV)
+        b = _bf(torch.randn(k, n) * 0.05).to(DEV)
+        got = ops.gemm(a, b)
+        want = a.float().cpu() @ b.float().cpu()
+        _assert_close(got, want, atol=3e-2, name=f"gemm m={m}")
+
+    def test_k_edge_and_bigger(self):
+        a = _bf(torch.randn(40, 72)).to(DEV)   # K not multiple of 32
+        b = _bf(torch.randn(72, 256) * 0.05).to(DEV)
+        got = ops.gemm(a, b)
+        want = a.float().cpu() @ b.float().cpu()
+        _assert_close(got, want, atol=3e-2, name="gemm k-edge")
+
+    def test_deterministic(self):
+        a = _bf(torch.randn(77, 2048)).to(DEV)
+        b = _bf(torch.randn(2048, 4096) * 0.02).to(DEV)
+        x = ops.gemm(a, b)
+        for _ in range(3):
+            assert torch.equal(ops.gemm(a, b), x)
