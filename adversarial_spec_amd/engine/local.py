@@ -167,7 +167,10 @@ class LocalEngine:
                 # same device-state step either way; capture only pays off
                 # for longer generations (ADVSPEC_NO_GRAPH forces eager —
                 # same kernels, bitwise-identical tokens)
+                # TP decode stays eager: capturing RCCL collectives in HIP
+                # graphs is not supported reliably on this stack.
                 use_graph = (max_new >= 8
+                             and self.model.tp is None
                              and not os.environ.get("ADVSPEC_NO_GRAPH"))
                 out_ids = self._decode_graphed(
                     logits, cache, max_new, temperature, stop_ids, deadline,
@@ -247,14 +250,14 @@ class LocalEngine:
             rng_state = gs["rng_state"]
             tok_hist = gs["tok_hist"]
             tok_slot = gs["tok_slot"]
-            logits_buf = gs["logits_buf"]
+            W = gs["ws"]
             graph = gs["graph"]
             pos_state.fill_(prompt_len)
             step_state.zero_()
             rng_state.fill_(self._next_seed() | 1)
             tok_hist.fill_(-1)
             tok_slot.zero_()
-            logits_buf.copy_(logits.reshape(-1))
+            W.logits.copy_(logits.reshape(1, -1))
             warm = 0
         else:
             pos_state = torch.tensor([prompt_len], dtype=torch.int32, device=dev)
@@ -263,14 +266,17 @@ class LocalEngine:
                                      device=dev)
             tok_hist = torch.full((max_new + 2,), -1, dtype=torch.int32, device=dev)
             tok_slot = torch.zeros(1, dtype=torch.int32, device=dev)
-            logits_buf = logits.reshape(-1).contiguous().clone()
+            W = self.model.new_decode_ws()
+            W.logits.copy_(logits.reshape(1, -1))
+            logits_buf = W.logits.view(-1)
 
             def step():
+                # ZERO allocations (capture-safe): sample from W.logits,
+                # feed the id through W.tok_long, forward writes W.logits.
                 hip.sample_state(logits_buf, temperature, rng_state, tok_hist,
                                  step_state, tok_slot)
-                lg = self.model.decode_one_graph(tok_slot, cache, pos_state,
-                                                 max_total)
-                logits_buf.copy_(lg.reshape(-1))
+                W.tok_long.copy_(tok_slot.view(1))
+                self.model.decode_step_ws(cache, pos_state, max_total, W)
                 hip.bump(pos_state, step_state)
 
             warm = 0
@@ -308,18 +314,26 @@ class LocalEngine:
                 # attention scratch is keyed by the KV-cache pointer, not the
                 # stream, so the capture-stream choice carries no aliasing
                 # risk.
-                graph = torch.cuda.CUDAGraph()
-                with _CAPTURE_LOCK:
-                    with torch.cuda.graph(graph,
-                                          stream=torch.cuda.Stream(device=dev),
-                                          capture_error_mode="thread_local"):
-                        step()
-                self._graph_state = {
-                    "key": key, "graph": graph, "pos_state": pos_state,
-                    "step_state": step_state, "rng_state": rng_state,
-                    "tok_hist": tok_hist, "tok_slot": tok_slot,
-                    "logits_buf": logits_buf,
-                }
+                if os.environ.get("ADVSPEC_DBG_NOCAPTURE"):
+                    graph = None  # debug: warm ran, decode eagerly
+                else:
+                    mode = os.environ.get("ADVSPEC_DBG_CAPMODE", "thread_local")
+                    cap_stream = (None if os.environ.get("ADVSPEC_DBG_DEFSTREAM")
+                                  else torch.cuda.Stream(device=dev))
+                    graph = torch.cuda.CUDAGraph()
+                    with _CAPTURE_LOCK:
+                        with torch.cuda.graph(graph, stream=cap_stream,
+                                              capture_error_mode=mode):
+                            step()
+                if os.environ.get("ADVSPEC_DBG_NOREPLAY"):
+                    graph = None  # debug: captured but decode eagerly
+                if graph is not None:
+                    self._graph_state = {
+                        "key": key, "graph": graph, "pos_state": pos_state,
+                        "step_state": step_state, "rng_state": rng_state,
+                        "tok_hist": tok_hist, "tok_slot": tok_slot,
+                        "ws": W,
+                    }
 
         CHECK = 32
         done = False

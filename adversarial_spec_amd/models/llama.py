@@ -77,6 +77,38 @@ class PagedKVCache:
         self.seq_len = 0
 
 
+class DecodeWorkspace:
+    """Preallocated buffers for the single-token decode step.
+
+    The captured HIP-graph step must perform ZERO torch allocations:
+    allocator traffic during stream capture corrupts the caching
+    allocator's bookkeeping on this torch/ROCm combo (later engines'
+    freshly-allocated prefill buffers alias, producing flaky NaN logits —
+    round-1 debugging). A static workspace also removes per-step allocator
+    overhead from eager decode.
+    """
+
+    def __init__(self, model: "LlamaModel") -> None:
+        c = model.config
+        dev, dt = model.device, model.dtype
+        h, kh, hd = c.n_heads, c.n_kv_heads, c.head_dim
+
+        def mk(*shape):
+            return torch.empty(*shape, device=dev, dtype=dt)
+
+        self.resid = mk(1, c.dim)
+        self.resid2 = mk(1, c.dim)
+        self.normed = mk(1, c.dim)
+        self.qkv = mk(1, (h + 2 * kh) * hd)
+        self.attn = mk(h, hd)
+        self.attn_out = mk(1, c.dim)
+        self.gu = mk(1, 2 * c.ffn_dim)
+        self.act = mk(1, c.ffn_dim)
+        self.mlp_out = mk(1, c.dim)
+        self.logits = mk(1, c.vocab_size)
+        self.tok_long = torch.empty(1, dtype=torch.long, device=dev)
+
+
 class LlamaModel:
     """Decoder-only transformer (RMSNorm / RoPE / GQA / SwiGLU)."""
 
@@ -319,6 +351,47 @@ class LlamaModel:
         logits = self._forward(tok, cache, cache.seq_len)
         cache.seq_len += 1
         return logits
+
+    def new_decode_ws(self) -> DecodeWorkspace:
+        return DecodeWorkspace(self)
+
+    def decode_step_ws(self, cache: PagedKVCache, pos_state: torch.Tensor,
+                       max_seq_bound: int, W: DecodeWorkspace) -> torch.Tensor:
+        """Allocation-free single-token decode: token id in W.tok_long,
+        logits written into W.logits. Safe to capture in a HIP graph (all
+        dynamic state in device words, zero allocator traffic)."""
+        c = self.config
+        h, kh, hd = c.n_heads, c.n_kv_heads, c.head_dim
+        torch.index_select(self.embed, 0, W.tok_long, out=W.resid)
+        ops.rmsnorm(W.resid, self.layers[0].attn_norm, c.norm_eps,
+                    out=W.normed)
+        for i, L in enumerate(self.layers):
+            ops.gemv(W.normed, L.wqkv, out=W.qkv)
+            q = W.qkv[:, : h * hd].view(1, h, hd)
+            k = W.qkv[:, h * hd : (h + kh) * hd].view(1, kh, hd)
+            v = W.qkv[:, (h + kh) * hd :].view(1, kh, hd)
+            ops.rope_kv(q, k, v, self.cos, self.sin, cache.k[i], cache.v[i],
+                        cache.page_table, 0, pos_state=pos_state)
+            ops.attn_decode_paged(
+                q[0], cache.k[i], cache.v[i], cache.page_table,
+                max_seq_bound, self.scale, pos_state=pos_state, out=W.attn,
+            )
+            ops.gemv(W.attn.view(1, h * hd), L.wo, out=W.attn_out)
+            if self.tp is not None and self.tp.size > 1:
+                self.tp.all_reduce_(W.attn_out)
+            ops.add_rmsnorm(W.resid, W.attn_out, L.mlp_norm, c.norm_eps,
+                            out_resid=W.resid2, out_y=W.normed)
+            ops.gemv(W.normed, L.w_gate_up, out=W.gu)
+            ops.swiglu(W.gu[:, : c.ffn_dim], W.gu[:, c.ffn_dim :], out=W.act)
+            ops.gemv(W.act, L.w_down, out=W.mlp_out)
+            if self.tp is not None and self.tp.size > 1:
+                self.tp.all_reduce_(W.mlp_out)
+            nxt = (self.layers[i + 1].attn_norm if i + 1 < c.n_layers
+                   else self.final_norm)
+            ops.add_rmsnorm(W.resid2, W.mlp_out, nxt, c.norm_eps,
+                            out_resid=W.resid, out_y=W.normed)
+        ops.gemv(W.normed, self.lm_head, out=W.logits)
+        return W.logits
 
     def decode_one_graph(self, tok_slot: torch.Tensor, cache: PagedKVCache,
                          pos_state: torch.Tensor, max_seq_bound: int) -> torch.Tensor:

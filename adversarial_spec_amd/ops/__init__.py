@@ -59,17 +59,20 @@ def _on_gpu(x: torch.Tensor) -> bool:
 # Public ops. Shapes documented in torch_ref (the contract is identical).
 # --------------------------------------------------------------------------
 
-def rmsnorm(x: torch.Tensor, w: torch.Tensor, eps: float) -> torch.Tensor:
+def rmsnorm(x: torch.Tensor, w: torch.Tensor, eps: float,
+            out: Optional[torch.Tensor] = None) -> torch.Tensor:
     if _on_gpu(x):
-        return _require_hip().rmsnorm(x, w, eps)
+        return _require_hip().rmsnorm(x, w, eps, out)
     return torch_ref.rmsnorm(x, w, eps)
 
 
 def add_rmsnorm(
-    resid: torch.Tensor, delta: torch.Tensor, w: torch.Tensor, eps: float
+    resid: torch.Tensor, delta: torch.Tensor, w: torch.Tensor, eps: float,
+    out_resid: Optional[torch.Tensor] = None,
+    out_y: Optional[torch.Tensor] = None,
 ) -> Tuple[torch.Tensor, torch.Tensor]:
     if _on_gpu(resid):
-        return _require_hip().add_rmsnorm(resid, delta, w, eps)
+        return _require_hip().add_rmsnorm(resid, delta, w, eps, out_resid, out_y)
     return torch_ref.add_rmsnorm(resid, delta, w, eps)
 
 
@@ -86,9 +89,10 @@ def rope(
     return torch_ref.rope(q, k, cos, sin, pos0)
 
 
-def swiglu(gate: torch.Tensor, up: torch.Tensor) -> torch.Tensor:
+def swiglu(gate: torch.Tensor, up: torch.Tensor,
+           out: Optional[torch.Tensor] = None) -> torch.Tensor:
     if _on_gpu(gate):
-        return _require_hip().swiglu(gate, up)
+        return _require_hip().swiglu(gate, up, out)
     return torch_ref.swiglu(gate, up)
 
 
@@ -116,6 +120,7 @@ def attn_decode_paged(
     seq_len: int,
     scale: Optional[float] = None,
     pos_state: Optional[torch.Tensor] = None,
+    out: Optional[torch.Tensor] = None,
 ) -> torch.Tensor:
     if _on_gpu(q):
         import math
@@ -125,7 +130,8 @@ def attn_decode_paged(
             # graph mode: true length = *pos_state + 1 in-kernel; seq_len is
             # the static max bound sizing the split geometry.
             return _require_hip().attn_decode_paged_ds(
-                q, k_cache, v_cache, page_table, pos_state, seq_len, s
+                q, k_cache, v_cache, page_table, pos_state, seq_len, s,
+                out,
             )
         return _require_hip().attn_decode_paged(q, k_cache, v_cache, page_table, seq_len, s)
     return torch_ref.attn_decode_paged(q, k_cache, v_cache, page_table, seq_len, scale)
@@ -173,12 +179,13 @@ def rope_kv(
     return q2, k2
 
 
-def gemv(x: torch.Tensor, w: torch.Tensor) -> torch.Tensor:
+def gemv(x: torch.Tensor, w: torch.Tensor,
+         out: Optional[torch.Tensor] = None) -> torch.Tensor:
     """Batch-1 matmul y = x @ w. On GPU, a hand-written weight-streaming
     kernel (decode's dominant cost is reading w once from HBM3E); falls
     back to torch.matmul when the column count is not 64-aligned."""
     if _on_gpu(x) and w.size(1) % 64 == 0 and hip_available():
-        return _require_hip().gemv(x, w)
+        return _require_hip().gemv(x, w, out)
     return x @ w
 
 

@@ -65,6 +65,13 @@ struct StreamWS {
   torch::Tensor attn_ws;   // fp32 scratch: decode-attention m/l/acc
   std::vector<torch::Tensor> retired;
 };
+// debug knob: opt out of workspace retention (fresh scratch per call)
+static bool ws_fresh() {
+  static int v = -1;
+  if (v < 0) v = getenv("ADVSPEC_DBG_WSFRESH") ? 1 : 0;
+  return v == 1;
+}
+
 static std::mutex g_ws_mu;
 static std::unordered_map<void*, StreamWS> g_ws;
 
@@ -91,7 +98,8 @@ static ushort_t* uptr_mut(torch::Tensor& t) {
 
 // --------------------------------------------------------------------------
 
-torch::Tensor rmsnorm(torch::Tensor x, torch::Tensor w, double eps) {
+torch::Tensor rmsnorm(torch::Tensor x, torch::Tensor w, double eps,
+                      c10::optional<torch::Tensor> out) {
   CHECK_BF16_CUDA(x);
   CHECK_BF16_CUDA(w);
   auto xc = x.contiguous();
@@ -99,16 +107,16 @@ torch::Tensor rmsnorm(torch::Tensor x, torch::Tensor w, double eps) {
   const int d = xc.size(-1);
   const int t = xc.numel() / d;
   TORCH_CHECK(d % 8 == 0, "rmsnorm: d must be a multiple of 8");
-  auto y = torch::empty_like(xc);
+  auto y = out.has_value() ? *out : torch::empty_like(xc);
   rmsnorm_kernel<<<t, 256, 0, cur_stream()>>>(uptr(xc), uptr(wc), uptr_mut(y),
                                               d, (float)eps);
   return y;
 }
 
-std::tuple<torch::Tensor, torch::Tensor> add_rmsnorm(torch::Tensor resid,
-                                                     torch::Tensor delta,
-                                                     torch::Tensor w,
-                                                     double eps) {
+std::tuple<torch::Tensor, torch::Tensor> add_rmsnorm(
+    torch::Tensor resid, torch::Tensor delta, torch::Tensor w, double eps,
+    c10::optional<torch::Tensor> out_resid,
+    c10::optional<torch::Tensor> out_y) {
   CHECK_BF16_CUDA(resid);
   CHECK_BF16_CUDA(delta);
   CHECK_BF16_CUDA(w);
@@ -118,8 +126,8 @@ std::tuple<torch::Tensor, torch::Tensor> add_rmsnorm(torch::Tensor resid,
   const int d = rc.size(-1);
   const int t = rc.numel() / d;
   TORCH_CHECK(d % 8 == 0, "add_rmsnorm: d must be a multiple of 8");
-  auto r_out = torch::empty_like(rc);
-  auto y = torch::empty_like(rc);
+  auto r_out = out_resid.has_value() ? *out_resid : torch::empty_like(rc);
+  auto y = out_y.has_value() ? *out_y : torch::empty_like(rc);
   add_rmsnorm_kernel<<<t, 256, 0, cur_stream()>>>(
       uptr(rc), uptr(dc), uptr(wc), uptr_mut(r_out), uptr_mut(y), d, (float)eps);
   return {r_out, y};
@@ -191,7 +199,8 @@ void rope_kv(torch::Tensor q, torch::Tensor k, torch::Tensor v,
       k.stride(0), v.stride(0), page, pp);
 }
 
-torch::Tensor swiglu(torch::Tensor gate, torch::Tensor up) {
+torch::Tensor swiglu(torch::Tensor gate, torch::Tensor up,
+                     c10::optional<torch::Tensor> out_opt) {
   CHECK_BF16_CUDA(gate);
   CHECK_BF16_CUDA(up);
   TORCH_CHECK(gate.dim() == 2 && up.dim() == 2);
@@ -201,7 +210,7 @@ torch::Tensor swiglu(torch::Tensor gate, torch::Tensor up) {
               "swiglu: gate/up must share a row stride (gate_up halves)");
   const int t = gate.size(0), f = gate.size(1);
   TORCH_CHECK(f % 8 == 0);
-  auto out = torch::empty({t, f}, gate.options());
+  auto out = out_opt.has_value() ? *out_opt : torch::empty({t, f}, gate.options());
   const long nvec = (long)t * f / 8;
   const int blocks = (int)std::min<long>((nvec + 255) / 256, 8192);
   swiglu_kernel<<<blocks, 256, 0, cur_stream()>>>(
@@ -324,7 +333,12 @@ torch::Tensor attn_decode_paged(torch::Tensor q, torch::Tensor kc,
   const long khnsg = (long)kh * n_splits * group;
   auto stream = cur_stream();
   float* ws;
-  {
+  torch::Tensor ws_local;
+  if (ws_fresh()) {
+    ws_local = torch::empty({khnsg * (2 + hd)},
+                            torch::TensorOptions().dtype(at::kFloat).device(q.device()));
+    ws = ws_local.data_ptr<float>();
+  } else {
     std::lock_guard<std::mutex> lk(g_ws_mu);
     auto& w = g_ws[(void*)kc.data_ptr()];
     ws = ws_f32(w.attn_ws, w.retired, khnsg * (2 + hd), q.device());
@@ -352,7 +366,8 @@ void sample_to(torch::Tensor logits, double temp, int64_t seed,
 
 // Decode GEMV: y = x @ w for batch-1 x. Streams w once at HBM rate
 // (hipBLASLt batch-1 measured 0.9-1.7 TB/s; this path targets ~5 TB/s).
-torch::Tensor gemv(torch::Tensor x, torch::Tensor w) {
+torch::Tensor gemv(torch::Tensor x, torch::Tensor w,
+                   c10::optional<torch::Tensor> out_opt) {
   CHECK_BF16_CUDA(x);
   CHECK_BF16_CUDA(w);
   TORCH_CHECK(w.dim() == 2 && w.is_contiguous());
@@ -360,9 +375,11 @@ torch::Tensor gemv(torch::Tensor x, torch::Tensor w) {
   TORCH_CHECK((long)x.numel() == (long)K, "gemv: x numel == K");
   TORCH_CHECK(N % 64 == 0, "gemv: N % 64 == 0");
   auto xc = x.contiguous();
-  auto y = torch::empty(x.dim() == 2 ? std::vector<int64_t>{1, N}
-                                     : std::vector<int64_t>{N},
-                        x.options());
+  auto y = out_opt.has_value()
+               ? *out_opt
+               : torch::empty(x.dim() == 2 ? std::vector<int64_t>{1, N}
+                                           : std::vector<int64_t>{N},
+                              x.options());
   launch_gemv(uptr(xc), uptr(w), uptr_mut(y), K, N, cur_stream());
   return y;
 }
@@ -403,7 +420,8 @@ int64_t sample(torch::Tensor logits, double temp, double top_p, int64_t seed) {
 torch::Tensor attn_decode_paged_ds(torch::Tensor q, torch::Tensor kc,
                                    torch::Tensor vc, torch::Tensor page_table,
                                    torch::Tensor pos_state, int64_t max_seq,
-                                   double scale) {
+                                   double scale,
+                                   c10::optional<torch::Tensor> out_opt) {
   CHECK_BF16_CUDA(q);
   auto qc = q.contiguous();
   const int hq = qc.size(0), hd = qc.size(1);
@@ -412,11 +430,16 @@ torch::Tensor attn_decode_paged_ds(torch::Tensor q, torch::Tensor kc,
   int n_splits = std::max(1, std::min((int)((max_seq + 63) / 64), 1024 / kh));
   int split_len = (int)((max_seq + n_splits - 1) / n_splits + 63) / 64 * 64;
   n_splits = (int)((max_seq + split_len - 1) / split_len);
-  auto out = torch::empty({hq, hd}, qc.options());
+  auto out = out_opt.has_value() ? *out_opt : torch::empty({hq, hd}, qc.options());
   const long khnsg = (long)kh * n_splits * group;
   auto stream = cur_stream();
   float* ws;
-  {
+  torch::Tensor ws_local;
+  if (ws_fresh()) {
+    ws_local = torch::empty({khnsg * (2 + hd)},
+                            torch::TensorOptions().dtype(at::kFloat).device(q.device()));
+    ws = ws_local.data_ptr<float>();
+  } else {
     std::lock_guard<std::mutex> lk(g_ws_mu);
     auto& w = g_ws[(void*)kc.data_ptr()];
     ws = ws_f32(w.attn_ws, w.retired, khnsg * (2 + hd), q.device());
@@ -448,10 +471,15 @@ void bump(torch::Tensor pos_state, torch::Tensor step_state) {
 }
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
-  m.def("rmsnorm", &rmsnorm, "fused RMSNorm (bf16, gfx950)");
-  m.def("add_rmsnorm", &add_rmsnorm, "fused residual add + RMSNorm");
+  m.def("rmsnorm", &rmsnorm, "fused RMSNorm (bf16, gfx950)",
+        py::arg("x"), py::arg("w"), py::arg("eps"),
+        py::arg("out") = py::none());
+  m.def("add_rmsnorm", &add_rmsnorm, "fused residual add + RMSNorm",
+        py::arg("resid"), py::arg("delta"), py::arg("w"), py::arg("eps"),
+        py::arg("out_resid") = py::none(), py::arg("out_y") = py::none());
   m.def("rope_inplace", &rope_inplace, "RoPE (interleaved pairs, table-driven)");
-  m.def("swiglu", &swiglu, "fused silu(gate)*up");
+  m.def("swiglu", &swiglu, "fused silu(gate)*up",
+        py::arg("gate"), py::arg("up"), py::arg("out") = py::none());
   m.def("kv_write", &kv_write, "paged KV scatter");
   m.def("attn_prefill", &attn_prefill, "causal prefill attention");
   m.def("attn_prefill_simple", &attn_prefill_simple, "non-MFMA prefill (anchor)");
@@ -459,7 +487,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("attn_decode_paged", &attn_decode_paged, "paged decode attention");
   m.def("sample", &sample, "fused temperature softmax sample");
   m.def("sample_to", &sample_to, "async on-device sample into out[idx]");
-  m.def("gemv", &gemv, "batch-1 decode GEMV (weight streaming)");
+  m.def("gemv", &gemv, "batch-1 decode GEMV (weight streaming)",
+        py::arg("x"), py::arg("w"), py::arg("out") = py::none());
   m.def("gemm", &gemm, "tiled MFMA GEMM (bf16, fp32 accum)");
   m.def("rope_inplace_ds", &rope_inplace_ds, "graph-mode RoPE (device pos)");
   m.def("rope_kv", &rope_kv, "fused RoPE + paged KV scatter",
@@ -468,7 +497,10 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         py::arg("pos0"), py::arg("pos_state") = py::none());
   m.def("kv_write_ds", &kv_write_ds, "graph-mode KV scatter (device pos)");
   m.def("attn_decode_paged_ds", &attn_decode_paged_ds,
-        "graph-mode paged decode attention (device pos)");
+        "graph-mode paged decode attention (device pos)",
+        py::arg("q"), py::arg("kc"), py::arg("vc"), py::arg("page_table"),
+        py::arg("pos_state"), py::arg("max_seq"), py::arg("scale"),
+        py::arg("out") = py::none());
   m.def("sample_state", &sample_state, "graph-mode on-device sampling");
   m.def("bump", &bump, "graph-mode pos/step bump");
 }
