@@ -72,6 +72,14 @@ class LocalEngine:
         else:
             self.model.init_random()
             self.tokenizer = build_tokenizer(self.config.vocab_size)
+        # ORDER the engine stream after weight init: weights/tables are
+        # initialized on the CREATION stream (usually the default stream),
+        # but all generate() work runs on self.stream. Without this event
+        # dependency the first prefill races the init kernels and reads
+        # unwritten weight memory (garbage bf16 ~1e33 -> NaN logits; flaky
+        # by device load — the round-1 "NaN first prefill" bug).
+        if self.stream is not None:
+            self.stream.wait_stream(torch.cuda.current_stream(self.device))
         self._gen_lock = threading.Lock()
         self._sample_gen = torch.Generator().manual_seed(_seed_from_name(self.name) ^ 0x5EED)
         self._seed_counter = _seed_from_name(self.name) ^ 0x5EED
@@ -282,15 +290,17 @@ class LocalEngine:
             warm = 0
             graph = None
             if use_graph:
-                # warm up on a side stream (torch graph-capture requirement);
-                # these are REAL decode steps — tokens land in tok_hist[0..1].
+                # Warm up ON THE ENGINE STREAM (real decode steps — tokens
+                # land in tok_hist[0..1]); warming loads the decode kernels'
+                # code objects before capture. The usual warm-on-a-side-
+                # stream recipe exists to keep allocator state out of the
+                # capture, but our step is allocation-free — and side-stream
+                # warms measured as the trigger for caching-allocator
+                # aliasing on this torch/ROCm combo (flaky NaN prefills in
+                # LATER engines).
                 warm = min(2, max_new)
-                s = torch.cuda.Stream(device=dev)
-                s.wait_stream(torch.cuda.current_stream(dev))
-                with torch.cuda.stream(s):
-                    for _ in range(warm):
-                        step()
-                torch.cuda.current_stream(dev).wait_stream(s)
+                for _ in range(warm):
+                    step()
                 if os.environ.get("ADVSPEC_DEBUG_NAN"):
                     print(f"[nan-debug] {self.name} warm toks:",
                           tok_hist[:warm].cpu().tolist(), "logits_buf NaNs:",
@@ -314,19 +324,12 @@ class LocalEngine:
                 # attention scratch is keyed by the KV-cache pointer, not the
                 # stream, so the capture-stream choice carries no aliasing
                 # risk.
-                if os.environ.get("ADVSPEC_DBG_NOCAPTURE"):
-                    graph = None  # debug: warm ran, decode eagerly
-                else:
-                    mode = os.environ.get("ADVSPEC_DBG_CAPMODE", "thread_local")
-                    cap_stream = (None if os.environ.get("ADVSPEC_DBG_DEFSTREAM")
-                                  else torch.cuda.Stream(device=dev))
-                    graph = torch.cuda.CUDAGraph()
-                    with _CAPTURE_LOCK:
-                        with torch.cuda.graph(graph, stream=cap_stream,
-                                              capture_error_mode=mode):
-                            step()
-                if os.environ.get("ADVSPEC_DBG_NOREPLAY"):
-                    graph = None  # debug: captured but decode eagerly
+                graph = torch.cuda.CUDAGraph()
+                with _CAPTURE_LOCK:
+                    with torch.cuda.graph(graph,
+                                          stream=torch.cuda.Stream(device=dev),
+                                          capture_error_mode="thread_local"):
+                        step()
                 if graph is not None:
                     self._graph_state = {
                         "key": key, "graph": graph, "pos_state": pos_state,

@@ -65,13 +65,6 @@ struct StreamWS {
   torch::Tensor attn_ws;   // fp32 scratch: decode-attention m/l/acc
   std::vector<torch::Tensor> retired;
 };
-// debug knob: opt out of workspace retention (fresh scratch per call)
-static bool ws_fresh() {
-  static int v = -1;
-  if (v < 0) v = getenv("ADVSPEC_DBG_WSFRESH") ? 1 : 0;
-  return v == 1;
-}
-
 static std::mutex g_ws_mu;
 static std::unordered_map<void*, StreamWS> g_ws;
 
@@ -333,12 +326,7 @@ torch::Tensor attn_decode_paged(torch::Tensor q, torch::Tensor kc,
   const long khnsg = (long)kh * n_splits * group;
   auto stream = cur_stream();
   float* ws;
-  torch::Tensor ws_local;
-  if (ws_fresh()) {
-    ws_local = torch::empty({khnsg * (2 + hd)},
-                            torch::TensorOptions().dtype(at::kFloat).device(q.device()));
-    ws = ws_local.data_ptr<float>();
-  } else {
+  {
     std::lock_guard<std::mutex> lk(g_ws_mu);
     auto& w = g_ws[(void*)kc.data_ptr()];
     ws = ws_f32(w.attn_ws, w.retired, khnsg * (2 + hd), q.device());
@@ -434,12 +422,7 @@ torch::Tensor attn_decode_paged_ds(torch::Tensor q, torch::Tensor kc,
   const long khnsg = (long)kh * n_splits * group;
   auto stream = cur_stream();
   float* ws;
-  torch::Tensor ws_local;
-  if (ws_fresh()) {
-    ws_local = torch::empty({khnsg * (2 + hd)},
-                            torch::TensorOptions().dtype(at::kFloat).device(q.device()));
-    ws = ws_local.data_ptr<float>();
-  } else {
+  {
     std::lock_guard<std::mutex> lk(g_ws_mu);
     auto& w = g_ws[(void*)kc.data_ptr()];
     ws = ws_f32(w.attn_ws, w.retired, khnsg * (2 + hd), q.device());
