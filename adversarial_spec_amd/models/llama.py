@@ -43,12 +43,13 @@ from .config import LlamaConfig
 
 @dataclass
 class LayerWeights:
+    # all projection weights ROW-MAJOR [out, in] (HF layout; x @ W^T)
     attn_norm: torch.Tensor  # [d]
-    wqkv: torch.Tensor  # [d, (h + 2*kh) * hd]
-    wo: torch.Tensor  # [h*hd, d]
+    wqkv: torch.Tensor  # [(h + 2*kh) * hd, d]
+    wo: torch.Tensor  # [d, h*hd]
     mlp_norm: torch.Tensor  # [d]
-    w_gate_up: torch.Tensor  # [d, 2*ffn]
-    w_down: torch.Tensor  # [ffn, d]
+    w_gate_up: torch.Tensor  # [2*ffn, d]
+    w_down: torch.Tensor  # [d, ffn]
 
 
 class PagedKVCache:
@@ -171,14 +172,14 @@ class LlamaModel:
         proj_std = 0.02 / math.sqrt(2 * c.n_layers)
         self.embed = t(c.vocab_size, d)
         self.final_norm = torch.ones(d, device=self.device, dtype=self.dtype)
-        self.lm_head = t(d, c.vocab_size)
+        self.lm_head = t(c.vocab_size, d)
         tp = self.tp
         self.layers = []
         for _ in range(c.n_layers):
-            wqkv = t(d, (c.n_heads + 2 * c.n_kv_heads) * hd)
-            wo = t(c.n_heads * hd, d, std=proj_std)
-            w_gate_up = t(d, 2 * c.ffn_dim)
-            w_down = t(c.ffn_dim, d, std=proj_std)
+            wqkv = t((c.n_heads + 2 * c.n_kv_heads) * hd, d)
+            wo = t(d, c.n_heads * hd, std=proj_std)
+            w_gate_up = t(2 * c.ffn_dim, d)
+            w_down = t(d, c.ffn_dim, std=proj_std)
             if tp is not None and tp.size > 1:
                 from ..parallel.tp import (
                     shard_down,
@@ -241,30 +242,33 @@ class LlamaModel:
         def g(key: str) -> torch.Tensor:
             return tensors[key].to(torch.float32)
 
+        # HF checkpoints store projections [out, in] — exactly our wire
+        # layout, so no transposes: only q/k row permutation for the
+        # interleaved-pair RoPE convention and the qkv / gate-up fusions.
         self.embed = g("model.embed_tokens.weight").to(self.dtype).to(self.device)
         self.final_norm = g("model.norm.weight").to(self.dtype).to(self.device)
         lm = tensors.get("lm_head.weight")
         if lm is None:  # tied embeddings
             lm = tensors["model.embed_tokens.weight"]
-        self.lm_head = lm.to(torch.float32).t().contiguous().to(self.dtype).to(self.device)
+        self.lm_head = lm.to(torch.float32).contiguous().to(self.dtype).to(self.device)
         self.layers = []
         for i in range(c.n_layers):
             p = f"model.layers.{i}."
             wq = perm_rope_rows(g(p + "self_attn.q_proj.weight"), c.n_heads)
             wk = perm_rope_rows(g(p + "self_attn.k_proj.weight"), c.n_kv_heads)
             wv = g(p + "self_attn.v_proj.weight")
-            wqkv = torch.cat([wq, wk, wv], dim=0).t().contiguous()
+            wqkv = torch.cat([wq, wk, wv], dim=0).contiguous()
             w_gate_up = torch.cat(
                 [g(p + "mlp.gate_proj.weight"), g(p + "mlp.up_proj.weight")], dim=0
-            ).t().contiguous()
+            ).contiguous()
             self.layers.append(
                 LayerWeights(
                     attn_norm=g(p + "input_layernorm.weight").to(self.dtype).to(self.device),
                     wqkv=wqkv.to(self.dtype).to(self.device),
-                    wo=g(p + "self_attn.o_proj.weight").t().contiguous().to(self.dtype).to(self.device),
+                    wo=g(p + "self_attn.o_proj.weight").contiguous().to(self.dtype).to(self.device),
                     mlp_norm=g(p + "post_attention_layernorm.weight").to(self.dtype).to(self.device),
                     w_gate_up=w_gate_up.to(self.dtype).to(self.device),
-                    w_down=g(p + "mlp.down_proj.weight").t().contiguous().to(self.dtype).to(self.device),
+                    w_down=g(p + "mlp.down_proj.weight").contiguous().to(self.dtype).to(self.device),
                 )
             )
         return self

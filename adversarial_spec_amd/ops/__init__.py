@@ -181,22 +181,23 @@ def rope_kv(
 
 def gemv(x: torch.Tensor, w: torch.Tensor,
          out: Optional[torch.Tensor] = None) -> torch.Tensor:
-    """Batch-1 matmul y = x @ w. On GPU, a hand-written weight-streaming
-    kernel (decode's dominant cost is reading w once from HBM3E); falls
-    back to torch.matmul when the column count is not 64-aligned."""
-    if _on_gpu(x) and w.size(1) % 64 == 0 and hip_available():
+    """Batch-1 matmul y = x @ w^T, weights ROW-MAJOR [out, in] (HF layout).
+    On GPU, a hand-written weight-streaming kernel (decode's dominant cost
+    is reading w once from HBM3E)."""
+    if _on_gpu(x):
         return _require_hip().gemv(x, w, out)
-    return x @ w
+    return x @ w.t()
 
 
 def gemm(x: torch.Tensor, w: torch.Tensor) -> torch.Tensor:
-    """Prefill matmul C = x @ w. On GPU the in-tree tiled MFMA kernel —
-    library GEMMs (hipBLASLt/rocBLAS Tensile kernels) intermittently return
-    garbage on skinny-m bf16 shapes in recycled-memory states, and the
-    in-tree kernel is deterministic and workspace-free. CPU uses torch."""
-    if _on_gpu(x) and w.size(1) % 16 == 0:
+    """Prefill matmul C = x @ w^T, weights ROW-MAJOR [out, in] (HF layout).
+    On GPU the in-tree tiled MFMA kernel — library GEMMs (hipBLASLt/rocBLAS
+    Tensile kernels) intermittently return garbage on skinny-m bf16 shapes
+    in recycled-memory states, and the in-tree kernel is deterministic and
+    workspace-free. CPU uses torch."""
+    if _on_gpu(x):
         return _require_hip().gemm(x, w)
-    return x @ w
+    return x @ w.t()
 
 
 def sample(

@@ -356,12 +356,13 @@ void sample_to(torch::Tensor logits, double temp, int64_t seed,
 // (hipBLASLt batch-1 measured 0.9-1.7 TB/s; this path targets ~5 TB/s).
 torch::Tensor gemv(torch::Tensor x, torch::Tensor w,
                    c10::optional<torch::Tensor> out_opt) {
+  // y = x @ w^T with w stored row-major [N, K] (HF layout)
   CHECK_BF16_CUDA(x);
   CHECK_BF16_CUDA(w);
   TORCH_CHECK(w.dim() == 2 && w.is_contiguous());
-  const int K = w.size(0), N = w.size(1);
+  const int N = w.size(0), K = w.size(1);
   TORCH_CHECK((long)x.numel() == (long)K, "gemv: x numel == K");
-  TORCH_CHECK(N % 64 == 0, "gemv: N % 64 == 0");
+  TORCH_CHECK(K % 8 == 0, "gemv: K % 8 == 0");
   auto xc = x.contiguous();
   auto y = out_opt.has_value()
                ? *out_opt
@@ -375,14 +376,14 @@ torch::Tensor gemv(torch::Tensor x, torch::Tensor w,
 // Tiled MFMA GEMM: C = A @ B, bf16, fp32 accumulation. Replaces library
 // GEMMs on the prefill path (deterministic, workspace-free; see gemm.hip).
 torch::Tensor gemm(torch::Tensor a, torch::Tensor b) {
+  // C = a @ b^T with b stored row-major [N, K] (HF layout)
   CHECK_BF16_CUDA(a);
   CHECK_BF16_CUDA(b);
   TORCH_CHECK(a.dim() == 2 && b.dim() == 2, "gemm: 2-D inputs");
-  TORCH_CHECK(a.size(1) == b.size(0), "gemm: K mismatch");
+  TORCH_CHECK(a.size(1) == b.size(1), "gemm: K mismatch");
   auto ac = a.contiguous();
   auto bc = b.contiguous();
-  const int M = ac.size(0), K = ac.size(1), N = bc.size(1);
-  TORCH_CHECK(N % 16 == 0, "gemm: N % 16 == 0");
+  const int M = ac.size(0), K = ac.size(1), N = bc.size(0);
   auto c = torch::empty({M, N}, ac.options());
   launch_gemm(uptr(ac), uptr(bc), uptr_mut(c), M, N, K, cur_stream());
   return c;

@@ -1,25 +1,28 @@
-// Tiled MFMA GEMM for CDNA4 (gfx950): C[M,N] = A[M,K] @ B[K,N],
+// Tiled MFMA GEMM for CDNA4 (gfx950): C[M,N] = A[M,K] @ B[N,K]^T,
 // bf16 in/out, fp32 accumulation via v_mfma_f32_16x16x32_bf16.
 //
-// This replaces library GEMMs (hipBLASLt/rocBLAS) on the prefill path.
-// Besides the perf headroom (Tensile's generic kernels measured ~0.9
-// PFLOP/s on these shapes), the libraries' skinny-m bf16 kernels
-// intermittently return garbage (~1e33 finite values) in recycled-memory
-// states — round-1 debugging traced every "NaN prefill" to that. An
-// in-tree GEMM is deterministic, workspace-free and graph-capture safe.
+// Weights are stored ROW-MAJOR [out, in] (= HF checkpoint layout, no
+// load-time transpose), so the A and B tiles stage IDENTICALLY: both are
+// [128 rows][64 k] images whose 16 B fragment reads are contiguous
+// ds_read_b128 — no LDS transpose pass anywhere.
 //
-// Tiling:
-//   block = 256 threads (4 waves), tile M=128 x N=128, K stepped by 32;
-//   each wave owns a 64x64 quadrant = 4x4 MFMA 16x16x32 tiles.
-//   A tile [128][32] staged row-major in LDS (16 B/lane frag reads);
-//   B tile staged TRANSPOSED [128 cols][32 k] with rows padded 32->40
-//   elements so the 16 B frag reads and the staging stores stay off each
-//   other's banks (same trick as the attention kernels' Vt image).
-//   Stores: direct bf16 scalar stores (32 KB/block vs MBs of loads).
+// This replaces library GEMMs (hipBLASLt/rocBLAS) on the prefill path:
+// besides the perf headroom, Tensile's skinny-m bf16 kernels return
+// garbage intermittently in recycled-memory states (round-1 debugging).
+// In-tree, deterministic, workspace-free, graph-capture safe.
 //
-// Edge handling: M/K guarded by zero-padding in LDS, N guarded on load
-// and store, so any (M, K%32==0... actually any K) with N%16==0 works;
-// model shapes are all N%128==0.
+// Structure = the cdna_hip_programming.md "step 3" ladder
+// (128x128 tile, BK=64, double-buffered `global_load_lds` staging,
+// one barrier per K-step, ~900 TF/s at 4096^3 on this chip):
+//   - block 256 threads = 4 waves in a 2x2 quadrant grid, each wave owns
+//     a 64x64 output quadrant = 4x4 MFMA 16x16 tiles;
+//   - staging via __builtin_amdgcn_global_load_lds width 16: LDS stays
+//     lane-linear, the bank-conflict XOR swizzle is applied to the
+//     PER-LANE GLOBAL SOURCE address (chunk ^= row&7 within each 128 B
+//     row — stays inside the cacheline, so coalescing is preserved);
+//   - fragment reads de-swizzle with the same XOR.
+// Edge tiles (M/N/K not tile-aligned) take a guarded register-staging
+// path into the same swizzled image, so the MFMA loop is shared.
 
 #include "common.h"
 
@@ -28,26 +31,77 @@ typedef __attribute__((__vector_size__(4 * sizeof(float)))) float f32x4v;
 
 #define GT_M 128
 #define GT_N 128
-#define GT_K 32
-#define BPITCH 40  // padded row length of the transposed B image
+#define GT_K 64  // elements; one row of the LDS image = 128 B
 
-extern "C" __global__ void __launch_bounds__(256, 2)
-gemm_mfma_kernel(const ushort_t *__restrict__ a, const ushort_t *__restrict__ b,
-                 ushort_t *__restrict__ c_out, int M, int N, int K) {
+DEVINL void glds16(const ushort_t *g, ushort_t *l) {
+  __builtin_amdgcn_global_load_lds(
+      (const __attribute__((address_space(1))) void *)g,
+      (__attribute__((address_space(3))) void *)l, 16, 0, 0);
+}
+
+// Stage one [128][64] bf16 tile via glds. Each wave-instruction covers
+// 1 KiB = 8 rows x 128 B; 4 waves x 4 issues cover the 16 KiB image.
+// src row r, 16B-chunk c is fetched from global chunk (c ^ (r & 7)) so the
+// lane-linear LDS placement realizes the swizzled image.
+DEVINL void stage_glds(const ushort_t *__restrict__ src, long row_stride,
+                       ushort_t *__restrict__ img, int tid) {
+  const int lane = tid & (WAVE - 1);
+  const int wid = tid / WAVE;
+#pragma unroll
+  for (int i = 0; i < 4; ++i) {
+    const int r0 = (wid * 4 + i) * 8;
+    const int row = r0 + (lane >> 3);
+    const int c = lane & 7;
+    const ushort_t *g = src + (size_t)row * row_stride + (size_t)(c ^ (row & 7)) * 8;
+    // LDS dest: wave-uniform base + lane*16 (lane-linear within the 1 KiB)
+    glds16(g, img + (size_t)r0 * GT_K);
+  }
+}
+
+// Guarded register staging for edge tiles: same swizzled image layout.
+// 1024 16 B chunk-pieces (128 rows x 8 chunks), 4 per thread.
+DEVINL void stage_edge(const ushort_t *__restrict__ src, long row_stride,
+                       int rows_left, int k_left, ushort_t *__restrict__ img,
+                       int tid) {
+#pragma unroll
+  for (int p = 0; p < 4; ++p) {
+    const int piece = tid + 256 * p;   // 0..1023
+    const int row = piece >> 3;        // 0..127
+    const int c = piece & 7;           // 16B chunk
+    bf16x8 v;
+    const int gk = c * 8;
+    if (row < rows_left && gk < k_left) {
+      const ushort_t *g = src + (size_t)row * row_stride + gk;
+      if (gk + 8 <= k_left) {
+        v = *(const bf16x8 *)g;
+      } else {
+#pragma unroll
+        for (int j = 0; j < 8; ++j) v.u[j] = (gk + j < k_left) ? g[j] : 0;
+      }
+    } else {
+#pragma unroll
+      for (int j = 0; j < 8; ++j) v.u[j] = 0;
+    }
+    ((bf16x8 *)(img + (size_t)row * GT_K))[c ^ (row & 7)] = v;
+  }
+}
+
+__global__ void __launch_bounds__(256, 2)
+gemm_nt_kernel(const ushort_t *__restrict__ a, const ushort_t *__restrict__ b,
+               ushort_t *__restrict__ c_out, int M, int N, int K) {
   const int tid = threadIdx.x;
   const int lane = tid & (WAVE - 1);
   const int wid = tid / WAVE;
-  const int lrow = lane & 15;   // 0..15: row (A/D) or col (B)
-  const int lhi = lane >> 4;    // 0..3: 8-element k-chunk / 4-row group
+  const int lrow = lane & 15;
+  const int lhi = lane >> 4;
 
-  const int n0 = blockIdx.x * GT_N;   // this block's first output col
-  const int m0 = blockIdx.y * GT_M;   // this block's first output row
-  // wave quadrant within the 128x128 tile: 2x2 of 64x64
-  const int wm = (wid >> 1) * 64;
+  const int n0 = blockIdx.x * GT_N;
+  const int m0 = blockIdx.y * GT_M;
+  const int wm = (wid >> 1) * 64;  // wave quadrant
   const int wn = (wid & 1) * 64;
 
-  __shared__ __attribute__((aligned(16))) ushort_t ldsA[GT_M * GT_K];
-  __shared__ __attribute__((aligned(16))) ushort_t ldsBt[GT_N * BPITCH];
+  __shared__ __attribute__((aligned(16))) ushort_t imgA[2][GT_M * GT_K];
+  __shared__ __attribute__((aligned(16))) ushort_t imgB[2][GT_N * GT_K];
 
   f32x4v acc[4][4];
 #pragma unroll
@@ -55,86 +109,63 @@ gemm_mfma_kernel(const ushort_t *__restrict__ a, const ushort_t *__restrict__ b,
 #pragma unroll
     for (int j = 0; j < 4; ++j) acc[i][j] = (f32x4v){0.f, 0.f, 0.f, 0.f};
 
-  for (int k0 = 0; k0 < K; k0 += GT_K) {
-    // ---- stage A [128][32]: 256 threads x 2 16B pieces ----
-    // piece p covers row (tid + 256*p) / 2... each row = 64 B = 4 pieces;
-    // 128 rows * 4 = 512 pieces, 2 per thread.
-    __syncthreads();  // previous iteration's frag reads done
-#pragma unroll
-    for (int p = 0; p < 2; ++p) {
-      const int piece = tid + 256 * p;
-      const int row = piece >> 2;          // 0..127
-      const int c8 = piece & 3;            // 16B chunk within the 64 B row
-      const int gm = m0 + row;
-      const int gk = k0 + c8 * 8;
-      bf16x8 va;
-      if (gm < M && gk < K) {
-        if (gk + 8 <= K) {
-          va = *(const bf16x8 *)(a + (size_t)gm * K + gk);
-        } else {
-#pragma unroll
-          for (int j = 0; j < 8; ++j)
-            va.u[j] = (gk + j < K) ? a[(size_t)gm * K + gk + j] : 0;
-        }
-      } else {
-#pragma unroll
-        for (int j = 0; j < 8; ++j) va.u[j] = 0;
-      }
-      // XOR swizzle the 16B chunk index by the row's low bits
-      ((bf16x8 *)(ldsA + row * GT_K))[c8 ^ (row & 3)] = va;
-    }
-    // ---- stage Bt [128 cols][40-pitch k]: read B rows coalesced,
-    // scatter transposed (8 scalar stores per 16 B read) ----
-#pragma unroll
-    for (int p = 0; p < 2; ++p) {
-      const int piece = tid + 256 * p;
-      const int krow = piece >> 4;         // 0..31: B row (k index)
-      const int c8 = piece & 15;           // 16B chunk within 128 cols
-      const int gk = k0 + krow;
-      const int gn = n0 + c8 * 8;
-      bf16x8 vb;
-      if (gk < K && gn < N) {
-        if (gn + 8 <= N) {
-          vb = *(const bf16x8 *)(b + (size_t)gk * N + gn);
-        } else {
-#pragma unroll
-          for (int j = 0; j < 8; ++j)
-            vb.u[j] = (gn + j < N) ? b[(size_t)gk * N + gn + j] : 0;
-        }
-      } else {
-#pragma unroll
-        for (int j = 0; j < 8; ++j) vb.u[j] = 0;
-      }
-#pragma unroll
-      for (int j = 0; j < 8; ++j)
-        ldsBt[(size_t)(c8 * 8 + j) * BPITCH + krow] = vb.u[j];
-    }
-    __syncthreads();
+  const int ntiles = (K + GT_K - 1) / GT_K;
+  // block-uniform edge predicates: only edge blocks (and the last K tile
+  // when K % 64 != 0) pay the guarded register-staging path
+  const bool a_edge = (M - m0) < GT_M;
+  const bool b_edge = (N - n0) < GT_N;
 
-    // ---- 16 MFMAs: 4 m-tiles x 4 n-tiles, full K=32 per mfma ----
-    bf16x8v afr[4], bfr[4];
-#pragma unroll
-    for (int i = 0; i < 4; ++i) {
-      const int row = wm + i * 16 + lrow;
-      const bf16x8 t =
-          ((const bf16x8 *)(ldsA + row * GT_K))[lhi ^ (row & 3)];
-#pragma unroll
-      for (int j = 0; j < 8; ++j) afr[i][j] = (short)t.u[j];
+  auto stage = [&](int kt, int buf) {
+    const ushort_t *asrc = a + (size_t)m0 * K + (size_t)kt * GT_K;
+    const ushort_t *bsrc = b + (size_t)n0 * K + (size_t)kt * GT_K;
+    const bool k_edge = (kt + 1) * GT_K > K;
+    if (a_edge || k_edge) {
+      stage_edge(asrc, K, M - m0, K - kt * GT_K, imgA[buf], tid);
+    } else {
+      stage_glds(asrc, K, imgA[buf], tid);
     }
-#pragma unroll
-    for (int j2 = 0; j2 < 4; ++j2) {
-      const int col = wn + j2 * 16 + lrow;
-      const bf16x8 t =
-          *(const bf16x8 *)(ldsBt + (size_t)col * BPITCH + lhi * 8);
-#pragma unroll
-      for (int j = 0; j < 8; ++j) bfr[j2][j] = (short)t.u[j];
+    if (b_edge || k_edge) {
+      stage_edge(bsrc, K, N - n0, K - kt * GT_K, imgB[buf], tid);
+    } else {
+      stage_glds(bsrc, K, imgB[buf], tid);
     }
+  };
+
+  stage(0, 0);
+  __syncthreads();  // drains the glds queue (vmcnt0 inside the barrier)
+
+  for (int kt = 0; kt < ntiles; ++kt) {
+    const int buf = kt & 1;
+    if (kt + 1 < ntiles) stage(kt + 1, buf ^ 1);  // prefetch next tile
+
+    // ---- 2 k-subtiles of 32: 8 frag reads + 16 MFMAs each ----
 #pragma unroll
-    for (int i = 0; i < 4; ++i)
+    for (int s = 0; s < 2; ++s) {
+      bf16x8v afr[4], bfr[4];
 #pragma unroll
-      for (int j2 = 0; j2 < 4; ++j2)
-        acc[i][j2] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-            afr[i], bfr[j2], acc[i][j2], 0, 0, 0);
+      for (int i = 0; i < 4; ++i) {
+        const int row = wm + i * 16 + lrow;
+        const int ch = (s * 4 + lhi) ^ (row & 7);
+        const bf16x8 t = ((const bf16x8 *)(imgA[buf] + (size_t)row * GT_K))[ch];
+#pragma unroll
+        for (int j = 0; j < 8; ++j) afr[i][j] = (short)t.u[j];
+      }
+#pragma unroll
+      for (int j2 = 0; j2 < 4; ++j2) {
+        const int row = wn + j2 * 16 + lrow;
+        const int ch = (s * 4 + lhi) ^ (row & 7);
+        const bf16x8 t = ((const bf16x8 *)(imgB[buf] + (size_t)row * GT_K))[ch];
+#pragma unroll
+        for (int j = 0; j < 8; ++j) bfr[j2][j] = (short)t.u[j];
+      }
+#pragma unroll
+      for (int i = 0; i < 4; ++i)
+#pragma unroll
+        for (int j2 = 0; j2 < 4; ++j2)
+          acc[i][j2] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              afr[i], bfr[j2], acc[i][j2], 0, 0, 0);
+    }
+    __syncthreads();  // compute done + prefetch glds drained
   }
 
   // ---- store C: D frag lane l -> col l%16, rows (l/16)*4 + 0..3 ----
@@ -147,8 +178,8 @@ gemm_mfma_kernel(const ushort_t *__restrict__ a, const ushort_t *__restrict__ b,
 #pragma unroll
       for (int r = 0; r < 4; ++r) {
         const int row = m0 + wm + i * 16 + lhi * 4 + r;
-        if (row < M)
-          c_out[(size_t)row * N + col] = f32_to_bf16(acc[i][j2][r]);
+        if (row >= M) continue;
+        c_out[(size_t)row * N + col] = f32_to_bf16(acc[i][j2][r]);
       }
     }
   }
@@ -157,5 +188,5 @@ gemm_mfma_kernel(const ushort_t *__restrict__ a, const ushort_t *__restrict__ b,
 extern "C" void launch_gemm(const ushort_t *a, const ushort_t *b, ushort_t *c,
                             int M, int N, int K, hipStream_t stream) {
   dim3 grid((N + GT_N - 1) / GT_N, (M + GT_M - 1) / GT_M);
-  gemm_mfma_kernel<<<grid, 256, 0, stream>>>(a, b, c, M, N, K);
+  gemm_nt_kernel<<<grid, 256, 0, stream>>>(a, b, c, M, N, K);
 }

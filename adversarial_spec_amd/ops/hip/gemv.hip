@@ -1,80 +1,69 @@
-// Decode GEMV: y[1,N] = x[1,K] @ W[K,N], bf16 in/out, fp32 accumulation.
+// Decode GEMV: y[1,N] = x[1,K] @ W[N,K]^T, bf16 in/out, fp32 accumulation.
 //
-// Batch-1 decode is bound by streaming the weight matrix once from HBM3E;
-// hipBLASLt's batch-1 kernels measured 0.9-1.7 TB/s on this path (rocprof,
-// profiles/), so the hot decode projections use this hand-written streamer:
-// 16 B/lane coalesced weight reads with an 8-deep k unroll (8 independent
-// loads in flight per thread) and a DIRECT bf16 write — no split-K.
+// Weights are ROW-MAJOR [out, in] (HF layout; same convention as the
+// MFMA GEMM), so each output element is a contiguous row dot-product —
+// the weight matrix streams once from HBM3E in full 256 B row segments.
 //
-// Why no split-K: a cross-block combine (either a second kernel or a
-// last-block-arrives reduction) was measured strictly worse on this chip:
-// the separate combine kernel costs a ~5.5 us execution floor per GEMV
-// (~0.7 ms/token over 33 projections), and the fused last-block variant
-// needs device-scope fences, which on the 8-XCD MI355X trigger cross-L2
-// traffic that slowed the WHOLE device ~5x (profiles/, round 1). Instead
-// each block owns 64 output columns over the FULL K; the debate engine
-// runs 3+ co-resident opponents on separate HIP streams, so the chip is
-// filled by opponent-level concurrency rather than intra-GEMV splits.
-//
-// Tile: one block = 256 threads covers 64 output columns x all of K.
-//   thread t: vec-column (t % 8) (8 bf16 cols), k-lane (t / 8) of 32.
+// Tiling: block = 256 threads (4 waves); each wave owns 4 output rows via
+// 16-lane groups; a lane reads 16 B (8 bf16) per k-step, 4-deep unrolled
+// so >= 4 independent loads are in flight per lane. Cross-lane reduce via
+// 16-lane shfl; no split-K, no workspace (the debate engine runs 3+
+// co-resident opponents on separate HIP streams, so the chip is filled by
+// opponent-level concurrency — and a cross-block combine needs
+// device-scope fences that thrash cross-XCD L2, measured ~5x whole-device
+// slowdown in round 1).
 
 #include "common.h"
+
+// 16-lane-group sum (lanes p, p+1, .., p+15 with stride 1)
+DEVINL float group16_sum(float v) {
+#pragma unroll
+  for (int off = 8; off > 0; off >>= 1) v += __shfl_xor(v, off, WAVE);
+  return v;
+}
 
 extern "C" __global__ void __launch_bounds__(256)
 gemv_kernel(const ushort_t *__restrict__ x, const ushort_t *__restrict__ w,
             ushort_t *__restrict__ y, int K, int N) {
-  const int cb = blockIdx.x;   // column block (64 cols)
-  const int t = threadIdx.x;
-  const int vc = t & 7;        // vec-column 0..7 (8 bf16 each)
-  const int kl = t >> 3;       // k lane 0..31
-  const int c0 = cb * 64;
+  const int lane = threadIdx.x & (WAVE - 1);
+  const int wid = threadIdx.x / WAVE;
+  const int rg = lane >> 4;       // row group within wave: 0..3
+  const int sl = lane & 15;       // k-slice lane: 0..15
+  const int n = blockIdx.x * 16 + wid * 4 + rg;
+  if (n >= N) return;
 
-  float acc[8];
-#pragma unroll
-  for (int j = 0; j < 8; ++j) acc[j] = 0.f;
+  const ushort_t *wr = w + (size_t)n * K;
+  const int nc = K / 8;           // 16 B chunks per row (K % 8 == 0)
 
-  // 8-deep k unroll: 8 independent 16 B weight loads in flight per thread
-  // (4-deep measured latency-bound at ~47% of HBM roofline).
-  int k = kl;
-  for (; k + 224 < K; k += 256) {
-    bf16x8 wv[8];
-    float xv[8];
+  float acc = 0.f;
+  int c = sl;
+  // 4-deep unroll: 4 independent 16 B row reads + 4 x reads in flight
+  for (; c + 48 < nc; c += 64) {
+    bf16x8 wv[4], xv[4];
 #pragma unroll
-    for (int u = 0; u < 8; ++u) {
-      const int kk = k + 32 * u;
-      xv[u] = bf16_to_f32(x[kk]);
-      wv[u] = ((const bf16x8 *)(w + (size_t)kk * N + c0))[vc];
+    for (int u = 0; u < 4; ++u) {
+      wv[u] = ((const bf16x8 *)wr)[c + 16 * u];
+      xv[u] = ((const bf16x8 *)x)[c + 16 * u];
     }
 #pragma unroll
-    for (int u = 0; u < 8; ++u) {
+    for (int u = 0; u < 4; ++u)
 #pragma unroll
-      for (int j = 0; j < 8; ++j) acc[j] += xv[u] * bf16_to_f32(wv[u].u[j]);
-    }
+      for (int j = 0; j < 8; ++j)
+        acc += bf16_to_f32(xv[u].u[j]) * bf16_to_f32(wv[u].u[j]);
   }
-  for (; k < K; k += 32) {
-    const float xv = bf16_to_f32(x[k]);
-    const bf16x8 wv = ((const bf16x8 *)(w + (size_t)k * N + c0))[vc];
+  for (; c < nc; c += 16) {
+    const bf16x8 wv = ((const bf16x8 *)wr)[c];
+    const bf16x8 xv = ((const bf16x8 *)x)[c];
 #pragma unroll
-    for (int j = 0; j < 8; ++j) acc[j] += xv * bf16_to_f32(wv.u[j]);
+    for (int j = 0; j < 8; ++j)
+      acc += bf16_to_f32(xv.u[j]) * bf16_to_f32(wv.u[j]);
   }
 
-  // reduce the 32 k-lanes per vec-column via LDS, then write bf16 directly.
-  __shared__ float red[256 * 8];
-#pragma unroll
-  for (int j = 0; j < 8; ++j) red[t * 8 + j] = acc[j];
-  __syncthreads();
-  if (t < 64) {
-    const int vcc = t >> 3, j = t & 7;
-    float sum = 0.f;
-#pragma unroll 8
-    for (int klane = 0; klane < 32; ++klane)
-      sum += red[(vcc + 8 * klane) * 8 + j];
-    y[c0 + vcc * 8 + j] = f32_to_bf16(sum);
-  }
+  const float sum = group16_sum(acc);
+  if (sl == 0) y[n] = f32_to_bf16(sum);
 }
 
 extern "C" void launch_gemv(const ushort_t *x, const ushort_t *w, ushort_t *y,
                             int K, int N, hipStream_t stream) {
-  gemv_kernel<<<dim3(N / 64), 256, 0, stream>>>(x, w, y, K, N);
+  gemv_kernel<<<dim3((N + 15) / 16), 256, 0, stream>>>(x, w, y, K, N);
 }

@@ -51,43 +51,45 @@ def shard_config(config: LlamaConfig, tp: int) -> LlamaConfig:
 
 
 def shard_qkv(wqkv: torch.Tensor, config: LlamaConfig, tp: int, rank: int) -> torch.Tensor:
-    """Column-shard the fused [d, (h+2kh)*hd] QKV weight by heads."""
+    """Output-shard the fused [(h+2kh)*hd, d] QKV weight by heads
+    (weights are row-major [out, in])."""
     h, kh, hd = config.n_heads, config.n_kv_heads, config.head_dim
     hl, khl = h // tp, kh // tp
-    q = wqkv[:, : h * hd]
-    k = wqkv[:, h * hd : (h + kh) * hd]
-    v = wqkv[:, (h + kh) * hd :]
+    q = wqkv[: h * hd]
+    k = wqkv[h * hd : (h + kh) * hd]
+    v = wqkv[(h + kh) * hd :]
     return torch.cat(
         [
-            q[:, rank * hl * hd : (rank + 1) * hl * hd],
-            k[:, rank * khl * hd : (rank + 1) * khl * hd],
-            v[:, rank * khl * hd : (rank + 1) * khl * hd],
+            q[rank * hl * hd : (rank + 1) * hl * hd],
+            k[rank * khl * hd : (rank + 1) * khl * hd],
+            v[rank * khl * hd : (rank + 1) * khl * hd],
         ],
-        dim=1,
+        dim=0,
     ).contiguous()
 
 
 def shard_o(wo: torch.Tensor, config: LlamaConfig, tp: int, rank: int) -> torch.Tensor:
-    """Row-shard [h*hd, d] by this rank's head block."""
+    """Input-shard [d, h*hd] by this rank's head block (row-parallel: the
+    partial outputs are summed by the all-reduce)."""
     h, hd = config.n_heads, config.head_dim
     hl = h // tp
-    return wo[rank * hl * hd : (rank + 1) * hl * hd].contiguous()
+    return wo[:, rank * hl * hd : (rank + 1) * hl * hd].contiguous()
 
 
 def shard_gate_up(w: torch.Tensor, config: LlamaConfig, tp: int, rank: int) -> torch.Tensor:
-    """Column-shard the fused [d, 2*ffn] gate/up weight per half."""
+    """Output-shard the fused [2*ffn, d] gate/up weight per half."""
     f = config.ffn_dim
     fl = f // tp
-    gate = w[:, :f][:, rank * fl : (rank + 1) * fl]
-    up = w[:, f:][:, rank * fl : (rank + 1) * fl]
-    return torch.cat([gate, up], dim=1).contiguous()
+    gate = w[:f][rank * fl : (rank + 1) * fl]
+    up = w[f:][rank * fl : (rank + 1) * fl]
+    return torch.cat([gate, up], dim=0).contiguous()
 
 
 def shard_down(w: torch.Tensor, config: LlamaConfig, tp: int, rank: int) -> torch.Tensor:
-    """Row-shard [ffn, d]."""
+    """Input-shard [d, ffn] (row-parallel)."""
     f = config.ffn_dim
     fl = f // tp
-    return w[rank * fl : (rank + 1) * fl].contiguous()
+    return w[:, rank * fl : (rank + 1) * fl].contiguous()
 
 
 class TPContext:

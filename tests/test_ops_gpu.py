@@ -286,26 +286,44 @@ class TestRopeKV:
 
 
 class TestGemm:
+    # weights row-major [out, in]: C = a @ b^T (NT)
     @pytest.mark.parametrize("m", [3, 36, 56, 128, 300])
     def test_vs_torch_fp32(self, m):
         """In-tree MFMA GEMM vs fp32 torch on model-shaped projections."""
         k, n = 512, 1024
         a = _bf(torch.randn(m, k)).to(DEV)
-        b = _bf(torch.randn(k, n) * 0.05).to(DEV)
+        b = _bf(torch.randn(n, k) * 0.05).to(DEV)
         got = ops.gemm(a, b)
-        want = a.float().cpu() @ b.float().cpu()
+        want = a.float().cpu() @ b.float().cpu().t()
         _assert_close(got, want, atol=3e-2, name=f"gemm m={m}")
 
     def test_k_edge_and_bigger(self):
-        a = _bf(torch.randn(40, 72)).to(DEV)   # K not multiple of 32
-        b = _bf(torch.randn(72, 256) * 0.05).to(DEV)
+        a = _bf(torch.randn(40, 72)).to(DEV)   # K not multiple of 64
+        b = _bf(torch.randn(250, 72) * 0.05).to(DEV)  # N not multiple of 128
         got = ops.gemm(a, b)
-        want = a.float().cpu() @ b.float().cpu()
+        want = a.float().cpu() @ b.float().cpu().t()
         _assert_close(got, want, atol=3e-2, name="gemm k-edge")
+
+    def test_full_tile_shape(self):
+        a = _bf(torch.randn(256, 2048)).to(DEV)
+        b = _bf(torch.randn(1024, 2048) * 0.02).to(DEV)
+        got = ops.gemm(a, b)
+        want = a.float().cpu() @ b.float().cpu().t()
+        _assert_close(got, want, atol=5e-2, name="gemm full-tile")
 
     def test_deterministic(self):
         a = _bf(torch.randn(77, 2048)).to(DEV)
-        b = _bf(torch.randn(2048, 4096) * 0.02).to(DEV)
+        b = _bf(torch.randn(4096, 2048) * 0.02).to(DEV)
         x = ops.gemm(a, b)
         for _ in range(3):
             assert torch.equal(ops.gemm(a, b), x)
+
+
+class TestGemv:
+    @pytest.mark.parametrize("n,k", [(1024, 512), (6144, 4096), (1000, 264)])
+    def test_vs_torch_fp32(self, n, k):
+        x = _bf(torch.randn(1, k)).to(DEV)
+        w = _bf(torch.randn(n, k) * 0.05).to(DEV)
+        got = ops.gemv(x, w)
+        want = x.float().cpu() @ w.float().cpu().t()
+        _assert_close(got, want, atol=3e-2, name=f"gemv {n}x{k}")

@@ -33,32 +33,33 @@ class TestShardHelpers:
             shard_config(CFG, 3)
 
     def test_qkv_shards_concatenate(self):
+        # weights row-major [out, in]: shards partition OUTPUT rows by head
         h, kh, hd = CFG.n_heads, CFG.n_kv_heads, CFG.head_dim
-        w = torch.randn(CFG.dim, (h + 2 * kh) * hd)
+        w = torch.randn((h + 2 * kh) * hd, CFG.dim)
         s0 = shard_qkv(w, CFG, 2, 0)
         s1 = shard_qkv(w, CFG, 2, 1)
         # reassemble: q halves then k halves then v halves
-        q = torch.cat([s0[:, : 4 * hd], s1[:, : 4 * hd]], dim=1)
-        k = torch.cat([s0[:, 4 * hd : 5 * hd], s1[:, 4 * hd : 5 * hd]], dim=1)
-        v = torch.cat([s0[:, 5 * hd :], s1[:, 5 * hd :]], dim=1)
-        assert torch.equal(torch.cat([q, k, v], dim=1), w)
+        q = torch.cat([s0[: 4 * hd], s1[: 4 * hd]], dim=0)
+        k = torch.cat([s0[4 * hd : 5 * hd], s1[4 * hd : 5 * hd]], dim=0)
+        v = torch.cat([s0[5 * hd :], s1[5 * hd :]], dim=0)
+        assert torch.equal(torch.cat([q, k, v], dim=0), w)
 
-    def test_o_rows_partition(self):
-        w = torch.randn(CFG.n_heads * CFG.head_dim, CFG.dim)
+    def test_o_input_partition(self):
+        w = torch.randn(CFG.dim, CFG.n_heads * CFG.head_dim)
         parts = [shard_o(w, CFG, 2, r) for r in range(2)]
-        assert torch.equal(torch.cat(parts, dim=0), w)
+        assert torch.equal(torch.cat(parts, dim=1), w)
 
     def test_gate_up_down(self):
-        w = torch.randn(CFG.dim, 2 * CFG.ffn_dim)
+        w = torch.randn(2 * CFG.ffn_dim, CFG.dim)
         s0 = shard_gate_up(w, CFG, 2, 0)
         s1 = shard_gate_up(w, CFG, 2, 1)
         f = CFG.ffn_dim
-        gate = torch.cat([s0[:, : f // 2], s1[:, : f // 2]], dim=1)
-        up = torch.cat([s0[:, f // 2 :], s1[:, f // 2 :]], dim=1)
-        assert torch.equal(torch.cat([gate, up], dim=1), w)
-        wd = torch.randn(CFG.ffn_dim, CFG.dim)
+        gate = torch.cat([s0[: f // 2], s1[: f // 2]], dim=0)
+        up = torch.cat([s0[f // 2 :], s1[f // 2 :]], dim=0)
+        assert torch.equal(torch.cat([gate, up], dim=0), w)
+        wd = torch.randn(CFG.dim, CFG.ffn_dim)
         assert torch.equal(
-            torch.cat([shard_down(wd, CFG, 2, r) for r in range(2)], dim=0), wd
+            torch.cat([shard_down(wd, CFG, 2, r) for r in range(2)], dim=1), wd
         )
 
 

@@ -45,10 +45,10 @@ def probe_gemv():
                       (4096, 28672, "gate_up"), (14336, 4096, "down"),
                       (4096, 128256, "lm_head")]:
         x = torch.randn(1, K, dtype=torch.bfloat16, device="cuda")
-        w = torch.randn(K, N, dtype=torch.bfloat16, device="cuda")
-        t = bench_gpu(lambda: hip.gemv(x, w), iters=50)
+        w = torch.randn(N, K, dtype=torch.bfloat16, device="cuda")
+        t = bench_gpu(lambda: hip.gemv(x, w, None), iters=50)
         tb = K * N * 2 / t / 1e12
-        tt = bench_gpu(lambda: x @ w, iters=50)
+        tt = bench_gpu(lambda: x @ w.t(), iters=50)
         print(f"gemv {tag:8s} [{K}x{N}]: {t*1e6:7.1f} us {tb:5.2f} TB/s "
               f"(hipBLASLt {tt*1e6:7.1f} us {K*N*2/tt/1e12:5.2f} TB/s)")
 
