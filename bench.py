@@ -204,8 +204,9 @@ def main() -> int:
 
     if rank == 0:
         print(json.dumps({
-            "metric": "critiques/sec (debate round: 8k-token spec prefill + "
-                      f"{args.decode_tokens}-token critique decode + RCCL consensus)",
+            "metric": f"critiques/sec (debate round: {args.spec_tokens}-token "
+                      f"spec prefill + {args.decode_tokens}-token critique "
+                      "decode + RCCL consensus)",
             "value": value,
             "unit": "critiques/s",
             "n_gpus": world,
