@@ -72,6 +72,9 @@ class LocalEngine:
         else:
             self.model.init_random()
             self.tokenizer = build_tokenizer(self.config.vocab_size)
+        if spec.get("dtype") == "fp8":
+            # BASELINE config 5: e4m3 MFMA prefill + fp8 weight streaming
+            self.model.quantize_fp8()
         # ORDER the engine stream after weight init: weights/tables are
         # initialized on the CREATION stream (usually the default stream),
         # but all generate() work runs on self.stream. Without this event

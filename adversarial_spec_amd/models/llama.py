@@ -42,6 +42,13 @@ from .config import LlamaConfig
 
 
 @dataclass
+class QuantW:
+    """Rowwise e4m3-quantized projection: q uint8 [out, in], scale f32 [out]."""
+    q: torch.Tensor
+    s: torch.Tensor
+
+
+@dataclass
 class LayerWeights:
     # all projection weights ROW-MAJOR [out, in] (HF layout; x @ W^T)
     attn_norm: torch.Tensor  # [d]
@@ -108,6 +115,10 @@ class DecodeWorkspace:
         self.mlp_out = mk(1, c.dim)
         self.logits = mk(1, c.vocab_size)
         self.tok_long = torch.empty(1, dtype=torch.long, device=dev)
+        # fp8 decode scratch: rowwise-quantized activation + scale
+        kmax = max(c.dim, h * hd, c.ffn_dim)
+        self.x8 = torch.empty(1, kmax, dtype=torch.uint8, device=dev)
+        self.xs = torch.empty(1, dtype=torch.float32, device=dev)
 
 
 class LlamaModel:
@@ -138,8 +149,11 @@ class LlamaModel:
         )
         self.embed: Optional[torch.Tensor] = None  # [vocab, d]
         self.final_norm: Optional[torch.Tensor] = None  # [d]
-        self.lm_head: Optional[torch.Tensor] = None  # [d, vocab]
+        self.lm_head: Optional[torch.Tensor] = None  # [vocab, d]
         self.layers: list[LayerWeights] = []
+        self.fp8 = False  # set by quantize_fp8()
+        self.layers_q: list = []  # per-layer QuantW mirrors when fp8
+        self.lm_head_q = None
 
     # -- weight initialisation ---------------------------------------------
 
@@ -273,6 +287,28 @@ class LlamaModel:
             )
         return self
 
+    def quantize_fp8(self) -> "LlamaModel":
+        """Switch the projection weights to rowwise OCP e4m3 (BASELINE
+        config 5: fp8 MFMA prefill; fp8 also halves decode weight
+        streaming). Norms, embeddings and the attention path stay bf16;
+        scales factor out of every dot product, so dequantization is exact
+        in the kernel epilogue."""
+        from dataclasses import dataclass as _dc  # noqa: F401
+
+        self.layers_q = []
+        for L in self.layers:
+            q = {}
+            for f in ("wqkv", "wo", "w_gate_up", "w_down"):
+                qt, sc = ops.quantize_fp8_rowwise(getattr(L, f))
+                q[f] = QuantW(qt.to(self.device), sc.to(self.device))
+                setattr(L, f, None)  # free the bf16 copy
+            self.layers_q.append(q)
+        qt, sc = ops.quantize_fp8_rowwise(self.lm_head)
+        self.lm_head_q = QuantW(qt.to(self.device), sc.to(self.device))
+        self.lm_head = None
+        self.fp8 = True
+        return self
+
     # -- inference ----------------------------------------------------------
 
     def new_cache(self, max_seq: Optional[int] = None) -> PagedKVCache:
@@ -296,12 +332,21 @@ class LlamaModel:
         resid = self.embed[tokens]  # [t, d]
 
         # batch-1 decode projections stream weights through the hand-written
-        # GEMV kernel; prefill (t>1) uses the in-tree tiled MFMA GEMM.
+        # GEMV kernel; prefill (t>1) uses the in-tree tiled MFMA GEMM; fp8
+        # mode routes every projection through the e4m3 MFMA kernel.
         mm = ops.gemv if t == 1 else ops.gemm
+        fp8 = self.fp8
+
+        def proj(x, L_, Qd, name):
+            if fp8:
+                qw = Qd[name]
+                return ops.gemm_fp8(x, qw.q, qw.s)
+            return mm(x, getattr(L_, name))
 
         normed = ops.rmsnorm(resid, self.layers[0].attn_norm, c.norm_eps)
         for i, L in enumerate(self.layers):
-            qkv = mm(normed, L.wqkv)  # [t, (h+2kh)*hd]
+            Qd = self.layers_q[i] if fp8 else None
+            qkv = proj(normed, L, Qd, "wqkv")  # [t, (h+2kh)*hd]
             q = qkv[:, : h * hd].view(t, h, hd)
             k = qkv[:, h * hd : (h + kh) * hd].view(t, kh, hd)
             v = qkv[:, (h + kh) * hd :].view(t, kh, hd)
@@ -318,20 +363,24 @@ class LlamaModel:
                     q[0], cache.k[i], cache.v[i], cache.page_table, seq,
                     self.scale, pos_state=pos_state,
                 ).unsqueeze(0)
-            attn_out = mm(attn.reshape(t, h * hd), L.wo)
+            attn_out = proj(attn.reshape(t, h * hd), L, Qd, "wo")
             if self.tp is not None and self.tp.size > 1:
                 self.tp.all_reduce_(attn_out)  # row-parallel wo partial sums
             resid, normed = ops.add_rmsnorm(resid, attn_out, L.mlp_norm, c.norm_eps)
-            gu = mm(normed, L.w_gate_up)
+            gu = proj(normed, L, Qd, "w_gate_up")
             act = ops.swiglu(gu[:, : c.ffn_dim], gu[:, c.ffn_dim :])
-            mlp_out = mm(act, L.w_down)
+            mlp_out = proj(act, L, Qd, "w_down")
             if self.tp is not None and self.tp.size > 1:
                 self.tp.all_reduce_(mlp_out)  # row-parallel down partial sums
             next_norm = (
                 self.layers[i + 1].attn_norm if i + 1 < c.n_layers else self.final_norm
             )
             resid, normed = ops.add_rmsnorm(resid, mlp_out, next_norm, c.norm_eps)
-        logits = ops.gemv(normed[-1:].contiguous(), self.lm_head)  # [1, vocab]
+        if fp8:
+            logits = ops.gemm_fp8(normed[-1:].contiguous(), self.lm_head_q.q,
+                                  self.lm_head_q.s)
+        else:
+            logits = ops.gemv(normed[-1:].contiguous(), self.lm_head)  # [1, vocab]
         return logits[0]
 
     def prefill(self, tokens: torch.Tensor, cache: PagedKVCache) -> torch.Tensor:
@@ -366,11 +415,21 @@ class LlamaModel:
         dynamic state in device words, zero allocator traffic)."""
         c = self.config
         h, kh, hd = c.n_heads, c.n_kv_heads, c.head_dim
+        fp8 = self.fp8
+
+        def proj(x, L_, Qd, name, out):
+            if fp8:
+                qw = Qd[name]
+                ops.gemv_fp8(x, qw.q, qw.s, W.x8, W.xs, out.view(1, -1))
+                return out
+            return ops.gemv(x, getattr(L_, name), out=out)
+
         torch.index_select(self.embed, 0, W.tok_long, out=W.resid)
         ops.rmsnorm(W.resid, self.layers[0].attn_norm, c.norm_eps,
                     out=W.normed)
         for i, L in enumerate(self.layers):
-            ops.gemv(W.normed, L.wqkv, out=W.qkv)
+            Qd = self.layers_q[i] if fp8 else None
+            proj(W.normed, L, Qd, "wqkv", W.qkv)
             q = W.qkv[:, : h * hd].view(1, h, hd)
             k = W.qkv[:, h * hd : (h + kh) * hd].view(1, kh, hd)
             v = W.qkv[:, (h + kh) * hd :].view(1, kh, hd)
@@ -380,21 +439,25 @@ class LlamaModel:
                 q[0], cache.k[i], cache.v[i], cache.page_table,
                 max_seq_bound, self.scale, pos_state=pos_state, out=W.attn,
             )
-            ops.gemv(W.attn.view(1, h * hd), L.wo, out=W.attn_out)
+            proj(W.attn.view(1, h * hd), L, Qd, "wo", W.attn_out)
             if self.tp is not None and self.tp.size > 1:
                 self.tp.all_reduce_(W.attn_out)
             ops.add_rmsnorm(W.resid, W.attn_out, L.mlp_norm, c.norm_eps,
                             out_resid=W.resid2, out_y=W.normed)
-            ops.gemv(W.normed, L.w_gate_up, out=W.gu)
+            proj(W.normed, L, Qd, "w_gate_up", W.gu)
             ops.swiglu(W.gu[:, : c.ffn_dim], W.gu[:, c.ffn_dim :], out=W.act)
-            ops.gemv(W.act, L.w_down, out=W.mlp_out)
+            proj(W.act, L, Qd, "w_down", W.mlp_out)
             if self.tp is not None and self.tp.size > 1:
                 self.tp.all_reduce_(W.mlp_out)
             nxt = (self.layers[i + 1].attn_norm if i + 1 < c.n_layers
                    else self.final_norm)
             ops.add_rmsnorm(W.resid2, W.mlp_out, nxt, c.norm_eps,
                             out_resid=W.resid, out_y=W.normed)
-        ops.gemv(W.normed, self.lm_head, out=W.logits)
+        if fp8:
+            ops.gemv_fp8(W.normed, self.lm_head_q.q, self.lm_head_q.s,
+                         W.x8, W.xs, W.logits)
+        else:
+            ops.gemv(W.normed, self.lm_head, out=W.logits)
         return W.logits
 
     def decode_one_graph(self, tok_slot: torch.Tensor, cache: PagedKVCache,

@@ -200,6 +200,40 @@ def gemm(x: torch.Tensor, w: torch.Tensor) -> torch.Tensor:
     return x @ w.t()
 
 
+def quantize_fp8_rowwise(w: torch.Tensor) -> Tuple[torch.Tensor, torch.Tensor]:
+    """Rowwise OCP e4m3 quantization: returns (q uint8 [N,K], scale f32 [N]).
+    Pure torch (runs once at engine init, CPU or GPU)."""
+    wf = w.float()
+    scale = wf.abs().amax(dim=1).clamp_min(1e-12) / 448.0
+    q = (wf / scale[:, None]).to(torch.float8_e4m3fn).view(torch.uint8)
+    return q.contiguous(), scale.contiguous()
+
+
+def dequantize_fp8(q: torch.Tensor, scale: torch.Tensor) -> torch.Tensor:
+    return q.view(torch.float8_e4m3fn).float() * scale[:, None]
+
+
+def gemm_fp8(x: torch.Tensor, w_q: torch.Tensor, w_scale: torch.Tensor) -> torch.Tensor:
+    """C = x @ W^T with W rowwise-e4m3-quantized; x is rowwise-quantized
+    on the fly. GPU: mfma_f32_16x16x32_fp8_fp8 kernel. CPU: dequantized
+    reference (tests)."""
+    if _on_gpu(x):
+        return _require_hip().gemm_fp8(x, w_q, w_scale)
+    xf = x.float()
+    xs = xf.abs().amax(dim=1).clamp_min(1e-12) / 448.0
+    x8 = (xf / xs[:, None]).to(torch.float8_e4m3fn).float() * xs[:, None]
+    return (x8 @ dequantize_fp8(w_q, w_scale).t()).to(x.dtype)
+
+
+def gemv_fp8(x: torch.Tensor, w_q: torch.Tensor, w_scale: torch.Tensor,
+             x8_buf: torch.Tensor, xs_buf: torch.Tensor,
+             out: torch.Tensor) -> torch.Tensor:
+    """Allocation-free fp8 decode GEMV (graph-capture safe): quantizes x
+    into the provided scratch, writes bf16 into out."""
+    _require_hip().gemv_fp8(x, w_q, w_scale, x8_buf, xs_buf, out)
+    return out
+
+
 def sample(
     logits: torch.Tensor,
     temperature: float = 0.7,

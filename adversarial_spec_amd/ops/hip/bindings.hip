@@ -46,6 +46,11 @@ void launch_gemv(const ushort_t*, const ushort_t*, ushort_t*, int, int,
                  hipStream_t);
 void launch_gemm(const ushort_t*, const ushort_t*, ushort_t*, int, int, int,
                  hipStream_t);
+void launch_gemm_fp8(const uint8_t*, const float*, const uint8_t*, const float*,
+                     ushort_t*, int, int, int, hipStream_t);
+void launch_gemv_fp8(const uint8_t*, const float*, const uint8_t*, const float*,
+                     ushort_t*, int, int, hipStream_t);
+void launch_quant_fp8(const ushort_t*, uint8_t*, float*, int, int, hipStream_t);
 }
 
 static hipStream_t cur_stream() {
@@ -389,6 +394,52 @@ torch::Tensor gemm(torch::Tensor a, torch::Tensor b) {
   return c;
 }
 
+// Row-quantize bf16 [M,K] to OCP e4m3 + per-row scale (scale = rowmax/448).
+void quant_fp8(torch::Tensor x, torch::Tensor q, torch::Tensor scale) {
+  CHECK_BF16_CUDA(x);
+  TORCH_CHECK(q.scalar_type() == at::kByte && scale.scalar_type() == at::kFloat);
+  auto xc = x.contiguous();
+  const int K = xc.size(-1), M = xc.numel() / K;
+  TORCH_CHECK(K % 8 == 0, "quant_fp8: K % 8 == 0");
+  launch_quant_fp8(uptr(xc), q.data_ptr<uint8_t>(), scale.data_ptr<float>(),
+                   M, K, cur_stream());
+}
+
+// fp8 GEMM: C = (x quantized rowwise) @ (w8 * wsc)^T; w8 row-major [N,K]
+// e4m3 + per-row scale. x is quantized in here (eager prefill path).
+torch::Tensor gemm_fp8(torch::Tensor x, torch::Tensor w8, torch::Tensor wsc) {
+  CHECK_BF16_CUDA(x);
+  TORCH_CHECK(w8.scalar_type() == at::kByte && w8.is_contiguous());
+  auto xc = x.contiguous();
+  const int M = xc.size(0), K = xc.size(1), N = w8.size(0);
+  TORCH_CHECK(w8.size(1) == K, "gemm_fp8: K mismatch");
+  auto opts8 = torch::TensorOptions().dtype(at::kByte).device(x.device());
+  auto optsf = torch::TensorOptions().dtype(at::kFloat).device(x.device());
+  auto x8 = torch::empty({M, (long)K}, opts8);
+  auto xs = torch::empty({M}, optsf);
+  launch_quant_fp8(uptr(xc), x8.data_ptr<uint8_t>(), xs.data_ptr<float>(),
+                   M, K, cur_stream());
+  auto c = torch::empty({M, N}, xc.options());
+  launch_gemm_fp8(x8.data_ptr<uint8_t>(), xs.data_ptr<float>(),
+                  w8.data_ptr<uint8_t>(), wsc.data_ptr<float>(), uptr_mut(c),
+                  M, N, K, cur_stream());
+  return c;
+}
+
+// fp8 decode GEMV (allocation-free: caller provides the x8/xs scratch and
+// the output — all preallocated in the DecodeWorkspace for graph capture).
+void gemv_fp8(torch::Tensor x, torch::Tensor w8, torch::Tensor wsc,
+              torch::Tensor x8, torch::Tensor xs, torch::Tensor out) {
+  CHECK_BF16_CUDA(x);
+  auto xc = x.contiguous();
+  const int K = xc.numel(), N = w8.size(0);
+  launch_quant_fp8(uptr(xc), x8.data_ptr<uint8_t>(), xs.data_ptr<float>(),
+                   1, K, cur_stream());
+  launch_gemv_fp8(x8.data_ptr<uint8_t>(), xs.data_ptr<float>(),
+                  w8.data_ptr<uint8_t>(), wsc.data_ptr<float>(),
+                  uptr_mut(out), K, N, cur_stream());
+}
+
 int64_t sample(torch::Tensor logits, double temp, double top_p, int64_t seed) {
   CHECK_BF16_CUDA(logits);
   TORCH_CHECK(top_p >= 1.0, "kernel sample handles top_p == 1 (nucleus is a cold path)");
@@ -474,6 +525,9 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("gemv", &gemv, "batch-1 decode GEMV (weight streaming)",
         py::arg("x"), py::arg("w"), py::arg("out") = py::none());
   m.def("gemm", &gemm, "tiled MFMA GEMM (bf16, fp32 accum)");
+  m.def("gemm_fp8", &gemm_fp8, "fp8 e4m3 MFMA GEMM (rowwise scales)");
+  m.def("gemv_fp8", &gemv_fp8, "fp8 decode GEMV (rowwise scales)");
+  m.def("quant_fp8", &quant_fp8, "rowwise bf16 -> e4m3 quantizer");
   m.def("rope_inplace_ds", &rope_inplace_ds, "graph-mode RoPE (device pos)");
   m.def("rope_kv", &rope_kv, "fused RoPE + paged KV scatter",
         py::arg("q"), py::arg("k"), py::arg("v"), py::arg("cost"),

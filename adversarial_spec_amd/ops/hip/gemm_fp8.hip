@@ -1,0 +1,314 @@
+// fp8 (OCP e4m3) MFMA GEMM + GEMV for CDNA4 (gfx950).
+//
+//   C[M,N] = (A8[M,K] * a_scale[M]) @ (B8[N,K] * b_scale[N])^T
+//
+// BASELINE config 5: the 32k-token long-context prefill runs the
+// projection GEMMs on v_mfma_f32_16x16x32_fp8_fp8 with row-wise scales
+// (scales factor out of the dot product, so dequantization is exact in
+// the epilogue). fp8 also HALVES the weight bytes streamed by the
+// bandwidth-bound decode GEMV — the decode path uses the same quantized
+// weights.
+//
+// Same step-3 structure as gemm.hip (128x128 tile, BK=64 elements,
+// double-buffered global_load_lds staging, source-XOR swizzle) with
+// half-width rows: an image row is 64 B (4 x 16 B chunks), fragments are
+// 8 fp8 = 8 B per lane read as the low/high half of a swizzled 16 B chunk.
+
+#include "common.h"
+
+typedef __attribute__((__vector_size__(4 * sizeof(float)))) float f32x4v;
+
+#define F8_M 128
+#define F8_N 128
+#define F8_K 64   // elements per K-step; one image row = 64 B
+
+DEVINL void glds16_f8(const uint8_t *g, uint8_t *l) {
+  __builtin_amdgcn_global_load_lds(
+      (const __attribute__((address_space(1))) void *)g,
+      (__attribute__((address_space(3))) void *)l, 16, 0, 0);
+}
+
+// Stage a [128][64] fp8 tile (8 KiB) via glds: 8 wave-instructions of
+// 1 KiB (16 rows x 64 B), 2 per wave. Global chunk XOR (row & 3) realizes
+// the swizzled image with lane-linear LDS placement.
+DEVINL void stage_glds_f8(const uint8_t *__restrict__ src, long row_stride,
+                          uint8_t *__restrict__ img, int tid) {
+  const int lane = tid & (WAVE - 1);
+  const int wid = tid / WAVE;
+#pragma unroll
+  for (int i = 0; i < 2; ++i) {
+    const int r0 = (wid * 2 + i) * 16;
+    const int row = r0 + (lane >> 2);
+    const int c = lane & 3;
+    const uint8_t *g = src + (size_t)row * row_stride + (size_t)(c ^ (row & 3)) * 16;
+    glds16_f8(g, img + (size_t)r0 * F8_K);
+  }
+}
+
+DEVINL void stage_edge_f8(const uint8_t *__restrict__ src, long row_stride,
+                          int rows_left, int k_left,
+                          uint8_t *__restrict__ img, int tid) {
+#pragma unroll
+  for (int p = 0; p < 2; ++p) {
+    const int piece = tid + 256 * p;   // 512 chunk-pieces of 16 B
+    const int row = piece >> 2;        // 0..127
+    const int c = piece & 3;           // 16B chunk
+    uint8_t v[16];
+    const int gk = c * 16;
+    if (row < rows_left && gk < k_left) {
+      const uint8_t *g = src + (size_t)row * row_stride + gk;
+#pragma unroll
+      for (int j = 0; j < 16; ++j) v[j] = (gk + j < k_left) ? g[j] : 0;
+    } else {
+#pragma unroll
+      for (int j = 0; j < 16; ++j) v[j] = 0;
+    }
+    uint8_t *dst = img + (size_t)row * F8_K + (size_t)(c ^ (row & 3)) * 16;
+#pragma unroll
+    for (int j = 0; j < 16; ++j) dst[j] = v[j];
+  }
+}
+
+__global__ void __launch_bounds__(256, 2)
+gemm_fp8_kernel(const uint8_t *__restrict__ a, const float *__restrict__ asc,
+                const uint8_t *__restrict__ b, const float *__restrict__ bsc,
+                ushort_t *__restrict__ c_out, int M, int N, int K) {
+  const int tid = threadIdx.x;
+  const int lane = tid & (WAVE - 1);
+  const int wid = tid / WAVE;
+  const int lrow = lane & 15;
+  const int lhi = lane >> 4;
+
+  const int n0 = blockIdx.x * F8_N;
+  const int m0 = blockIdx.y * F8_M;
+  const int wm = (wid >> 1) * 64;
+  const int wn = (wid & 1) * 64;
+
+  __shared__ __attribute__((aligned(16))) uint8_t imgA[2][F8_M * F8_K];
+  __shared__ __attribute__((aligned(16))) uint8_t imgB[2][F8_N * F8_K];
+
+  f32x4v acc[4][4];
+#pragma unroll
+  for (int i = 0; i < 4; ++i)
+#pragma unroll
+    for (int j = 0; j < 4; ++j) acc[i][j] = (f32x4v){0.f, 0.f, 0.f, 0.f};
+
+  const int ntiles = (K + F8_K - 1) / F8_K;
+  const bool a_edge = (M - m0) < F8_M;
+  const bool b_edge = (N - n0) < F8_N;
+
+  auto stage = [&](int kt, int buf) {
+    const uint8_t *asrc = a + (size_t)m0 * K + (size_t)kt * F8_K;
+    const uint8_t *bsrc = b + (size_t)n0 * K + (size_t)kt * F8_K;
+    const bool k_edge = (kt + 1) * F8_K > K;
+    if (a_edge || k_edge) {
+      stage_edge_f8(asrc, K, M - m0, K - kt * F8_K, imgA[buf], tid);
+    } else {
+      stage_glds_f8(asrc, K, imgA[buf], tid);
+    }
+    if (b_edge || k_edge) {
+      stage_edge_f8(bsrc, K, N - n0, K - kt * F8_K, imgB[buf], tid);
+    } else {
+      stage_glds_f8(bsrc, K, imgB[buf], tid);
+    }
+  };
+
+  stage(0, 0);
+  __syncthreads();
+
+  for (int kt = 0; kt < ntiles; ++kt) {
+    const int buf = kt & 1;
+    if (kt + 1 < ntiles) stage(kt + 1, buf ^ 1);
+
+    // 2 k-subtiles of 32 elements: frag = 8 B per lane (low/high half of
+    // a swizzled 16 B chunk)
+#pragma unroll
+    for (int s = 0; s < 2; ++s) {
+      long afr[4], bfr[4];
+#pragma unroll
+      for (int i = 0; i < 4; ++i) {
+        const int row = wm + i * 16 + lrow;
+        const int byte_off = s * 32 + lhi * 8;           // within the row
+        const int ch = (byte_off >> 4) ^ (row & 3);      // 16B chunk
+        const int half = (byte_off >> 3) & 1;
+        afr[i] = ((const long *)(imgA[buf] + (size_t)row * F8_K + ch * 16))[half];
+      }
+#pragma unroll
+      for (int j2 = 0; j2 < 4; ++j2) {
+        const int row = wn + j2 * 16 + lrow;
+        const int byte_off = s * 32 + lhi * 8;
+        const int ch = (byte_off >> 4) ^ (row & 3);
+        const int half = (byte_off >> 3) & 1;
+        bfr[j2] = ((const long *)(imgB[buf] + (size_t)row * F8_K + ch * 16))[half];
+      }
+#pragma unroll
+      for (int i = 0; i < 4; ++i)
+#pragma unroll
+        for (int j2 = 0; j2 < 4; ++j2)
+          acc[i][j2] = __builtin_amdgcn_mfma_f32_16x16x32_fp8_fp8(
+              afr[i], bfr[j2], acc[i][j2], 0, 0, 0);
+    }
+    __syncthreads();
+  }
+
+  // epilogue: dequantize with the row scales (exact: scales factor out)
+#pragma unroll
+  for (int i = 0; i < 4; ++i) {
+#pragma unroll
+    for (int j2 = 0; j2 < 4; ++j2) {
+      const int col = n0 + wn + j2 * 16 + lrow;
+      if (col >= N) continue;
+      const float bs = bsc[col];
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        const int row = m0 + wm + i * 16 + lhi * 4 + r;
+        if (row >= M) continue;
+        c_out[(size_t)row * N + col] =
+            f32_to_bf16(acc[i][j2][r] * asc[row] * bs);
+      }
+    }
+  }
+}
+
+// ---------------------------------------------------------------------------
+// fp8 decode GEMV: y[1,N] = (x8 * xs) @ (W8[N,K] * ws[N])^T — same tiling
+// as gemv.hip but HALF the streamed bytes (the decode roofline).
+// ---------------------------------------------------------------------------
+
+DEVINL float fp8_to_f32(uint8_t v) {
+  // OCP e4m3 (1s 4e 3m, bias 7) -> f32 by bit manipulation: for normals
+  // the f32 exponent is e+(127-7) and the mantissa top 3 bits are m, so
+  // bits = sign | (em + (120<<3)) << 20. Subnormals: m * 2^-9.
+  const uint32_t s = (uint32_t)(v & 0x80u) << 24;
+  const uint32_t em = v & 0x7fu;
+  uint32_t bits;
+  if (em >= 8u) {
+    bits = s | ((em + 960u) << 20);
+  } else {
+    bits = s | __builtin_bit_cast(uint32_t, (float)em * 0.001953125f);
+  }
+  return __builtin_bit_cast(float, bits);
+}
+
+extern "C" __global__ void __launch_bounds__(256)
+gemv_fp8_kernel(const uint8_t *__restrict__ x, const float *__restrict__ xs,
+                const uint8_t *__restrict__ w, const float *__restrict__ wsc,
+                ushort_t *__restrict__ y, int K, int N) {
+  const int lane = threadIdx.x & (WAVE - 1);
+  const int wid = threadIdx.x / WAVE;
+  const int rg = lane >> 4;
+  const int sl = lane & 15;
+  const int n = blockIdx.x * 16 + wid * 4 + rg;
+  if (n >= N) return;
+
+  const uint8_t *wr = w + (size_t)n * K;
+  const int nc = K / 16;  // 16 B chunks of 16 fp8 (K % 16 == 0)
+
+  float acc = 0.f;
+  int c = sl;
+  for (; c + 48 < nc; c += 64) {
+#pragma unroll
+    for (int u = 0; u < 4; ++u) {
+      const uint8_t *wp = wr + (size_t)(c + 16 * u) * 16;
+      const uint8_t *xp = x + (size_t)(c + 16 * u) * 16;
+      uint4 wv = *(const uint4 *)wp;
+      uint4 xv = *(const uint4 *)xp;
+      const uint8_t *w8 = (const uint8_t *)&wv;
+      const uint8_t *x8 = (const uint8_t *)&xv;
+#pragma unroll
+      for (int j = 0; j < 16; ++j)
+        acc += fp8_to_f32(x8[j]) * fp8_to_f32(w8[j]);
+    }
+  }
+  for (; c < nc; c += 16) {
+    uint4 wv = *(const uint4 *)(wr + (size_t)c * 16);
+    uint4 xv = *(const uint4 *)(x + (size_t)c * 16);
+    const uint8_t *w8 = (const uint8_t *)&wv;
+    const uint8_t *x8 = (const uint8_t *)&xv;
+#pragma unroll
+    for (int j = 0; j < 16; ++j)
+      acc += fp8_to_f32(x8[j]) * fp8_to_f32(w8[j]);
+  }
+
+#pragma unroll
+  for (int off = 8; off > 0; off >>= 1) acc += __shfl_xor(acc, off, WAVE);
+  if (sl == 0) y[n] = f32_to_bf16(acc * xs[0] * wsc[n]);
+}
+
+// ---------------------------------------------------------------------------
+// Row-wise fp8 quantizer: in[M,K] bf16 -> out[M,K] e4m3 + scale[M] f32
+// (scale = rowmax/448; allocation-free out-variant for the decode graph).
+// One block per row.
+// ---------------------------------------------------------------------------
+
+DEVINL uint8_t f32_to_fp8(float f) {
+  // round-to-nearest via float -> e4m3 with saturation to +-448
+  const uint32_t s = (__builtin_bit_cast(uint32_t, f) >> 31) & 1;
+  float a = fabsf(f);
+  if (a > 448.f) a = 448.f;
+  if (a < 0.0009765625f) {  // < 2^-10: rounds to 0 or smallest subnormal
+    const uint32_t m = (uint32_t)(a * 512.f + 0.5f);  // in units of 2^-9
+    return (uint8_t)((s << 7) | m);
+  }
+  int e;
+  float mant = frexpf(a, &e);        // a = mant * 2^e, mant in [0.5, 1)
+  // e4m3 value = (1.m3) * 2^(E-7): normalize to exponent e-1
+  int E = e - 1 + 7;
+  float mm = mant * 2.f;             // [1, 2)
+  int m3 = (int)((mm - 1.f) * 8.f + 0.5f);
+  if (m3 == 8) { m3 = 0; E += 1; }
+  if (E <= 0) {                      // subnormal
+    const uint32_t m = (uint32_t)(a * 512.f + 0.5f);
+    return (uint8_t)((s << 7) | (m > 7 ? 7 : m));
+  }
+  if (E > 15 || (E == 15 && m3 > 6)) { E = 15; m3 = 6; }  // clamp to 448
+  return (uint8_t)((s << 7) | (E << 3) | m3);
+}
+
+extern "C" __global__ void __launch_bounds__(256)
+quant_fp8_row_kernel(const ushort_t *__restrict__ x, uint8_t *__restrict__ q,
+                     float *__restrict__ scale, int K) {
+  __shared__ float scratch[16];
+  const int row = blockIdx.x;
+  const ushort_t *xr = x + (size_t)row * K;
+  uint8_t *qr = q + (size_t)row * K;
+  const int nv = K / 8;
+
+  float amax = 0.f;
+  for (int i = threadIdx.x; i < nv; i += blockDim.x) {
+    const f32x8 v = unpack8(((const bf16x8 *)xr)[i]);
+#pragma unroll
+    for (int j = 0; j < 8; ++j) amax = fmaxf(amax, fabsf(v.v[j]));
+  }
+  amax = block_reduce_max(amax, scratch);
+  const float s = (amax > 0.f) ? amax / 448.f : 1.f;
+  const float inv = 1.f / s;
+  if (threadIdx.x == 0) scale[row] = s;
+
+  for (int i = threadIdx.x; i < nv; i += blockDim.x) {
+    const f32x8 v = unpack8(((const bf16x8 *)xr)[i]);
+#pragma unroll
+    for (int j = 0; j < 8; ++j) qr[i * 8 + j] = f32_to_fp8(v.v[j] * inv);
+  }
+}
+
+extern "C" void launch_gemm_fp8(const uint8_t *a, const float *asc,
+                                const uint8_t *b, const float *bsc,
+                                ushort_t *c, int M, int N, int K,
+                                hipStream_t stream) {
+  dim3 grid((N + F8_N - 1) / F8_N, (M + F8_M - 1) / F8_M);
+  gemm_fp8_kernel<<<grid, 256, 0, stream>>>(a, asc, b, bsc, c, M, N, K);
+}
+
+extern "C" void launch_gemv_fp8(const uint8_t *x, const float *xs,
+                                const uint8_t *w, const float *wsc,
+                                ushort_t *y, int K, int N,
+                                hipStream_t stream) {
+  gemv_fp8_kernel<<<dim3((N + 15) / 16), 256, 0, stream>>>(x, xs, w, wsc, y,
+                                                           K, N);
+}
+
+extern "C" void launch_quant_fp8(const ushort_t *x, uint8_t *q, float *scale,
+                                 int M, int K, hipStream_t stream) {
+  quant_fp8_row_kernel<<<dim3(M), 256, 0, stream>>>(x, q, scale, K);
+}

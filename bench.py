@@ -60,6 +60,8 @@ def parse_args():
                    help="comma list of archs cycled over opponents for the "
                         "heterogeneous config (e.g. llama-3-8b,mistral-7b)")
     p.add_argument("--temperature", type=float, default=0.7)
+    p.add_argument("--dtype", default="bf16", choices=["bf16", "fp8"],
+                   help="projection compute dtype (fp8 = BASELINE config 5)")
     p.add_argument("--tp", type=int, default=1,
                    help="tensor-parallel degree: all ranks form ONE sharded "
                         "opponent (BASELINE config 5: llama-3-70b --tp 8)")
@@ -96,7 +98,8 @@ def main() -> int:
         n_opp = 1
         engines = [
             LocalEngine(
-                {"name": f"{args.model}-tp", "arch": args.model},
+                {"name": f"{args.model}-tp", "arch": args.model,
+                 "dtype": args.dtype},
                 device=device, tp=TPContext.from_default_group(),
             )
         ]
@@ -109,7 +112,8 @@ def main() -> int:
             arch = archs[(rank * n_opp + i) % len(archs)]
             engines.append(
                 LocalEngine(
-                    {"name": f"{arch}-r{rank}o{i}", "arch": arch},
+                    {"name": f"{arch}-r{rank}o{i}", "arch": arch,
+                     "dtype": args.dtype},
                     device=device,
                 )
             )
@@ -216,7 +220,7 @@ def main() -> int:
             "higher_is_better": True,
             "scaling": "weak",
             "vs_baseline": None,
-            "dtype": "bf16",
+            "dtype": args.dtype,
             "data": "synthetic",
             "config": {
                 "model": args.model,

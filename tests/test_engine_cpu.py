@@ -126,3 +126,34 @@ class TestSyntheticSpec:
     def test_looks_like_markdown(self):
         s = synthetic_spec(2000, seed=0)
         assert s.startswith("# ") and "## " in s
+
+
+class TestFp8CPU:
+    def test_quantize_roundtrip_close(self):
+        import torch
+
+        from adversarial_spec_amd import ops
+
+        w = torch.randn(64, 128) * 0.05
+        q, s = ops.quantize_fp8_rowwise(w)
+        back = ops.dequantize_fp8(q, s)
+        assert (back - w.float()).abs().max().item() < 0.05 * w.abs().max().item()
+
+    def test_fp8_model_forward_close_to_fp32(self):
+        import torch
+
+        from adversarial_spec_amd.models import LlamaModel
+        from adversarial_spec_amd.models.config import get_config
+
+        cfg = get_config("tiny")
+        m = LlamaModel(cfg, device="cpu", seed=3).init_random()
+        toks = torch.arange(2, 30)
+        c1 = m.new_cache(64)
+        ref = m.prefill(toks, c1)
+        m2 = LlamaModel(cfg, device="cpu", seed=3).init_random().quantize_fp8()
+        c2 = m2.new_cache(64)
+        got = m2.prefill(toks, c2)
+        cos = torch.nn.functional.cosine_similarity(
+            ref.unsqueeze(0), got.unsqueeze(0)
+        ).item()
+        assert cos > 0.98, f"fp8 logits cosine {cos}"

