@@ -17,6 +17,23 @@
 #include "common.h"
 
 typedef __attribute__((__vector_size__(4 * sizeof(float)))) float f32x4v;
+typedef __attribute__((__vector_size__(2 * sizeof(float)))) float f32x2v;
+
+// hardware fp8->f32: converts byte pair (word ? bytes 2,3 : bytes 0,1)
+DEVINL float dot16_fp8(const uint4 &xv, const uint4 &wv) {
+  const int *x32 = (const int *)&xv;
+  const int *w32 = (const int *)&wv;
+  float acc = 0.f;
+#pragma unroll
+  for (int q = 0; q < 4; ++q) {
+    const f32x2v xl = __builtin_amdgcn_cvt_pk_f32_fp8(x32[q], false);
+    const f32x2v xh = __builtin_amdgcn_cvt_pk_f32_fp8(x32[q], true);
+    const f32x2v wl = __builtin_amdgcn_cvt_pk_f32_fp8(w32[q], false);
+    const f32x2v wh = __builtin_amdgcn_cvt_pk_f32_fp8(w32[q], true);
+    acc += xl[0] * wl[0] + xl[1] * wl[1] + xh[0] * wh[0] + xh[1] * wh[1];
+  }
+  return acc;
+}
 
 #define F8_M 128
 #define F8_N 128
@@ -207,27 +224,19 @@ gemv_fp8_kernel(const uint8_t *__restrict__ x, const float *__restrict__ xs,
   float acc = 0.f;
   int c = sl;
   for (; c + 48 < nc; c += 64) {
+    uint4 wv[4], xv[4];
 #pragma unroll
     for (int u = 0; u < 4; ++u) {
-      const uint8_t *wp = wr + (size_t)(c + 16 * u) * 16;
-      const uint8_t *xp = x + (size_t)(c + 16 * u) * 16;
-      uint4 wv = *(const uint4 *)wp;
-      uint4 xv = *(const uint4 *)xp;
-      const uint8_t *w8 = (const uint8_t *)&wv;
-      const uint8_t *x8 = (const uint8_t *)&xv;
-#pragma unroll
-      for (int j = 0; j < 16; ++j)
-        acc += fp8_to_f32(x8[j]) * fp8_to_f32(w8[j]);
+      wv[u] = *(const uint4 *)(wr + (size_t)(c + 16 * u) * 16);
+      xv[u] = *(const uint4 *)(x + (size_t)(c + 16 * u) * 16);
     }
+#pragma unroll
+    for (int u = 0; u < 4; ++u) acc += dot16_fp8(xv[u], wv[u]);
   }
   for (; c < nc; c += 16) {
-    uint4 wv = *(const uint4 *)(wr + (size_t)c * 16);
-    uint4 xv = *(const uint4 *)(x + (size_t)c * 16);
-    const uint8_t *w8 = (const uint8_t *)&wv;
-    const uint8_t *x8 = (const uint8_t *)&xv;
-#pragma unroll
-    for (int j = 0; j < 16; ++j)
-      acc += fp8_to_f32(x8[j]) * fp8_to_f32(w8[j]);
+    const uint4 wv = *(const uint4 *)(wr + (size_t)c * 16);
+    const uint4 xv = *(const uint4 *)(x + (size_t)c * 16);
+    acc += dot16_fp8(xv, wv);
   }
 
 #pragma unroll

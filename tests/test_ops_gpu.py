@@ -351,8 +351,8 @@ class TestFp8:
         x = _bf(torch.randn(m, k)).to(DEV)
         w = _bf(torch.randn(n, k) * 0.05).to(DEV)
         wq, wsc = ops.quantize_fp8_rowwise(w)
-        got = ops.gemm_fp8(x, wq.to(DEV), wsc.to(DEV))
-        want = ops.gemm_fp8(x.cpu(), wq, wsc)  # CPU dequant reference
+        got = ops.gemm_fp8(x, wq, wsc)
+        want = ops.gemm_fp8(x.cpu(), wq.cpu(), wsc.cpu())  # CPU dequant ref
         _assert_close(got, want.to(torch.bfloat16), atol=8e-2, rtol=5e-2,
                       name=f"gemm_fp8 m={m}")
 
