@@ -375,6 +375,8 @@ class TestFp8Engine:
     def test_fp8_generate_close_to_bf16(self):
         """fp8 opponent produces a similar logits trajectory: compare the
         first greedy token with the bf16 engine (same seed/name)."""
+        from adversarial_spec_amd.engine.local import LocalEngine
+
         e_bf = LocalEngine({"name": "f8cmp", "arch": "debug-1b"}, device=DEV)
         e_f8 = LocalEngine({"name": "f8cmp", "arch": "debug-1b",
                             "dtype": "fp8"}, device=DEV)
