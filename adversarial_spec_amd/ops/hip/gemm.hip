@@ -33,16 +33,6 @@ typedef __attribute__((__vector_size__(4 * sizeof(float)))) float f32x4v;
 #define GT_N 128
 #define GT_K 64  // elements; one row of the LDS image = 128 B
 
-
-// XCD-aware workgroup remap (bijective; guide formula): consecutive
-// logical tiles land on the SAME XCD so column-neighbor tiles share that
-// XCD's L2 slice of A — +10% on HBM-bound shapes at this size.
-DEVINL int xcd_remap(int orig, int nwg) {
-  const int xcd = orig & 7;
-  const int q = nwg >> 3, r = nwg & 7;
-  return (xcd < r ? xcd * (q + 1) : r * (q + 1) + (xcd - r) * q) + (orig >> 3);
-}
-
 DEVINL void glds16(const ushort_t *g, ushort_t *l) {
   __builtin_amdgcn_global_load_lds(
       (const __attribute__((address_space(1))) void *)g,
@@ -105,10 +95,8 @@ gemm_nt_kernel(const ushort_t *__restrict__ a, const ushort_t *__restrict__ b,
   const int lrow = lane & 15;
   const int lhi = lane >> 4;
 
-  const int nwg = gridDim.x * gridDim.y;
-  const int flat = xcd_remap(blockIdx.y * gridDim.x + blockIdx.x, nwg);
-  const int n0 = (flat % gridDim.x) * GT_N;
-  const int m0 = (flat / gridDim.x) * GT_M;
+  const int n0 = blockIdx.x * GT_N;
+  const int m0 = blockIdx.y * GT_M;
   const int wm = (wid >> 1) * 64;  // wave quadrant
   const int wn = (wid & 1) * 64;
 
