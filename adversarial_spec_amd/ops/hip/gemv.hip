@@ -37,16 +37,18 @@ gemv_kernel(const ushort_t *__restrict__ x, const ushort_t *__restrict__ w,
 
   float acc = 0.f;
   int c = sl;
-  // 4-deep unroll: 4 independent 16 B row reads + 4 x reads in flight
-  for (; c + 48 < nc; c += 64) {
-    bf16x8 wv[4], xv[4];
+  // 8-deep unroll: 8 independent 16 B row reads in flight per lane (the
+  // small-N projections run ~1 wave/SIMD, so per-lane ILP is the only
+  // latency hiding available)
+  for (; c + 112 < nc; c += 128) {
+    bf16x8 wv[8], xv[8];
 #pragma unroll
-    for (int u = 0; u < 4; ++u) {
+    for (int u = 0; u < 8; ++u) {
       wv[u] = ((const bf16x8 *)wr)[c + 16 * u];
       xv[u] = ((const bf16x8 *)x)[c + 16 * u];
     }
 #pragma unroll
-    for (int u = 0; u < 4; ++u)
+    for (int u = 0; u < 8; ++u)
 #pragma unroll
       for (int j = 0; j < 8; ++j)
         acc += bf16_to_f32(xv[u].u[j]) * bf16_to_f32(wv[u].u[j]);
@@ -63,7 +65,55 @@ gemv_kernel(const ushort_t *__restrict__ x, const ushort_t *__restrict__ w,
   if (sl == 0) y[n] = f32_to_bf16(sum);
 }
 
+// 32-lanes-per-row variant for SMALL N (o/qkv-sized): the 16-lane kernel
+// leaves only ~1 wave/SIMD there, so halve the rows per wave and double
+// both the k-parallelism per row and the block count.
+extern "C" __global__ void __launch_bounds__(256)
+gemv_kernel_w32(const ushort_t *__restrict__ x, const ushort_t *__restrict__ w,
+                ushort_t *__restrict__ y, int K, int N) {
+  const int lane = threadIdx.x & (WAVE - 1);
+  const int wid = threadIdx.x / WAVE;
+  const int rg = lane >> 5;       // row group within wave: 0..1
+  const int sl = lane & 31;       // k-slice lane: 0..31
+  const int n = blockIdx.x * 8 + wid * 2 + rg;
+  if (n >= N) return;
+
+  const ushort_t *wr = w + (size_t)n * K;
+  const int nc = K / 8;
+
+  float acc = 0.f;
+  int c = sl;
+  for (; c + 224 < nc; c += 256) {
+    bf16x8 wv[8], xv[8];
+#pragma unroll
+    for (int u = 0; u < 8; ++u) {
+      wv[u] = ((const bf16x8 *)wr)[c + 32 * u];
+      xv[u] = ((const bf16x8 *)x)[c + 32 * u];
+    }
+#pragma unroll
+    for (int u = 0; u < 8; ++u)
+#pragma unroll
+      for (int j = 0; j < 8; ++j)
+        acc += bf16_to_f32(xv[u].u[j]) * bf16_to_f32(wv[u].u[j]);
+  }
+  for (; c < nc; c += 32) {
+    const bf16x8 wv = ((const bf16x8 *)wr)[c];
+    const bf16x8 xv = ((const bf16x8 *)x)[c];
+#pragma unroll
+    for (int j = 0; j < 8; ++j)
+      acc += bf16_to_f32(xv.u[j]) * bf16_to_f32(wv.u[j]);
+  }
+
+#pragma unroll
+  for (int off = 16; off > 0; off >>= 1) acc += __shfl_xor(acc, off, WAVE);
+  if (sl == 0) y[n] = f32_to_bf16(acc);
+}
+
 extern "C" void launch_gemv(const ushort_t *x, const ushort_t *w, ushort_t *y,
                             int K, int N, hipStream_t stream) {
-  gemv_kernel<<<dim3((N + 15) / 16), 256, 0, stream>>>(x, w, y, K, N);
+  if (N <= 8192) {
+    gemv_kernel_w32<<<dim3((N + 7) / 8), 256, 0, stream>>>(x, w, y, K, N);
+  } else {
+    gemv_kernel<<<dim3((N + 15) / 16), 256, 0, stream>>>(x, w, y, K, N);
+  }
 }
