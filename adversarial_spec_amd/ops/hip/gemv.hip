@@ -37,18 +37,17 @@ gemv_kernel(const ushort_t *__restrict__ x, const ushort_t *__restrict__ w,
 
   float acc = 0.f;
   int c = sl;
-  // 8-deep unroll: 8 independent 16 B row reads in flight per lane (the
-  // small-N projections run ~1 wave/SIMD, so per-lane ILP is the only
-  // latency hiding available)
-  for (; c + 112 < nc; c += 128) {
-    bf16x8 wv[8], xv[8];
+  // 4-deep unroll (this kernel only runs at large N where the grid itself
+  // saturates HBM; 8-deep measured slower there)
+  for (; c + 48 < nc; c += 64) {
+    bf16x8 wv[4], xv[4];
 #pragma unroll
-    for (int u = 0; u < 8; ++u) {
+    for (int u = 0; u < 4; ++u) {
       wv[u] = ((const bf16x8 *)wr)[c + 16 * u];
       xv[u] = ((const bf16x8 *)x)[c + 16 * u];
     }
 #pragma unroll
-    for (int u = 0; u < 8; ++u)
+    for (int u = 0; u < 4; ++u)
 #pragma unroll
       for (int j = 0; j < 8; ++j)
         acc += bf16_to_f32(xv[u].u[j]) * bf16_to_f32(wv[u].u[j]);
