@@ -67,15 +67,11 @@ attn_prefill_mfma_kernel(const ushort_t *__restrict__ q,
     const ushort_t *qr = q + ((size_t)(ok ? qrow : 0) * hq + h) * HD;
 #pragma unroll
     for (int c = 0; c < 4; ++c) {
-      bf16x8 t;
       if (ok) {
-        t = ((const bf16x8 *)qr)[c * 4 + lhi];  // dims c*32 + lhi*8 .. +8
+        qfrag[c] = ((const bf16x8v *)qr)[c * 4 + lhi];  // dims c*32+lhi*8..
       } else {
-#pragma unroll
-        for (int j = 0; j < 8; ++j) t.u[j] = 0;
+        qfrag[c] = (bf16x8v){0, 0, 0, 0, 0, 0, 0, 0};
       }
-#pragma unroll
-      for (int j = 0; j < 8; ++j) qfrag[c][j] = (short)t.u[j];
     }
   }
 
@@ -111,10 +107,15 @@ attn_prefill_mfma_kernel(const ushort_t *__restrict__ q,
         unsigned kbyte = (unsigned)flat * 16u;
         kbyte ^= ((unsigned)(krow & 7)) << 4;
         *(bf16x8 *)((char *)ldsK + kbyte) = kk;
-        // Vt: 8 scalar transposed stores (staging-only cost)
+        // Vt: 8 scalar transposed stores. Lane-ROTATED store order: all
+        // 16 lanes of a write group share d0%... bank (8-row stride x 80 B
+        // pitch = bank step 0 mod 32 -> 16-way conflict, PMC: 3.2
+        // conflict-cycles/busy-cycle); rotating the element index by the
+        // lane spreads each step over 8 banks (2-way).
         const int d0 = kcol8 * 8;
 #pragma unroll
-        for (int j = 0; j < 8; ++j) {
+        for (int jj = 0; jj < 8; ++jj) {
+          const int j = (jj + kcol8) & 7;
           ldsVt[(size_t)(d0 + j) * VPITCH + krow] = vv.u[j];
         }
       }
@@ -132,10 +133,7 @@ attn_prefill_mfma_kernel(const ushort_t *__restrict__ q,
         // B frag: B[kd][col] = K[keyr][c*32+kd8] from the swizzled image
         unsigned kbyte = (unsigned)keyr * 256u + (unsigned)(c * 4 + lhi) * 16u;
         kbyte ^= ((unsigned)(keyr & 7)) << 4;
-        const bf16x8 t = *(const bf16x8 *)((const char *)ldsK + kbyte);
-        bf16x8v bfr;
-#pragma unroll
-        for (int j = 0; j < 8; ++j) bfr[j] = (short)t.u[j];
+        const bf16x8v bfr = *(const bf16x8v *)((const char *)ldsK + kbyte);
         s[n] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(qfrag[c], bfr, s[n], 0, 0, 0);
       }
     }
@@ -190,21 +188,13 @@ attn_prefill_mfma_kernel(const ushort_t *__restrict__ q,
     }
 
     // ---- O += P V ----
-    bf16x8v pfrag;
-    {
-      const bf16x8 t = *(const bf16x8 *)(ldsP + lrow * VPITCH + lhi * 8);
-#pragma unroll
-      for (int j = 0; j < 8; ++j) pfrag[j] = (short)t.u[j];
-    }
+    const bf16x8v pfrag = *(const bf16x8v *)(ldsP + lrow * VPITCH + lhi * 8);
 #pragma unroll
     for (int d = 0; d < 8; ++d) {
       // B frag: B[kd][col] = V[kd8][d*16+col] = Vt[d*16+col][kd8] — one
       // contiguous 16B LDS read per lane.
-      const bf16x8 t =
-          *(const bf16x8 *)(ldsVt + (size_t)(d * 16 + lrow) * VPITCH + lhi * 8);
-      bf16x8v vfr;
-#pragma unroll
-      for (int j = 0; j < 8; ++j) vfr[j] = (short)t.u[j];
+      const bf16x8v vfr =
+          *(const bf16x8v *)(ldsVt + (size_t)(d * 16 + lrow) * VPITCH + lhi * 8);
       o[d] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(pfrag, vfr, o[d], 0, 0, 0);
     }
   }

@@ -146,17 +146,13 @@ gemm_nt_kernel(const ushort_t *__restrict__ a, const ushort_t *__restrict__ b,
       for (int i = 0; i < 4; ++i) {
         const int row = wm + i * 16 + lrow;
         const int ch = (s * 4 + lhi) ^ (row & 7);
-        const bf16x8 t = ((const bf16x8 *)(imgA[buf] + (size_t)row * GT_K))[ch];
-#pragma unroll
-        for (int j = 0; j < 8; ++j) afr[i][j] = (short)t.u[j];
+        afr[i] = ((const bf16x8v *)(imgA[buf] + (size_t)row * GT_K))[ch];
       }
 #pragma unroll
       for (int j2 = 0; j2 < 4; ++j2) {
         const int row = wn + j2 * 16 + lrow;
         const int ch = (s * 4 + lhi) ^ (row & 7);
-        const bf16x8 t = ((const bf16x8 *)(imgB[buf] + (size_t)row * GT_K))[ch];
-#pragma unroll
-        for (int j = 0; j < 8; ++j) bfr[j2][j] = (short)t.u[j];
+        bfr[j2] = ((const bf16x8v *)(imgB[buf] + (size_t)row * GT_K))[ch];
       }
 #pragma unroll
       for (int i = 0; i < 4; ++i)
