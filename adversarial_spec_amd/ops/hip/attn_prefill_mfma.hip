@@ -107,16 +107,18 @@ attn_prefill_mfma_kernel(const ushort_t *__restrict__ q,
         unsigned kbyte = (unsigned)flat * 16u;
         kbyte ^= ((unsigned)(krow & 7)) << 4;
         *(bf16x8 *)((char *)ldsK + kbyte) = kk;
-        // Vt: 8 scalar transposed stores. Lane-ROTATED store order: all
-        // 16 lanes of a write group share d0%... bank (8-row stride x 80 B
-        // pitch = bank step 0 mod 32 -> 16-way conflict, PMC: 3.2
-        // conflict-cycles/busy-cycle); rotating the element index by the
-        // lane spreads each step over 8 banks (2-way).
+        // Vt: 8 scalar transposed stores, with the KEY SLOT rotated by a
+        // dim-derived amount: the raw pattern puts all 16 lanes of a write
+        // group on one bank (8-row x 80 B stride = bank step 0 mod 32 ->
+        // 16-way conflict; PMC: 3.2 conflict-cycles/busy-cycle). Rotating
+        // krow by 8*(kcol8&3) spreads the group over 4 banks with purely
+        // static register indexing; reads below de-rotate and each 16 B
+        // vector stays inside an 8-aligned 32-slot window (no wrap).
         const int d0 = kcol8 * 8;
+        const int rot = (kcol8 & 3) * 8;
 #pragma unroll
-        for (int jj = 0; jj < 8; ++jj) {
-          const int j = (jj + kcol8) & 7;
-          ldsVt[(size_t)(d0 + j) * VPITCH + krow] = vv.u[j];
+        for (int j = 0; j < 8; ++j) {
+          ldsVt[(size_t)(d0 + j) * VPITCH + ((krow + rot) & 31)] = vv.u[j];
         }
       }
       __syncthreads();
@@ -193,8 +195,10 @@ attn_prefill_mfma_kernel(const ushort_t *__restrict__ q,
     for (int d = 0; d < 8; ++d) {
       // B frag: B[kd][col] = V[kd8][d*16+col] = Vt[d*16+col][kd8] — one
       // contiguous 16B LDS read per lane.
-      const bf16x8v vfr =
-          *(const bf16x8v *)(ldsVt + (size_t)(d * 16 + lrow) * VPITCH + lhi * 8);
+      const int vdim = d * 16 + lrow;
+      const int vrot = (((unsigned)vdim >> 3) & 3) * 8;
+      const bf16x8v vfr = *(const bf16x8v *)(
+          ldsVt + (size_t)vdim * VPITCH + ((lhi * 8 + vrot) & 31));
       o[d] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(pfrag, vfr, o[d], 0, 0, 0);
     }
   }
