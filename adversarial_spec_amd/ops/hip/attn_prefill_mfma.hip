@@ -28,10 +28,10 @@ typedef __attribute__((__vector_size__(8 * sizeof(short)))) short bf16x8v;
 typedef __attribute__((__vector_size__(4 * sizeof(float)))) float f32x4v;
 
 #define QROWS 16   // q rows per wave
-#define KVBLK 32   // keys per staged tile
+#define KVBLK 64   // keys per staged tile (64: halves softmax passes+syncs)
 #define NWAVE 8    // waves per block
 #define HD 128     // head dim (this kernel is hd=128 only)
-#define VPITCH 40  // padded row length (elements) of the Vt and P images
+#define VPITCH 72  // padded row length (elements) of the Vt and P images
 
 extern "C" __global__ void __launch_bounds__(512, 1)
 attn_prefill_mfma_kernel(const ushort_t *__restrict__ q,
@@ -52,8 +52,8 @@ attn_prefill_mfma_kernel(const ushort_t *__restrict__ q,
 
   __shared__ __attribute__((aligned(16))) ushort_t lds[
       KVBLK * HD + HD * VPITCH + NWAVE * QROWS * VPITCH];
-  ushort_t *ldsK = lds;                       // swizzled [32][128]
-  ushort_t *ldsVt = lds + KVBLK * HD;         // [128][40]
+  ushort_t *ldsK = lds;                       // swizzled [64][128]
+  ushort_t *ldsVt = lds + KVBLK * HD;         // [128][72]
   ushort_t *ldsP = ldsVt + HD * VPITCH + wid * QROWS * VPITCH;
 
   const int lrow = lane & 15;       // 0..15
@@ -87,10 +87,12 @@ attn_prefill_mfma_kernel(const ushort_t *__restrict__ q,
   if (causal) kmax = min(tk, kv_offset + qb0 + NWAVE * QROWS);
 
   for (int kt = 0; kt < kmax; kt += KVBLK) {
-    // ---- stage K (swizzled) and Vt (transposed, padded) ----
-    {
-      const int flat = tid;             // one 16B K piece each
-      const int krow = flat >> 4;       // key within tile
+    // ---- stage K (swizzled) and Vt (transposed, padded): 64 keys ----
+    __syncthreads();  // previous iteration's reads done
+#pragma unroll
+    for (int piece = 0; piece < 2; ++piece) {
+      const int flat = tid + 512 * piece;
+      const int krow = flat >> 4;       // key within tile: 0..63
       const int kcol8 = flat & 15;      // 16B chunk within the 256B row
       const int key = kt + krow;
       bf16x8 kk, vv;
@@ -101,33 +103,28 @@ attn_prefill_mfma_kernel(const ushort_t *__restrict__ q,
 #pragma unroll
         for (int j = 0; j < 8; ++j) { kk.u[j] = 0; vv.u[j] = 0; }
       }
-      __syncthreads();  // previous iteration's reads done
-      {
-        // K: swizzled 16B store
-        unsigned kbyte = (unsigned)flat * 16u;
-        kbyte ^= ((unsigned)(krow & 7)) << 4;
-        *(bf16x8 *)((char *)ldsK + kbyte) = kk;
-        // Vt: 8 scalar transposed stores, with the KEY SLOT rotated by a
-        // dim-derived amount: the raw pattern puts all 16 lanes of a write
-        // group on one bank (8-row x 80 B stride = bank step 0 mod 32 ->
-        // 16-way conflict; PMC: 3.2 conflict-cycles/busy-cycle). Rotating
-        // krow by 8*(kcol8&3) spreads the group over 4 banks with purely
-        // static register indexing; reads below de-rotate and each 16 B
-        // vector stays inside an 8-aligned 32-slot window (no wrap).
-        const int d0 = kcol8 * 8;
-        const int rot = (kcol8 & 3) * 8;
+      // K: swizzled 16B store
+      unsigned kbyte = (unsigned)flat * 16u;
+      kbyte ^= ((unsigned)(krow & 7)) << 4;
+      *(bf16x8 *)((char *)ldsK + kbyte) = kk;
+      // Vt: 8 scalar transposed stores with the KEY SLOT rotated by a
+      // dim-derived amount (raw pattern = one bank for a whole write
+      // group; rotation spreads it 8-wide with static indexing; reads
+      // de-rotate and each 16 B vector stays in an 8-aligned 64-slot
+      // window, no wrap).
+      const int d0 = kcol8 * 8;
+      const int rot = (kcol8 & 7) * 8;
 #pragma unroll
-        for (int j = 0; j < 8; ++j) {
-          ldsVt[(size_t)(d0 + j) * VPITCH + ((krow + rot) & 31)] = vv.u[j];
-        }
+      for (int j = 0; j < 8; ++j) {
+        ldsVt[(size_t)(d0 + j) * VPITCH + ((krow + rot) & 63)] = vv.u[j];
       }
-      __syncthreads();
     }
+    __syncthreads();
 
-    // ---- S = Q K^T for two 16-key subtiles ----
-    f32x4v s[2];
+    // ---- S = Q K^T for four 16-key subtiles ----
+    f32x4v s[4];
 #pragma unroll
-    for (int n = 0; n < 2; ++n) {
+    for (int n = 0; n < 4; ++n) {
       s[n] = (f32x4v){0.f, 0.f, 0.f, 0.f};
       const int keyr = n * 16 + lrow;
 #pragma unroll
@@ -146,13 +143,13 @@ attn_prefill_mfma_kernel(const ushort_t *__restrict__ q,
     for (int r = 0; r < 4; ++r) {
       const int qrow_abs = kv_offset + q0 + lhi * 4 + r;
 #pragma unroll
-      for (int n = 0; n < 2; ++n) {
+      for (int n = 0; n < 4; ++n) {
         const int key = kt + n * 16 + lrow;
         float sv = s[n][r] * scale;
         if (key >= tk || (causal && key > qrow_abs)) sv = -INFINITY;
         s[n][r] = sv;
       }
-      float mx = fmaxf(s[0][r], s[1][r]);
+      float mx = fmaxf(fmaxf(s[0][r], s[1][r]), fmaxf(s[2][r], s[3][r]));
 #pragma unroll
       for (int off = 8; off > 0; off >>= 1)
         mx = fmaxf(mx, __shfl_xor(mx, off, WAVE));
@@ -166,7 +163,7 @@ attn_prefill_mfma_kernel(const ushort_t *__restrict__ q,
           (m[r] == -INFINITY && mn == -INFINITY) ? 0.f : __expf(m[r] - mn);
       float psum = 0.f;
 #pragma unroll
-      for (int n = 0; n < 2; ++n) {
+      for (int n = 0; n < 4; ++n) {
         const float p = (s[n][r] == -INFINITY) ? 0.f : __expf(s[n][r] - mn);
         s[n][r] = p;
         psum += p;
@@ -182,24 +179,29 @@ attn_prefill_mfma_kernel(const ushort_t *__restrict__ q,
 
     // ---- P -> LDS (C layout -> A layout bounce; padded pitch) ----
 #pragma unroll
-    for (int n = 0; n < 2; ++n) {
+    for (int n = 0; n < 4; ++n) {
 #pragma unroll
       for (int r = 0; r < 4; ++r) {
         ldsP[(lhi * 4 + r) * VPITCH + n * 16 + lrow] = f32_to_bf16(s[n][r]);
       }
     }
 
-    // ---- O += P V ----
-    const bf16x8v pfrag = *(const bf16x8v *)(ldsP + lrow * VPITCH + lhi * 8);
+    // ---- O += P V (two 32-key halves) ----
 #pragma unroll
-    for (int d = 0; d < 8; ++d) {
-      // B frag: B[kd][col] = V[kd8][d*16+col] = Vt[d*16+col][kd8] — one
-      // contiguous 16B LDS read per lane.
-      const int vdim = d * 16 + lrow;
-      const int vrot = (((unsigned)vdim >> 3) & 3) * 8;
-      const bf16x8v vfr = *(const bf16x8v *)(
-          ldsVt + (size_t)vdim * VPITCH + ((lhi * 8 + vrot) & 31));
-      o[d] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(pfrag, vfr, o[d], 0, 0, 0);
+    for (int half = 0; half < 2; ++half) {
+      const bf16x8v pfrag =
+          *(const bf16x8v *)(ldsP + lrow * VPITCH + half * 32 + lhi * 8);
+#pragma unroll
+      for (int d = 0; d < 8; ++d) {
+        // B frag: B[kd][col] = V[kd8][d*16+col] = Vt[d*16+col][kd8] — one
+        // contiguous 16B LDS read per lane.
+        const int vdim = d * 16 + lrow;
+        const int vrot = (((unsigned)vdim >> 3) & 7) * 8;
+        const bf16x8v vfr = *(const bf16x8v *)(
+            ldsVt + (size_t)vdim * VPITCH +
+            ((half * 32 + lhi * 8 + vrot) & 63));
+        o[d] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(pfrag, vfr, o[d], 0, 0, 0);
+      }
     }
   }
 
