@@ -53,8 +53,11 @@ def parse_args():
     p.add_argument("--warmup", type=int, default=1, help="untimed rounds")
     p.add_argument("--opponents-per-gpu", type=int, default=3)
     p.add_argument("--spec-tokens", type=int, default=8192)
-    p.add_argument("--decode-tokens", type=int, default=128,
-                   help="critique length decoded per opponent per round")
+    p.add_argument("--decode-tokens", type=int, default=512,
+                   help="critique length decoded per opponent per round "
+                        "(the reference caps critiques at 8000 tokens, "
+                        "models.py:620; 512 is a representative critique+"
+                        "[SPEC] revision length for the headline round)")
     p.add_argument("--model", default="llama-3-8b")
     p.add_argument("--arch-mix", default=None,
                    help="comma list of archs cycled over opponents for the "
