@@ -89,11 +89,28 @@ __global__ void __launch_bounds__(256) attn_decode_split_kernel(
     for (int j = 0; j < 8; ++j) acc[gi][j] = 0.f;
   }
 
-  for (int p = start + sub; p < limit; p += subs) {
-    const int phys = page_table[p / page];
-    const size_t row = ((size_t)phys * page + (p % page)) * kh * hd;
-    const f32x8 kd = unpack8(((const bf16x8 *)(kc + row + (size_t)g * hd))[sl]);
-    const f32x8 vd = unpack8(((const bf16x8 *)(vc + row + (size_t)g * hd))[sl]);
+  // software prefetch: next position's K/V rows are loaded while the
+  // current position's softmax update computes (the serial m/l/acc chain
+  // otherwise exposes full HBM latency every iteration).
+  int p = start + sub;
+  bf16x8 kraw, vraw;
+  if (p < limit) {
+    const int phys0 = page_table[p / page];
+    const size_t row0 = ((size_t)phys0 * page + (p % page)) * kh * hd;
+    kraw = ((const bf16x8 *)(kc + row0 + (size_t)g * hd))[sl];
+    vraw = ((const bf16x8 *)(vc + row0 + (size_t)g * hd))[sl];
+  }
+  for (; p < limit; ) {
+    const int pn = p + subs;
+    bf16x8 kn, vn;
+    if (pn < limit) {
+      const int physn = page_table[pn / page];
+      const size_t rown = ((size_t)physn * page + (pn % page)) * kh * hd;
+      kn = ((const bf16x8 *)(kc + rown + (size_t)g * hd))[sl];
+      vn = ((const bf16x8 *)(vc + rown + (size_t)g * hd))[sl];
+    }
+    const f32x8 kd = unpack8(kraw);
+    const f32x8 vd = unpack8(vraw);
 
 #pragma unroll
     for (int gi = 0; gi < MG; ++gi) {
@@ -111,6 +128,9 @@ __global__ void __launch_bounds__(256) attn_decode_split_kernel(
         acc[gi][j] = acc[gi][j] * alpha + pex * vd.v[j];
       m[gi] = m_new;
     }
+    kraw = kn;
+    vraw = vn;
+    p = pn;
   }
 
   // merge position sub-groups within the wave (lanes xor LPP, 2*LPP, ...)
@@ -206,42 +226,45 @@ attn_decode_combine_kernel(const float *__restrict__ ws_m,
   const int dd = threadIdx.x;
   if (dd >= hd) return;
 
-  // 4-way unrolled split walk keeps >=4 independent loads in flight
-  // (the serial version was latency-bound at large split counts).
+  // 8-way unrolled split walk keeps >=8 independent loads in flight
+  // (the walk is latency-bound: one thread per (head, dim) output).
   float M = -INFINITY;
   {
-    float m4[4] = {-INFINITY, -INFINITY, -INFINITY, -INFINITY};
+    float m8[8] = {-INFINITY, -INFINITY, -INFINITY, -INFINITY,
+                   -INFINITY, -INFINITY, -INFINITY, -INFINITY};
     int s = 0;
-    for (; s + 4 <= n_splits; s += 4) {
+    for (; s + 8 <= n_splits; s += 8) {
 #pragma unroll
-      for (int u = 0; u < 4; ++u)
-        m4[u] = fmaxf(m4[u], ws_m[((size_t)g * n_splits + s + u) * group + gi]);
+      for (int u = 0; u < 8; ++u)
+        m8[u] = fmaxf(m8[u], ws_m[((size_t)g * n_splits + s + u) * group + gi]);
     }
     for (; s < n_splits; ++s)
-      m4[0] = fmaxf(m4[0], ws_m[((size_t)g * n_splits + s) * group + gi]);
-    M = fmaxf(fmaxf(m4[0], m4[1]), fmaxf(m4[2], m4[3]));
-  }
-  float L4[4] = {0.f, 0.f, 0.f, 0.f}, A4[4] = {0.f, 0.f, 0.f, 0.f};
-  int s = 0;
-  for (; s + 4 <= n_splits; s += 4) {
+      m8[0] = fmaxf(m8[0], ws_m[((size_t)g * n_splits + s) * group + gi]);
 #pragma unroll
-    for (int u = 0; u < 4; ++u) {
+    for (int u = 0; u < 8; ++u) M = fmaxf(M, m8[u]);
+  }
+  float L8[8] = {0.f}, A8[8] = {0.f};
+  int s = 0;
+  for (; s + 8 <= n_splits; s += 8) {
+#pragma unroll
+    for (int u = 0; u < 8; ++u) {
       const size_t base = ((size_t)g * n_splits + s + u) * group + gi;
       const float mw = ws_m[base];
       const float sc = (mw == -INFINITY) ? 0.f : __expf(mw - M);
-      L4[u] += ws_l[base] * sc;
-      A4[u] += ws_acc[base * hd + dd] * sc;
+      L8[u] += ws_l[base] * sc;
+      A8[u] += ws_acc[base * hd + dd] * sc;
     }
   }
   for (; s < n_splits; ++s) {
     const size_t base = ((size_t)g * n_splits + s) * group + gi;
     const float mw = ws_m[base];
     const float sc = (mw == -INFINITY) ? 0.f : __expf(mw - M);
-    L4[0] += ws_l[base] * sc;
-    A4[0] += ws_acc[base * hd + dd] * sc;
+    L8[0] += ws_l[base] * sc;
+    A8[0] += ws_acc[base * hd + dd] * sc;
   }
-  const float L = L4[0] + L4[1] + L4[2] + L4[3];
-  const float A = A4[0] + A4[1] + A4[2] + A4[3];
+  float L = 0.f, A = 0.f;
+#pragma unroll
+  for (int u = 0; u < 8; ++u) { L += L8[u]; A += A8[u]; }
   out[(size_t)h * hd + dd] = f32_to_bf16(L > 0.f ? A / L : 0.f);
 }
 

@@ -77,9 +77,28 @@ def probe_decode():
               f"{tm['decode']/max(otok,1):.2f} ms/tok), wall {wall:.2f} s")
 
 
+
+
+def probe_decode_attn():
+    from adversarial_spec_amd.ops import _advspec_hip as hip
+
+    kh, hd, group = 8, 128, 4
+    page = 256
+    for seq in (8192, 16384):
+        npg = (seq + page - 1) // page + 1
+        kc = torch.randn(npg, page, kh, hd, dtype=torch.bfloat16, device="cuda")
+        vc = torch.randn_like(kc)
+        pt = torch.arange(npg, dtype=torch.int32, device="cuda")
+        q = torch.randn(kh * group, hd, dtype=torch.bfloat16, device="cuda")
+        t = bench_gpu(lambda: hip.attn_decode_paged(q, kc, vc, pt, seq, 0.088),
+                      iters=100)
+        by = seq * kh * hd * 2 * 2
+        print(f"decode-attn seq={seq}: {t*1e6:7.1f} us {by/t/1e12:5.2f} TB/s")
+
 if __name__ == "__main__":
     assert torch.cuda.is_available()
     torch.manual_seed(0)
     probe_gemv()
     probe_prefill_attn()
+    probe_decode_attn()
     probe_decode()
