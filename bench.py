@@ -139,16 +139,24 @@ def main() -> int:
     pool = ThreadPoolExecutor(max_workers=max(1, n_opp))
 
     def run_one(eng):
-        text, _in, _out, _tm = eng.generate(
-            system_prompt,
-            user_message,
-            max_tokens=args.decode_tokens,
-            temperature=args.temperature,
-            timeout=600.0,
-        )
-        ids = eng.tokenizer.encode(text)[: max_gather_tokens]
-        agreed = detect_agreement(text)
-        return ids, agreed
+        # per-opponent fault isolation (reference semantics: an errored
+        # opponent degrades the round, never aborts it — and on the
+        # distributed path an un-launched collective would DEADLOCK the
+        # other ranks, so errors must still reach the gather)
+        try:
+            text, _in, _out, _tm = eng.generate(
+                system_prompt,
+                user_message,
+                max_tokens=args.decode_tokens,
+                temperature=args.temperature,
+                timeout=600.0,
+            )
+            ids = eng.tokenizer.encode(text)[: max_gather_tokens]
+            agreed = detect_agreement(text)
+            return ids, agreed, False
+        except Exception as e:  # noqa: BLE001
+            print(f"opponent {eng.name} failed: {e}", file=sys.stderr)
+            return [], False, True
 
     def one_round():
         """One debate round for this rank's opponents + consensus gather.
@@ -162,12 +170,12 @@ def main() -> int:
         local = []
         futures = [pool.submit(run_one, eng) for eng in engines]
         for i, fut in enumerate(futures):
-            ids, agreed = fut.result()
+            ids, agreed, error = fut.result()
             if gathers is not None:
-                gathers[i].launch(ids, agreed, False)
+                gathers[i].launch(ids, agreed, error)
             else:
                 local.append(
-                    pack_result(ids, agreed, False, max_gather_tokens,
+                    pack_result(ids, agreed, error, max_gather_tokens,
                                 torch.device(device))
                 )
         if gathers is not None:
