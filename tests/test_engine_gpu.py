@@ -119,3 +119,44 @@ class TestGenerationGPU:
         div = next((i for i, (p, q) in enumerate(zip(a, b)) if p != q), None)
         assert a == b, f"diverge at {div}: eager={a} graph={b}"
         assert b[2] > 0
+
+
+class TestEngineBehaviorGPU:
+    def test_deadline_returns_partial(self):
+        """--timeout semantics: the decode deadline returns a PARTIAL
+        critique instead of raising (reference translates provider timeouts
+        to retries; locally a partial decode is still usable)."""
+        import time
+
+        eng = LocalEngine({"name": "dl", "arch": "debug-1b"}, device=DEV)
+        t0 = time.monotonic()
+        text, in_tok, out_tok, tm = eng.generate(
+            "s", "deadline test", max_tokens=8000, temperature=0.7,
+            timeout=1.5,
+        )
+        wall = time.monotonic() - t0
+        assert out_tok > 0
+        assert out_tok < 8000  # stopped early
+        assert wall < 30
+
+    def test_long_context_generate(self):
+        """16k-token prompt exercises the >8k cache bucket + split
+        geometry used by BASELINE config 4."""
+        eng = LocalEngine({"name": "lc", "arch": "debug-1b"}, device=DEV)
+        long_user = "spec line\n" * 4000  # ~16k byte tokens
+        text, in_tok, out_tok, tm = eng.generate(
+            "sys", long_user, max_tokens=16, temperature=0.0, timeout=300,
+        )
+        assert in_tok > 8192
+        assert out_tok > 0
+
+    def test_rounds_reuse_cache_and_graph(self):
+        """Three rounds on one engine: cache+graph reuse must not leak
+        state across rounds (greedy round 1 == greedy round 3 for the
+        same prompt)."""
+        eng = LocalEngine({"name": "rr", "arch": "debug-1b"}, device=DEV)
+        outs = []
+        for prompt in ("round A", "round B", "round A"):
+            outs.append(eng.generate("s", prompt, max_tokens=12,
+                                     temperature=0.0, timeout=120)[0])
+        assert outs[0] == outs[2]
