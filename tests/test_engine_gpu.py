@@ -118,7 +118,6 @@ class TestGenerationGPU:
         a, b = captured
         div = next((i for i, (p, q) in enumerate(zip(a, b)) if p != q), None)
         assert a == b, f"diverge at {div}: eager={a} graph={b}"
-        assert b[2] > 0
 
 
 class TestEngineBehaviorGPU:
