@@ -67,7 +67,7 @@ PRESETS: dict[str, LlamaConfig] = {
     # A mid-size config for single-GPU kernel shakeout (hd=128 MFMA path).
     "debug-1b": LlamaConfig(
         name="debug-1b", dim=2048, n_layers=16, n_heads=16, n_kv_heads=8,
-        ffn_dim=8192, vocab_size=128256, max_seq_len=8192, rope_theta=500000.0,
+        ffn_dim=8192, vocab_size=128256, max_seq_len=32768, rope_theta=500000.0,
     ),
 }
 
