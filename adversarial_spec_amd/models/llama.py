@@ -354,9 +354,19 @@ class LlamaModel:
                                cache.v[i], cache.page_table, pos0,
                                pos_state=pos_state)
             if t > 1:
-                if pos0 != 0:
-                    raise NotImplementedError("chunked prefill lands with the 32k path")
-                attn = ops.attn_prefill(q, k, v, self.scale, causal=True)
+                if pos0 == 0:
+                    attn = ops.attn_prefill(q, k, v, self.scale, causal=True)
+                else:
+                    # chunked prefill: this chunk's queries attend to ALL
+                    # cached positions plus themselves. The page table is
+                    # identity (PagedKVCache allocates one pool), so the
+                    # cache view [0 : pos0+t] IS contiguous [seq, kh, hd] —
+                    # the same flash prefill kernel runs with kv_offset.
+                    kh_ = k.shape[1]
+                    k_all = cache.k[i].view(-1, kh_, hd)[: pos0 + t]
+                    v_all = cache.v[i].view(-1, kh_, hd)[: pos0 + t]
+                    attn = ops.attn_prefill(q, k_all, v_all, self.scale,
+                                            causal=True, kv_offset=pos0)
             else:
                 seq = max_seq_bound if pos_state is not None else pos0 + 1
                 attn = ops.attn_decode_paged(
@@ -383,12 +393,29 @@ class LlamaModel:
             logits = ops.gemv(normed[-1:].contiguous(), self.lm_head)  # [1, vocab]
         return logits[0]
 
-    def prefill(self, tokens: torch.Tensor, cache: PagedKVCache) -> torch.Tensor:
-        """Prefill the prompt; returns last-position logits [vocab]."""
-        if tokens.shape[0] > cache.max_seq:
-            raise ValueError(f"prompt {tokens.shape[0]} exceeds cache {cache.max_seq}")
-        logits = self._forward(tokens, cache, 0)
-        cache.seq_len = tokens.shape[0]
+    def prefill(self, tokens: torch.Tensor, cache: PagedKVCache,
+                chunk: Optional[int] = None) -> torch.Tensor:
+        """Prefill the prompt; returns last-position logits [vocab].
+
+        `chunk` caps the tokens processed per forward pass: 32k single-shot
+        prefill fits comfortably in 288 GB HBM3E, so chunking is an
+        ACTIVATION-memory valve for longer documents, not
+        the default. Chunk boundaries re-run attention against the cached
+        prefix (kv_offset), numerically identical to single-shot."""
+        n = tokens.shape[0]
+        if n > cache.max_seq:
+            raise ValueError(f"prompt {n} exceeds cache {cache.max_seq}")
+        if chunk is None or n <= chunk:
+            logits = self._forward(tokens, cache, 0)
+            cache.seq_len = n
+            return logits
+        pos = 0
+        logits = None
+        while pos < n:
+            step = min(chunk, n - pos)
+            logits = self._forward(tokens[pos : pos + step], cache, pos)
+            pos += step
+            cache.seq_len = pos
         return logits
 
     def decode_one(self, token, cache: PagedKVCache) -> torch.Tensor:

@@ -157,3 +157,39 @@ class TestFp8CPU:
             ref.unsqueeze(0), got.unsqueeze(0)
         ).item()
         assert cos > 0.98, f"fp8 logits cosine {cos}"
+
+
+class TestChunkedPrefill:
+    def test_chunked_matches_single_shot(self):
+        import torch
+
+        from adversarial_spec_amd.models import LlamaModel
+        from adversarial_spec_amd.models.config import get_config
+
+        cfg = get_config("tiny")
+        m = LlamaModel(cfg, device="cpu", seed=5).init_random()
+        toks = torch.arange(2, 120)  # 118 tokens
+        c1 = m.new_cache(256)
+        ref = m.prefill(toks, c1)
+        c2 = m.new_cache(256)
+        got = m.prefill(toks, c2, chunk=48)  # 48+48+22
+        assert torch.allclose(ref, got, atol=1e-4), (ref - got).abs().max()
+        # caches identical too
+        assert torch.allclose(c1.k[:, :, :118], c2.k[:, :, :118], atol=1e-5)
+
+    def test_chunked_then_decode(self):
+        import torch
+
+        from adversarial_spec_amd.models import LlamaModel
+        from adversarial_spec_amd.models.config import get_config
+
+        cfg = get_config("tiny")
+        m = LlamaModel(cfg, device="cpu", seed=6).init_random()
+        toks = torch.arange(3, 80)
+        c1 = m.new_cache(256)
+        m.prefill(toks, c1)
+        ref = m.decode_one(9, c1)
+        c2 = m.new_cache(256)
+        m.prefill(toks, c2, chunk=32)
+        got = m.decode_one(9, c2)
+        assert torch.allclose(ref, got, atol=1e-4)

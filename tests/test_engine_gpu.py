@@ -159,3 +159,18 @@ class TestEngineBehaviorGPU:
             outs.append(eng.generate("s", prompt, max_tokens=12,
                                      temperature=0.0, timeout=120)[0])
         assert outs[0] == outs[2]
+
+    def test_chunked_prefill_matches_single_shot_gpu(self):
+        """Chunk boundaries re-run the MFMA prefill kernel with kv_offset
+        against the cached prefix — must match single-shot bitwise-ish."""
+        m = LlamaModel(GPU_TINY, device=DEV, dtype=torch.bfloat16,
+                       seed=21).init_random()
+        toks = torch.arange(2, 300, device=DEV)  # 298 tokens
+        c1 = m.new_cache(512)
+        ref = m.prefill(toks, c1)
+        c2 = m.new_cache(512)
+        got = m.prefill(toks, c2, chunk=128)  # 128+128+42, kv_offset path
+        cos = torch.nn.functional.cosine_similarity(
+            ref.float().unsqueeze(0), got.float().unsqueeze(0)
+        ).item()
+        assert cos > 0.999, f"chunked/single cosine {cos}"
