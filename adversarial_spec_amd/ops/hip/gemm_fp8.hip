@@ -309,12 +309,55 @@ extern "C" void launch_gemm_fp8(const uint8_t *a, const float *asc,
   gemm_fp8_kernel<<<grid, 256, 0, stream>>>(a, asc, b, bsc, c, M, N, K);
 }
 
+// 32-lanes-per-row variant for SMALL N (same rationale as gemv.hip)
+extern "C" __global__ void __launch_bounds__(256)
+gemv_fp8_kernel_w32(const uint8_t *__restrict__ x, const float *__restrict__ xs,
+                    const uint8_t *__restrict__ w, const float *__restrict__ wsc,
+                    ushort_t *__restrict__ y, int K, int N) {
+  const int lane = threadIdx.x & (WAVE - 1);
+  const int wid = threadIdx.x / WAVE;
+  const int rg = lane >> 5;
+  const int sl = lane & 31;
+  const int n = blockIdx.x * 8 + wid * 2 + rg;
+  if (n >= N) return;
+
+  const uint8_t *wr = w + (size_t)n * K;
+  const int nc = K / 16;
+
+  float acc = 0.f;
+  int c = sl;
+  for (; c + 96 < nc; c += 128) {
+    uint4 wv[4], xv[4];
+#pragma unroll
+    for (int u = 0; u < 4; ++u) {
+      wv[u] = *(const uint4 *)(wr + (size_t)(c + 32 * u) * 16);
+      xv[u] = *(const uint4 *)(x + (size_t)(c + 32 * u) * 16);
+    }
+#pragma unroll
+    for (int u = 0; u < 4; ++u) acc += dot16_fp8(xv[u], wv[u]);
+  }
+  for (; c < nc; c += 32) {
+    const uint4 wv = *(const uint4 *)(wr + (size_t)c * 16);
+    const uint4 xv = *(const uint4 *)(x + (size_t)c * 16);
+    acc += dot16_fp8(xv, wv);
+  }
+
+#pragma unroll
+  for (int off = 16; off > 0; off >>= 1) acc += __shfl_xor(acc, off, WAVE);
+  if (sl == 0) y[n] = f32_to_bf16(acc * xs[0] * wsc[n]);
+}
+
 extern "C" void launch_gemv_fp8(const uint8_t *x, const float *xs,
                                 const uint8_t *w, const float *wsc,
                                 ushort_t *y, int K, int N,
                                 hipStream_t stream) {
-  gemv_fp8_kernel<<<dim3((N + 15) / 16), 256, 0, stream>>>(x, xs, w, wsc, y,
-                                                           K, N);
+  if (N <= 8192) {
+    gemv_fp8_kernel_w32<<<dim3((N + 7) / 8), 256, 0, stream>>>(x, xs, w, wsc,
+                                                               y, K, N);
+  } else {
+    gemv_fp8_kernel<<<dim3((N + 15) / 16), 256, 0, stream>>>(x, xs, w, wsc, y,
+                                                             K, N);
+  }
 }
 
 extern "C" void launch_quant_fp8(const ushort_t *x, uint8_t *q, float *scale,

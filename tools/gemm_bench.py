@@ -37,7 +37,7 @@ for M, K, N in [(8192, 4096, 6144), (8192, 4096, 28672), (56, 4096, 6144)]:
     fl = 2.0 * M * K * N
     print(f"M{M} K{K} N{N}: fp8 {t8*1e3:7.2f} ms {fl/t8/1e12:7.1f} TF/s")
 # decode gemv: bf16 vs fp8 (per-token weight streaming)
-for K, N, tag in [(4096, 28672, "gate_up"), (14336, 4096, "down"), (4096, 128256, "lm")]:
+for K, N, tag in [(4096, 6144, "qkv"), (4096, 4096, "o"), (4096, 28672, "gate_up"), (14336, 4096, "down"), (4096, 128256, "lm")]:
     x = torch.randn(1, K, device="cuda").bfloat16()
     w = (torch.randn(N, K, device="cuda") * 0.02).bfloat16()
     wq, wsc = ops.quantize_fp8_rowwise(w)
