@@ -169,19 +169,21 @@ class LocalEngine:
 
         out_ids: list[int] = []
         with timer.phase("decode"):
+            # TP decode uses the host-stepped loop (decode_one, covered by
+            # the gloo TP=2 parity tests): capturing RCCL collectives in
+            # HIP graphs is not supported reliably on this stack, and the
+            # per-layer all-reduce already bounds decode latency there.
             use_async = (
                 self.device.type == "cuda"
                 and top_p >= 1.0
+                and self.model.tp is None
                 and ops.hip_available()
             )
             if use_async:
                 # same device-state step either way; capture only pays off
                 # for longer generations (ADVSPEC_NO_GRAPH forces eager —
                 # same kernels, bitwise-identical tokens)
-                # TP decode stays eager: capturing RCCL collectives in HIP
-                # graphs is not supported reliably on this stack.
                 use_graph = (max_new >= 8
-                             and self.model.tp is None
                              and not os.environ.get("ADVSPEC_NO_GRAPH"))
                 out_ids = self._decode_graphed(
                     logits, cache, max_new, temperature, stop_ids, deadline,
