@@ -68,6 +68,10 @@ def parse_args():
     p.add_argument("--tp", type=int, default=1,
                    help="tensor-parallel degree: all ranks form ONE sharded "
                         "opponent (BASELINE config 5: llama-3-70b --tp 8)")
+    p.add_argument("--cpu-smoke", action="store_true",
+                   help="validation mode: tiny opponents on CPU over gloo — "
+                        "exercises the full multi-rank round flow (threaded "
+                        "opponents, ordered consensus gathers) without GPUs")
     return p.parse_args()
 
 
@@ -77,19 +81,26 @@ def main() -> int:
     rank = int(os.environ.get("RANK", "0"))
     local_rank = int(os.environ.get("LOCAL_RANK", "0"))
 
-    use_gpu = torch.cuda.is_available()
-    if not use_gpu:
-        print("bench.py requires an MI355X GPU", file=sys.stderr)
+    use_gpu = torch.cuda.is_available() and not args.cpu_smoke
+    if not use_gpu and not args.cpu_smoke:
+        print("bench.py requires an MI355X GPU (or --cpu-smoke)", file=sys.stderr)
         return 1
-    torch.cuda.set_device(local_rank)
-    device = f"cuda:{local_rank}"
+    if use_gpu:
+        torch.cuda.set_device(local_rank)
+        device = f"cuda:{local_rank}"
+    else:
+        device = "cpu"
+        args.model = "tiny"
+        args.arch_mix = None
+        args.spec_tokens = min(args.spec_tokens, 256)
+        args.decode_tokens = min(args.decode_tokens, 8)
 
     dist = None
     if world > 1:
         import torch.distributed as dist_mod
 
         dist = dist_mod
-        dist.init_process_group("nccl")
+        dist.init_process_group("nccl" if use_gpu else "gloo")
 
     tp_mode = args.tp > 1
     if tp_mode:
@@ -197,13 +208,15 @@ def main() -> int:
 
     if dist is not None:
         dist.barrier()
-    torch.cuda.synchronize()
+    if use_gpu:
+        torch.cuda.synchronize()
     t0 = time.perf_counter()
     for _ in range(args.steps):
         one_round()
     if dist is not None:
         dist.barrier()
-    torch.cuda.synchronize()
+    if use_gpu:
+        torch.cuda.synchronize()
     elapsed = time.perf_counter() - t0
 
     # max over ranks
