@@ -440,6 +440,9 @@ class LlamaModel:
         """Allocation-free single-token decode: token id in W.tok_long,
         logits written into W.logits. Safe to capture in a HIP graph (all
         dynamic state in device words, zero allocator traffic)."""
+        # GPU-only: the CPU op wrappers ignore out= (they return fresh
+        # tensors), which would silently drop every write into W.
+        assert self.device.type == "cuda", "decode_step_ws is GPU-only"
         c = self.config
         h, kh, hd = c.n_heads, c.n_kv_heads, c.head_dim
         fp8 = self.fp8
