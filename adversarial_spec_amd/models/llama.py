@@ -2,8 +2,10 @@
 
 Design (MI355X-first, not a port):
   - weights live as plain bf16 tensors in HBM3E (288 GB/GPU lets several
-    opponent models co-reside), pre-transposed to [in, out] so the hot
-    matmuls run as x @ W through hipBLASLt with no transposes;
+    opponent models co-reside) in ROW-MAJOR [out, in] — exactly the HF
+    checkpoint layout, zero load-time transposes; the in-tree MFMA GEMM
+    and row-dot GEMV compute x @ W^T natively (optionally e4m3-quantized
+    rowwise for the fp8 path);
   - QKV and gate/up projections are fused into single GEMMs;
   - the residual stream is updated by a fused add+RMSNorm HIP kernel (one
     HBM round-trip instead of two — the ~8 TB/s HBM is the usual bound);
@@ -221,9 +223,9 @@ class LlamaModel:
     def load_safetensors(self, path: str) -> "LlamaModel":
         """Load HF-format Llama weights from a local safetensors dir.
 
-        Fuses q/k/v and gate/up, transposes to [in, out], and permutes q/k
-        rows from HF's half-split RoPE layout to the interleaved-pair
-        convention the RoPE kernel uses.
+        Fuses q/k/v and gate/up (keeping HF's [out, in] row-major layout)
+        and permutes q/k rows from HF's half-split RoPE layout to the
+        interleaved-pair convention the RoPE kernel uses.
         """
         import glob as _glob
         import json as _json
