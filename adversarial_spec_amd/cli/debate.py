@@ -129,6 +129,9 @@ Bedrock:
                         help="Second extra argument (alias target)")
     parser.add_argument("--weights", help="Weights path for `local alias`")
     parser.add_argument("--gpu", type=int, default=None, help="GPU ordinal for `local alias`")
+    parser.add_argument("--model-dtype", choices=["bf16", "fp8"], default=None,
+                        help="Engine dtype for `local alias` (fp8 = e4m3 "
+                             "MFMA prefill + fp8 weight streaming)")
     # misc
     parser.add_argument("--timeout", type=int, default=600,
                         help="Timeout in seconds for model calls (default: 600)")
@@ -222,7 +225,7 @@ def handle_utility_command(args: argparse.Namespace) -> Optional[int]:
     if args.action == "local":
         return providers.handle_local_command(
             args.profile_name, args.bedrock_arg, args.extra_arg,
-            args.weights, args.gpu,
+            args.weights, args.gpu, args.model_dtype,
         )
     return None
 

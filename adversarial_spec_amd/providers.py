@@ -508,7 +508,8 @@ def handle_bedrock_command(subcommand: Optional[str], arg: Optional[str],
 
 def handle_local_command(subcommand: Optional[str], arg: Optional[str],
                          extra: Optional[str], weights: Optional[str],
-                         gpu: Optional[int]) -> int:
+                         gpu: Optional[int],
+                         dtype: Optional[str] = None) -> int:
     """`debate.py local {status,add-model,remove-model,alias,list-models}` —
     manages the MI355X model registry section of the global config."""
     config = load_global_config()
@@ -563,6 +564,8 @@ def handle_local_command(subcommand: Optional[str], arg: Optional[str],
             spec["weights"] = weights
         if gpu is not None:
             spec["gpu"] = gpu
+        if dtype:
+            spec["dtype"] = dtype
         local.setdefault("custom_aliases", {})[arg] = spec
         save_global_config(config)
         print(f"Local alias added: {arg} -> {json.dumps(spec)}")
