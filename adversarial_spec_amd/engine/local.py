@@ -376,64 +376,6 @@ class LocalEngine:
                         return True
             return False
 
-        if graph is not None and max_new - i > CHECK:
-            # PIPELINED replay: keep one chunk of replays queued AHEAD of
-            # the chunk being scanned, so the engine stream never idles on
-            # the host's stop-scan. Chunk k's tok_hist slice is copied
-            # D2H on a side stream gated by an event recorded after chunk
-            # k's replays — the host wait releases when chunk k completes,
-            # NOT when the already-queued chunk k+1 does.
-            copy_stream = torch.cuda.Stream(device=dev)
-            tok_host = torch.empty_like(tok_hist, device="cpu",
-                                        pin_memory=True)
-
-            def launch_chunk(start: int) -> tuple[int, torch.cuda.Event]:
-                n = min(CHECK, max_new - start)
-                reps = (n + _SPG - 1) // _SPG
-                for _ in range(reps):
-                    graph.replay()
-                ev = torch.cuda.Event()
-                ev.record(torch.cuda.current_stream(dev))
-                return start + reps * _SPG, ev
-
-            def scan_chunk(lo: int, hi: int) -> bool:
-                nonlocal tail
-                for tok in tok_host[lo:hi].tolist():
-                    if tok in stop_ids or tok < 0:
-                        return True
-                    out_ids.append(tok)
-                    if 0 <= tok < 256:
-                        tail = (tail + chr(tok))[-16:]
-                        if tail.endswith(_STOP_SUBSTR):
-                            return True
-                return False
-
-            nxt, ev = launch_chunk(i)
-            pend = (i, min(nxt, max_new), ev)
-            i = nxt
-            while not done:
-                if i < max_new:  # queue the NEXT chunk before scanning
-                    nxt, ev2 = launch_chunk(i)
-                    queued = (i, min(nxt, max_new), ev2)
-                    i = nxt
-                else:
-                    queued = None
-                lo, hi, ev = pend
-                with torch.cuda.stream(copy_stream):
-                    copy_stream.wait_event(ev)
-                    tok_host[lo:hi].copy_(tok_hist[lo:hi], non_blocking=True)
-                copy_stream.synchronize()
-                done = scan_chunk(lo, hi)
-                if time.monotonic() > deadline:
-                    done = True
-                if queued is None:
-                    break  # pend was the last chunk and is scanned
-                pend = queued
-            # on early stop the already-queued chunk's tokens are simply
-            # discarded (same semantics as the _SPG overshoot)
-            cache.seq_len = prompt_len + i
-            return out_ids
-
         while i < max_new and not done:
             n = min(CHECK, max_new - i)
             if graph is not None:
