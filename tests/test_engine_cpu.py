@@ -276,3 +276,27 @@ class TestSafetensorsLoader:
         m = LlamaModel(get_config("tiny"), device="cpu")
         with _pytest.raises(FileNotFoundError):
             m.load_safetensors("/nonexistent/dir")
+
+
+class TestFitPrompt:
+    def test_clamps_over_budget_keeping_head_and_tail(self):
+        from adversarial_spec_amd.engine.local import LocalEngine
+
+        eng = LocalEngine({"name": "fp", "arch": "tiny"}, device="cpu")
+        max_seq = eng.config.max_seq_len  # 2048
+        ids = list(range(1, 3000 + 1))
+        reserve = 128
+        out = eng._fit_prompt(ids, reserve)
+        budget = max_seq - reserve - 8
+        assert len(out) == budget
+        head = budget * 2 // 3
+        # head preserved verbatim, tail preserved verbatim (middle dropped)
+        assert out[:head] == ids[:head]
+        assert out[head:] == ids[-(budget - head):]
+
+    def test_under_budget_untouched(self):
+        from adversarial_spec_amd.engine.local import LocalEngine
+
+        eng = LocalEngine({"name": "fp2", "arch": "tiny"}, device="cpu")
+        ids = list(range(100))
+        assert eng._fit_prompt(ids, 256) == ids
