@@ -324,11 +324,10 @@ torch::Tensor attn_decode_paged(torch::Tensor q, torch::Tensor kc,
   // Size splits for full-chip occupancy: the split kernel runs one
   // 256-thread block per (kv_head, split); target ~1024 blocks so every CU
   // carries ~4 blocks (16 waves) to hide the KV-read latency.
-  // ~1024 blocks total (4 waves/SIMD): the split kernel's softmax chain
-  // exposes HBM latency beyond what the 1-deep prefetch hides, so wave
-  // OVERSUBSCRIPTION is the latency hiding; the 8-way-unrolled combine
-  // keeps the longer split walk cheap.
-  int n_splits = std::max(1, std::min((int)((seq_len + 63) / 64), 1024 / kh));
+  // ~512 blocks total: bigger splits keep the combine walk short (the
+  // combine is latency-bound at 1 thread per (head, dim) element) while
+  // the split kernel hides KV latency with in-loop prefetch.
+  int n_splits = std::max(1, std::min((int)((seq_len + 255) / 256), 512 / kh));
   int split_len = (int)((seq_len + n_splits - 1) / n_splits + 63) / 64 * 64;
   n_splits = (int)((seq_len + split_len - 1) / split_len);
   auto out = torch::empty({hq, hd}, qc.options());
@@ -471,7 +470,7 @@ torch::Tensor attn_decode_paged_ds(torch::Tensor q, torch::Tensor kc,
   const int hq = qc.size(0), hd = qc.size(1);
   const int page = kc.size(1), kh = kc.size(2);
   const int group = hq / kh;
-  int n_splits = std::max(1, std::min((int)((max_seq + 63) / 64), 1024 / kh));
+  int n_splits = std::max(1, std::min((int)((max_seq + 255) / 256), 512 / kh));
   int split_len = (int)((max_seq + n_splits - 1) / n_splits + 63) / 64 * 64;
   n_splits = (int)((max_seq + split_len - 1) / split_len);
   auto out = out_opt.has_value() ? *out_opt : torch::empty({hq, hd}, qc.options());
