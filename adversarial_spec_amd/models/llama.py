@@ -289,12 +289,12 @@ class LlamaModel:
             )
         return self
 
-    # fp8 default = MIXED precision: quantize the bandwidth-dominant
-    # projections (gate_up is ~54% of layer weight bytes; lm_head is 1 GB)
-    # and keep the small latency-bound ones bf16 — the fp8 decode GEMV is
-    # cvt-ALU-bound at small N and measured SLOWER than bf16 there
-    # (qkv 10.7 vs 7.7 us, o 9.9 vs 4.3), while gate_up/lm_head win.
-    FP8_PROJECTIONS = ("w_gate_up",)
+    # fp8 default = UNIFORM: under multi-opponent concurrency decode is
+    # HBM-bound, so total BYTES win — uniform fp8 (8 GB/8B model) measured
+    # 0.714 critiques/s vs 0.670 for a mixed scheme that kept the small
+    # latency-bound projections bf16 (solo-kernel ALU numbers do not
+    # transfer to the bandwidth-shared regime).
+    FP8_PROJECTIONS = ("wqkv", "wo", "w_gate_up", "w_down")
 
     def quantize_fp8(self, projections=None, lm_head: bool = True) -> "LlamaModel":
         """Switch projection weights to rowwise OCP e4m3 (BASELINE config
