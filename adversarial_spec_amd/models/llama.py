@@ -146,8 +146,12 @@ class LlamaModel:
         self.seed = seed
         c = config
         self.scale = 1.0 / math.sqrt(c.head_dim)
+        # +8 slack: the multi-step decode graph may overshoot max_new by
+        # up to _SPG-1 harmless steps whose RoPE reads land past the
+        # nominal window (tokens discarded, but the table read must stay
+        # in bounds)
         self.cos, self.sin = torch_ref.rope_tables(
-            c.head_dim, c.max_seq_len, c.rope_theta, self.device
+            c.head_dim, c.max_seq_len + 8, c.rope_theta, self.device
         )
         self.embed: Optional[torch.Tensor] = None  # [vocab, d]
         self.final_norm: Optional[torch.Tensor] = None  # [d]
