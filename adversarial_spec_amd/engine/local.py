@@ -404,7 +404,8 @@ _ENGINES_LOCK = threading.Lock()
 
 def get_engine(spec: dict[str, Any], device: Optional[str] = None) -> LocalEngine:
     """Process-wide engine cache: weights stay HBM-resident across rounds."""
-    key = (spec.get("name"), spec.get("arch"), spec.get("weights"), device or spec.get("gpu"))
+    key = (spec.get("name"), spec.get("arch"), spec.get("weights"),
+           spec.get("dtype"), device or spec.get("gpu"))
     with _ENGINES_LOCK:
         eng = _ENGINES.get(key)
         if eng is None:
