@@ -36,6 +36,30 @@ _CAPTURE_LOCK = threading.Lock()
 _SPG = 4
 
 
+class _StopScan:
+    """Sliding-window [/SPEC] detector over DECODED text.
+
+    Decodes the last few token ids with the engine's own tokenizer each
+    push, so the close tag is detected for byte-level AND BPE tokenizers
+    alike. (Round 1 built the tail only from ids < 256 — with an
+    HFTokenizer-backed checkpoint every BPE piece id is >= 256 and the tag
+    was never seen, burning up to 8000 wasted tokens per critique.)
+    """
+
+    __slots__ = ("_tok", "_ids")
+    WINDOW = 16  # [/SPEC] is 7 chars; 16 tokens always covers it
+
+    def __init__(self, tokenizer) -> None:
+        self._tok = tokenizer
+        self._ids: list[int] = []
+
+    def push(self, tok: int) -> bool:
+        self._ids.append(tok)
+        if len(self._ids) > self.WINDOW:
+            del self._ids[0]
+        return _STOP_SUBSTR in self._tok.decode(list(self._ids))
+
+
 def _seed_from_name(name: str) -> int:
     h = 2166136261
     for ch in name:
@@ -152,10 +176,21 @@ class LocalEngine:
         graph."""
         cache = self._cache
         if cache is None or cache.max_seq < need:
+            old = cache
             bucket = (need + 2047) // 2048 * 2048
             cache = self.model.new_cache(bucket)
             self._cache = cache
+            # Drop the captured graph BEFORE releasing the old cache's
+            # decode-attention scratch: once _graph_state is gone no replay
+            # can reference the retired scratch tensors, so the extension
+            # may erase the per-cache workspace map entries (otherwise a
+            # long-lived process leaks scratch per growth and can inherit a
+            # stale entry when the allocator reuses the freed address).
             self._graph_state = None
+            if old is not None and self.device.type == "cuda" and ops.hip_available():
+                from ..ops import _load_hip
+
+                _load_hip().ws_release(old.k)
         cache.seq_len = 0
         return cache
 
@@ -211,7 +246,7 @@ class LocalEngine:
                      stop_ids, deadline) -> list[int]:
         """Host-stepped decode (CPU path and nucleus sampling)."""
         out_ids: list[int] = []
-        tail = ""
+        scan = _StopScan(self.tokenizer)
         for _ in range(max_new):
             tok = ops.sample(
                 logits, temperature=temperature, top_p=top_p,
@@ -220,10 +255,8 @@ class LocalEngine:
             if tok in stop_ids:
                 break
             out_ids.append(tok)
-            if 0 <= tok < 256:
-                tail = (tail + chr(tok))[-16:]
-                if tail.endswith(_STOP_SUBSTR):
-                    break
+            if scan.push(tok):
+                break
             if time.monotonic() > deadline:
                 # deadline: return the partial critique rather than erroring
                 # the opponent (a partial decode is still a usable critique).
@@ -257,13 +290,20 @@ class LocalEngine:
         max_total = cache.max_seq
 
         gs = self._graph_state if use_graph else None
-        key = (id(cache), cache.max_seq, max_new, float(temperature))
+        # The key is the CACHE identity only: max_new is covered by sizing
+        # tok_hist to the cache's full capacity, and temperature lives in a
+        # device-side f32 word read in-kernel — so one captured graph serves
+        # every request shape on a warm cache (rounds that alternate
+        # max_new/temperature no longer recapture, and capture's device-wide
+        # synchronize no longer stalls co-resident opponents per round).
+        key = (id(cache), cache.max_seq)
         if gs is not None and gs["key"] == key:
             # Round k+1 on a warm engine: reset the device-side state words
             # and replay the graph captured in round 1 — no re-capture.
             pos_state = gs["pos_state"]
             step_state = gs["step_state"]
             rng_state = gs["rng_state"]
+            temp_state = gs["temp_state"]
             tok_hist = gs["tok_hist"]
             tok_slot = gs["tok_slot"]
             W = gs["ws"]
@@ -271,6 +311,7 @@ class LocalEngine:
             pos_state.fill_(prompt_len)
             step_state.zero_()
             rng_state.fill_(self._next_seed() | 1)
+            temp_state.fill_(float(temperature))
             tok_hist.fill_(-1)
             tok_slot.zero_()
             W.logits.copy_(logits.reshape(1, -1))
@@ -280,10 +321,13 @@ class LocalEngine:
             step_state = torch.zeros(1, dtype=torch.int32, device=dev)
             rng_state = torch.tensor([self._next_seed() | 1], dtype=torch.int32,
                                      device=dev)
-            # + _SPG slack: multi-step replays may overshoot max_new by up
-            # to _SPG-1 harmless steps
-            tok_hist = torch.full((max_new + _SPG + 2,), -1, dtype=torch.int32,
-                                  device=dev)
+            temp_state = torch.tensor([float(temperature)], dtype=torch.float32,
+                                      device=dev)
+            # sized to the cache capacity (not max_new) so the same graph
+            # state serves any decode length that fits this cache; + _SPG
+            # slack: multi-step replays may overshoot by _SPG-1 steps
+            tok_hist = torch.full((cache.max_seq + _SPG + 2,), -1,
+                                  dtype=torch.int32, device=dev)
             tok_slot = torch.zeros(1, dtype=torch.int32, device=dev)
             W = self.model.new_decode_ws()
             W.logits.copy_(logits.reshape(1, -1))
@@ -292,7 +336,7 @@ class LocalEngine:
             def step():
                 # ZERO allocations (capture-safe): sample from W.logits,
                 # feed the id through W.tok_long, forward writes W.logits.
-                hip.sample_state(logits_buf, temperature, rng_state, tok_hist,
+                hip.sample_state(logits_buf, temp_state, rng_state, tok_hist,
                                  step_state, tok_slot)
                 W.tok_long.copy_(tok_slot.view(1))
                 self.model.decode_step_ws(cache, pos_state, max_total, W)
@@ -349,31 +393,29 @@ class LocalEngine:
                     self._graph_state = {
                         "key": key, "graph": graph, "pos_state": pos_state,
                         "step_state": step_state, "rng_state": rng_state,
-                        "tok_hist": tok_hist, "tok_slot": tok_slot,
-                        "ws": W,
+                        "temp_state": temp_state, "tok_hist": tok_hist,
+                        "tok_slot": tok_slot, "ws": W,
                     }
 
         CHECK = 32
         done = False
         i = warm
         out_ids: list[int] = []
-        tail = ""
+        scan = _StopScan(self.tokenizer)
         scanned = 0
 
         def scan_until(upto: int) -> bool:
             """Pull tokens [scanned, upto) to host, extend out_ids; True if a
             stop condition fired."""
-            nonlocal scanned, tail
+            nonlocal scanned
             chunk = tok_hist[scanned:upto].cpu().tolist()
             scanned = upto
             for tok in chunk:
                 if tok in stop_ids or tok < 0:
                     return True
                 out_ids.append(tok)
-                if 0 <= tok < 256:
-                    tail = (tail + chr(tok))[-16:]
-                    if tail.endswith(_STOP_SUBSTR):
-                        return True
+                if scan.push(tok):
+                    return True
             return False
 
         while i < max_new and not done:
@@ -416,4 +458,14 @@ def get_engine(spec: dict[str, Any], device: Optional[str] = None) -> LocalEngin
 
 def clear_engines() -> None:
     with _ENGINES_LOCK:
+        for eng in _ENGINES.values():
+            # drop captured graphs first, then the per-cache attention
+            # scratch the extension holds for this engine's KV cache
+            eng._graph_state = None
+            cache = eng._cache
+            if (cache is not None and eng.device.type == "cuda"
+                    and ops.hip_available()):
+                from ..ops import _load_hip
+
+                _load_hip().ws_release(cache.k)
         _ENGINES.clear()

@@ -5,16 +5,17 @@
 // Structure (cdna_hip_programming.md §5/§B):
 //   grid  = (hq, ceil(tq/128));  block = 512 threads (8 waves)
 //   each wave owns a 16-row Q tile; the block shares LDS-staged K/V tiles
-//   of 32 keys.
+//   of KVBLK=64 keys (64 halves the softmax passes + syncs vs 32).
 //
 // LDS images (all carved from ONE array — compiler trap §5 4a):
-//   K  [32][128] bf16, XOR-swizzled byte^=((key&7)<<4): a 256-B row puts a
+//   K  [64][128] bf16, XOR-swizzled byte^=((key&7)<<4): a 256-B row puts a
 //      column read's whole 16-lane group on one bank without it
 //      (Guideline 4's D=128 hazard — 16-way, measured 5.1 ms/layer).
-//   Vt [128][40] bf16: V TRANSPOSED with rows padded 32->40 elements so the
-//      PV B-fragment is one ds_read_b128 per lane (row stride 80 B makes
-//      the 16 consecutive-row banks distinct) instead of 8 scalar reads.
-//   P  [16][40] bf16 per wave (C->A layout bounce), same 80-B row pitch.
+//   Vt [128][VPITCH=72] bf16: V TRANSPOSED with 64-key rows padded to 72
+//      elements (row stride 144 B) plus a per-row rotation of the key index
+//      (krow+rot)&63, so the PV B-fragment is one ds_read_b128 per lane
+//      with the 16 consecutive-row banks distinct, not 8 scalar reads.
+//   P  [16][72] bf16 per wave (C->A layout bounce), same 144-B row pitch.
 //
 // MFMA fragment maps (hardware-verified by tests/test_ops_gpu.py mfma probe):
 //   mfma_f32_16x16x32_bf16:

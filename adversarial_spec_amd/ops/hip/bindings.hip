@@ -39,8 +39,8 @@ __global__ void rope_kv_kernel(ushort_t*, ushort_t*, const ushort_t*,
                                const int*, int, int, int, int, int, long, long,
                                long, int, const int*);
 __global__ void sample_kernel(const ushort_t*, int, float, uint32_t, int*);
-__global__ void sample_state_kernel(const ushort_t*, int, float, uint32_t*, int*,
-                                    const int*, int*);
+__global__ void sample_state_kernel(const ushort_t*, int, const float*,
+                                    uint32_t*, int*, const int*, int*);
 __global__ void bump_kernel(int*, int*);
 void launch_gemv(const ushort_t*, const ushort_t*, ushort_t*, int, int,
                  hipStream_t);
@@ -491,16 +491,39 @@ torch::Tensor attn_decode_paged_ds(torch::Tensor q, torch::Tensor kc,
   return out;
 }
 
-void sample_state(torch::Tensor logits, double temp, torch::Tensor rng_state,
-                  torch::Tensor tok_hist, torch::Tensor step_state,
-                  torch::Tensor tok_slot) {
+void sample_state(torch::Tensor logits, torch::Tensor temp_state,
+                  torch::Tensor rng_state, torch::Tensor tok_hist,
+                  torch::Tensor step_state, torch::Tensor tok_slot) {
   CHECK_BF16_CUDA(logits);
+  TORCH_CHECK(temp_state.scalar_type() == at::kFloat && temp_state.is_cuda(),
+              "sample_state: temp_state must be a device f32 word");
   auto lc = logits.contiguous();
   sample_state_kernel<<<1, 1024, 0, cur_stream()>>>(
-      uptr(lc), (int)lc.numel(), (float)temp,
+      uptr(lc), (int)lc.numel(), temp_state.data_ptr<float>(),
       reinterpret_cast<uint32_t*>(rng_state.data_ptr<int>()),
       tok_hist.data_ptr<int>(), step_state.data_ptr<int>(),
       tok_slot.data_ptr<int>());
+}
+
+// Release the decode-attention scratch entries whose keys (per-layer KV
+// cache base pointers) lie inside a dropped cache allocation, and free the
+// retired grown tensors. Call ONLY after every captured graph referencing
+// the cache has been destroyed (engine/local.py drops _graph_state first);
+// without this, long-lived processes leak scratch per cache growth and can
+// inherit a stale entry when the allocator reuses a freed cache address
+// (ADVICE round 1).
+void ws_release(torch::Tensor kc_all) {
+  char* base = reinterpret_cast<char*>(kc_all.data_ptr());
+  const size_t bytes = (size_t)kc_all.numel() * kc_all.element_size();
+  std::lock_guard<std::mutex> lk(g_ws_mu);
+  for (auto it = g_ws.begin(); it != g_ws.end();) {
+    char* p = reinterpret_cast<char*>(it->first);
+    if (p >= base && p < base + bytes) {
+      it = g_ws.erase(it);
+    } else {
+      ++it;
+    }
+  }
 }
 
 void bump(torch::Tensor pos_state, torch::Tensor step_state) {
@@ -544,4 +567,6 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         py::arg("out") = py::none());
   m.def("sample_state", &sample_state, "graph-mode on-device sampling");
   m.def("bump", &bump, "graph-mode pos/step bump");
+  m.def("ws_release", &ws_release,
+        "drop decode-attention scratch for a dead KV cache");
 }

@@ -193,13 +193,18 @@ gemm_fp8_kernel(const uint8_t *__restrict__ a, const float *__restrict__ asc,
 // ---------------------------------------------------------------------------
 
 DEVINL float fp8_to_f32(uint8_t v) {
-  // OCP e4m3 (1s 4e 3m, bias 7) -> f32 by bit manipulation: for normals
+  // OCP e4m3fn (1s 4e 3m, bias 7) -> f32 by bit manipulation: for normals
   // the f32 exponent is e+(127-7) and the mantissa top 3 bits are m, so
   // bits = sign | (em + (120<<3)) << 20. Subnormals: m * 2^-9.
+  // em == 0x7f is NaN in e4m3fn (no infinities, +-NaN only): map it to a
+  // real f32 NaN so NaN-poisoned quantized weights surface on GPU exactly
+  // like the CPU float8_e4m3fn reference instead of decoding to ~480.
   const uint32_t s = (uint32_t)(v & 0x80u) << 24;
   const uint32_t em = v & 0x7fu;
   uint32_t bits;
-  if (em >= 8u) {
+  if (em == 0x7fu) {
+    bits = s | 0x7fc00000u;
+  } else if (em >= 8u) {
     bits = s | ((em + 960u) << 20);
   } else {
     bits = s | __builtin_bit_cast(uint32_t, (float)em * 0.001953125f);

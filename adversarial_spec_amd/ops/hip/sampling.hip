@@ -78,21 +78,25 @@ sample_kernel(const ushort_t *__restrict__ logits, int vocab, float temp,
   sample_body(logits, vocab, temp, seed, out);
 }
 
-// HIP-graph decode variant: the RNG seed, step index and token outputs all
-// live in device words so one captured graph replays per token.
+// HIP-graph decode variant: the RNG seed, step index, TEMPERATURE and token
+// outputs all live in device words so one captured graph replays per token
+// and serves every request temperature (the graph cache key must not
+// include temperature — ADVICE round 1).
+//   temp_state: f32[1], sampling temperature (<= 0 -> greedy)
 //   rng_state:  uint32[1], advanced each call
 //   step_state: int32[1], the decode step index (bumped by bump_kernel)
 //   tok_hist:   int32[max_new] history the host polls every N tokens
 //   tok_slot:   int32[1] fixed slot feeding the next embedding lookup
 extern "C" __global__ void __launch_bounds__(1024)
 sample_state_kernel(const ushort_t *__restrict__ logits, int vocab,
-                    float temp, uint32_t *__restrict__ rng_state,
+                    const float *__restrict__ temp_state,
+                    uint32_t *__restrict__ rng_state,
                     int *__restrict__ tok_hist,
                     const int *__restrict__ step_state,
                     int *__restrict__ tok_slot) {
   __shared__ int picked[1];
   const uint32_t seed = *rng_state;
-  sample_body(logits, vocab, temp, seed, picked);
+  sample_body(logits, vocab, *temp_state, seed, picked);
   __syncthreads();
   if (threadIdx.x == 0) {
     const int tok = picked[0];
