@@ -46,6 +46,8 @@ void launch_gemv(const ushort_t*, const ushort_t*, ushort_t*, int, int,
                  hipStream_t);
 void launch_gemm(const ushort_t*, const ushort_t*, ushort_t*, int, int, int,
                  hipStream_t);
+void launch_gemm256(const ushort_t*, const ushort_t*, ushort_t*, int, int, int,
+                    hipStream_t);
 void launch_gemm_fp8(const uint8_t*, const float*, const uint8_t*, const float*,
                      ushort_t*, int, int, int, hipStream_t);
 void launch_gemv_fp8(const uint8_t*, const float*, const uint8_t*, const float*,
@@ -393,7 +395,33 @@ torch::Tensor gemm(torch::Tensor a, torch::Tensor b) {
   auto bc = b.contiguous();
   const int M = ac.size(0), K = ac.size(1), N = bc.size(0);
   auto c = torch::empty({M, N}, ac.options());
-  launch_gemm(uptr(ac), uptr(bc), uptr_mut(c), M, N, K, cur_stream());
+  // Deep-pipelined 256^2 8-phase kernel for prefill-sized shapes (its
+  // counted-vmcnt schedule needs K % 128 == 0 and pays off when the M/N
+  // tiles fill); the 128^2 kernel covers everything else.
+  if (M > 128 && N >= 256 && K >= 512 && (K % 128) == 0) {
+    launch_gemm256(uptr(ac), uptr(bc), uptr_mut(c), M, N, K, cur_stream());
+  } else {
+    launch_gemm(uptr(ac), uptr(bc), uptr_mut(c), M, N, K, cur_stream());
+  }
+  return c;
+}
+
+// Force a specific GEMM kernel (128 = two-barrier tile, 256 = 8-phase
+// deep-pipelined) — A/B microbenches and the race screen; production code
+// uses gemm()'s dispatcher.
+torch::Tensor gemm_variant(torch::Tensor a, torch::Tensor b, int64_t which) {
+  CHECK_BF16_CUDA(a);
+  CHECK_BF16_CUDA(b);
+  auto ac = a.contiguous();
+  auto bc = b.contiguous();
+  const int M = ac.size(0), K = ac.size(1), N = bc.size(0);
+  auto c = torch::empty({M, N}, ac.options());
+  if (which == 256) {
+    TORCH_CHECK(K >= 512 && K % 128 == 0, "gemm256 needs K%128==0, K>=512");
+    launch_gemm256(uptr(ac), uptr(bc), uptr_mut(c), M, N, K, cur_stream());
+  } else {
+    launch_gemm(uptr(ac), uptr(bc), uptr_mut(c), M, N, K, cur_stream());
+  }
   return c;
 }
 
@@ -551,6 +579,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("gemv", &gemv, "batch-1 decode GEMV (weight streaming)",
         py::arg("x"), py::arg("w"), py::arg("out") = py::none());
   m.def("gemm", &gemm, "tiled MFMA GEMM (bf16, fp32 accum)");
+  m.def("gemm_variant", &gemm_variant, "force GEMM kernel 128/256 (A/B)");
   m.def("gemm_fp8", &gemm_fp8, "fp8 e4m3 MFMA GEMM (rowwise scales)");
   m.def("gemv_fp8", &gemv_fp8, "fp8 decode GEMV (rowwise scales)");
   m.def("quant_fp8", &quant_fp8, "rowwise bf16 -> e4m3 quantizer");

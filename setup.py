@@ -26,6 +26,7 @@ ext = CUDAExtension(
         os.path.join(HIP_DIR, "attn_prefill_mfma.hip"),
         os.path.join(HIP_DIR, "gemv.hip"),
         os.path.join(HIP_DIR, "gemm.hip"),
+        os.path.join(HIP_DIR, "gemm256.hip"),
         os.path.join(HIP_DIR, "gemm_fp8.hip"),
         os.path.join(HIP_DIR, "sampling.hip"),
     ],

@@ -1,0 +1,115 @@
+"""torchrun worker for the RCCL-on-hardware tests (tests/test_rccl_gpu.py).
+
+Launched as:
+    python -m torch.distributed.run --nnodes=1 --nproc-per-node 2 \
+        --master-addr 127.0.0.1 --master-port <p> tests/rccl_worker.py <mode>
+
+Two ranks SHARE one MI355X (device = rank % device_count): RCCL supports
+several ranks per device on ROCm, which lets a 1-GPU lease execute the real
+collectives the 8-GPU debate round uses (round-1 verdict: no RCCL
+collective had ever run on hardware).
+"""
+
+from __future__ import annotations
+
+import os
+import sys
+
+import torch
+import torch.distributed as dist
+
+
+def run_consensus(rank: int, world: int) -> None:
+    """Fused consensus all-gather over RCCL."""
+    from adversarial_spec_amd.parallel.consensus import (
+        AsyncRoundGather,
+        gather_round,
+    )
+
+    ids = [100 + rank, 200 + rank, 300]
+    results, all_agreed = gather_round(
+        ids, agreed=(rank % 2 == 0), error=False, max_tokens=64
+    )
+    assert len(results) == world
+    for r in range(world):
+        assert results[r].token_ids == [100 + r, 200 + r, 300], results[r]
+        assert results[r].agreed == (r % 2 == 0)
+    assert all_agreed == (world == 1)
+
+    # async variant on its own comm stream (the overlap path bench.py uses)
+    g = AsyncRoundGather(max_tokens=64)
+    g.launch([1, 2, rank], True, False)
+    res, ok = g.wait()
+    assert ok and len(res) == world
+    assert [r.token_ids for r in res] == [[1, 2, rr] for rr in range(world)]
+
+    # errored-rank exclusion (reference debate.py:845-853 semantics)
+    results, all_agreed = gather_round(
+        [7], agreed=False, error=(rank == world - 1), max_tokens=8
+    )
+    ok_ranks = [r for r in results if not r.error]
+    assert len(ok_ranks) == world - 1 if world > 1 else True
+
+    if rank == 0:
+        print(f"RCCL_CONSENSUS_OK backend={dist.get_backend()} world={world}")
+
+
+def run_tp(rank: int, world: int) -> None:
+    """TP=2 sharded forward over RCCL equals the unsharded model."""
+    from adversarial_spec_amd.models import LlamaModel
+    from adversarial_spec_amd.models.config import LlamaConfig
+    from adversarial_spec_amd.parallel.tp import TPContext
+
+    cfg = LlamaConfig(
+        name="tp-gpu-test", dim=256, n_layers=2, n_heads=8, n_kv_heads=2,
+        ffn_dim=512, vocab_size=512, max_seq_len=256, rope_theta=10000.0,
+    )  # head_dim = 256/8 = 32 (in the decode kernel's {32,64,128} set)
+    dev = torch.device("cuda", torch.cuda.current_device())
+    tp = TPContext(world, rank, None)
+    m_tp = LlamaModel(cfg, device=dev, seed=11, tp=tp).init_random()
+    m_full = LlamaModel(cfg, device=dev, seed=11).init_random()
+
+    toks = torch.arange(3, 40, device=dev)
+    c_tp = m_tp.new_cache(128)
+    c_full = m_full.new_cache(128)
+    l_tp = m_tp.prefill(toks, c_tp).float()
+    l_full = m_full.prefill(toks, c_full).float()
+    err = (l_tp - l_full).abs().max().item()
+    scale = l_full.abs().max().item() + 1e-6
+    assert err / scale < 5e-2, f"TP parity: max err {err} vs scale {scale}"
+
+    # decode parity for a few tokens (host-stepped TP decode loop)
+    tok_tp = tok_full = int(l_full.argmax().item())
+    for _ in range(4):
+        lt = m_tp.decode_one(tok_tp, c_tp).float()
+        lf = m_full.decode_one(tok_full, c_full).float()
+        e = (lt - lf).abs().max().item() / (lf.abs().max().item() + 1e-6)
+        assert e < 5e-2, f"TP decode parity: rel err {e}"
+        tok_tp = int(lt.argmax().item())
+        tok_full = int(lf.argmax().item())
+        assert tok_tp == tok_full
+
+    if rank == 0:
+        print(f"RCCL_TP_OK backend={dist.get_backend()} world={world}")
+
+
+def main() -> int:
+    mode = sys.argv[1] if len(sys.argv) > 1 else "consensus"
+    rank = int(os.environ["RANK"])
+    world = int(os.environ["WORLD_SIZE"])
+    torch.cuda.set_device(rank % torch.cuda.device_count())
+    dist.init_process_group("nccl")
+    try:
+        if mode == "consensus":
+            run_consensus(rank, world)
+        elif mode == "tp":
+            run_tp(rank, world)
+        else:
+            raise SystemExit(f"unknown mode {mode}")
+    finally:
+        dist.destroy_process_group()
+    return 0
+
+
+if __name__ == "__main__":
+    sys.exit(main())

@@ -319,6 +319,57 @@ class TestGemm:
             assert torch.equal(ops.gemm(a, b), x)
 
 
+class TestGemm256:
+    """8-phase 256^2 deep-pipelined kernel (gemm256.hip).
+
+    A NEW sync structure (counted vmcnt spanning raw barriers) gets the
+    guide's two-lane discipline: refcheck at several shapes + a multi-run
+    race screen, and A/B equivalence against the two-barrier 128^2 kernel.
+    """
+
+    def _hip(self):
+        from adversarial_spec_amd.ops import _load_hip
+
+        return _load_hip()
+
+    @pytest.mark.parametrize("m,n,k", [
+        (256, 256, 512),       # single tile, min K
+        (512, 512, 1024),      # multi-tile
+        (300, 770, 640),       # M and N edges (clamped rows/cols)
+        (8192 // 16, 6144 // 4, 4096 // 4),  # model-shaped, scaled
+    ])
+    def test_vs_torch_fp32(self, m, n, k):
+        hip = self._hip()
+        a = _bf(torch.randn(m, k)).to(DEV)
+        b = _bf(torch.randn(n, k) * 0.05).to(DEV)
+        got = hip.gemm_variant(a, b, 256)
+        want = a.float().cpu() @ b.float().cpu().t()
+        _assert_close(got, want, atol=5e-2, name=f"gemm256 {m}x{n}x{k}")
+
+    def test_matches_128_variant(self):
+        hip = self._hip()
+        a = _bf(torch.randn(512, 1024)).to(DEV)
+        b = _bf(torch.randn(768, 1024) * 0.02).to(DEV)
+        g256 = hip.gemm_variant(a, b, 256)
+        g128 = hip.gemm_variant(a, b, 128)
+        # identical K-order fp32 accumulation -> bitwise equal bf16 outputs
+        assert torch.equal(g256, g128)
+
+    def test_race_screen(self):
+        """Counted-vmcnt schedules race, not drift: rerun many times at two
+        shapes and require bitwise-stable output matching the reference."""
+        hip = self._hip()
+        for m, n, k in [(256, 512, 512), (512, 256, 1536)]:
+            a = _bf(torch.randn(m, k)).to(DEV)
+            b = _bf(torch.randn(n, k) * 0.05).to(DEV)
+            first = hip.gemm_variant(a, b, 256)
+            want = a.float().cpu() @ b.float().cpu().t()
+            _assert_close(first, want, atol=5e-2, name=f"race {m}x{n}x{k}")
+            for _ in range(10):
+                again = hip.gemm_variant(a, b, 256)
+                assert torch.equal(again, first), "gemm256 nondeterminism"
+
+
 class TestGemv:
     @pytest.mark.parametrize("n,k", [(1024, 512), (6144, 4096), (1000, 264)])
     def test_vs_torch_fp32(self, n, k):
