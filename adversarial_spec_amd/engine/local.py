@@ -305,7 +305,6 @@ class LocalEngine:
             rng_state = gs["rng_state"]
             temp_state = gs["temp_state"]
             tok_hist = gs["tok_hist"]
-            tok_slot = gs["tok_slot"]
             W = gs["ws"]
             graph = gs["graph"]
             pos_state.fill_(prompt_len)
@@ -313,7 +312,7 @@ class LocalEngine:
             rng_state.fill_(self._next_seed() | 1)
             temp_state.fill_(float(temperature))
             tok_hist.fill_(-1)
-            tok_slot.zero_()
+            W.tok_long.zero_()
             W.logits.copy_(logits.reshape(1, -1))
             warm = 0
         else:
@@ -328,17 +327,16 @@ class LocalEngine:
             # slack: multi-step replays may overshoot by _SPG-1 steps
             tok_hist = torch.full((cache.max_seq + _SPG + 2,), -1,
                                   dtype=torch.int32, device=dev)
-            tok_slot = torch.zeros(1, dtype=torch.int32, device=dev)
             W = self.model.new_decode_ws()
             W.logits.copy_(logits.reshape(1, -1))
             logits_buf = W.logits.view(-1)
 
             def step():
-                # ZERO allocations (capture-safe): sample from W.logits,
-                # feed the id through W.tok_long, forward writes W.logits.
+                # ZERO allocations (capture-safe): sample from W.logits
+                # writes W.tok_long (the embedding index) in-kernel, the
+                # forward writes W.logits back.
                 hip.sample_state(logits_buf, temp_state, rng_state, tok_hist,
-                                 step_state, tok_slot)
-                W.tok_long.copy_(tok_slot.view(1))
+                                 step_state, W.tok_long)
                 self.model.decode_step_ws(cache, pos_state, max_total, W)
                 hip.bump(pos_state, step_state)
 
@@ -394,7 +392,7 @@ class LocalEngine:
                         "key": key, "graph": graph, "pos_state": pos_state,
                         "step_state": step_state, "rng_state": rng_state,
                         "temp_state": temp_state, "tok_hist": tok_hist,
-                        "tok_slot": tok_slot, "ws": W,
+                        "ws": W,
                     }
 
         CHECK = 32

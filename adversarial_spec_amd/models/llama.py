@@ -392,8 +392,14 @@ class LlamaModel:
             if self.tp is not None and self.tp.size > 1:
                 self.tp.all_reduce_(attn_out)  # row-parallel wo partial sums
             resid, normed = ops.add_rmsnorm(resid, attn_out, L.mlp_norm, c.norm_eps)
-            gu = proj(normed, L, Qd, "w_gate_up")
-            act = ops.swiglu(gu[:, : c.ffn_dim], gu[:, c.ffn_dim :])
+            if t == 1 and Qd is None:
+                # decode: fused gate_up GEMV + SwiGLU (one launch)
+                act = torch.empty(1, c.ffn_dim, device=normed.device,
+                                  dtype=normed.dtype)
+                ops.gemv_gateup(normed, L.w_gate_up, act)
+            else:
+                gu = proj(normed, L, Qd, "w_gate_up")
+                act = ops.swiglu(gu[:, : c.ffn_dim], gu[:, c.ffn_dim :])
             mlp_out = proj(act, L, Qd, "w_down")
             if self.tp is not None and self.tp.size > 1:
                 self.tp.all_reduce_(mlp_out)  # row-parallel down partial sums
@@ -489,8 +495,13 @@ class LlamaModel:
                 self.tp.all_reduce_(W.attn_out)
             ops.add_rmsnorm(W.resid, W.attn_out, L.mlp_norm, c.norm_eps,
                             out_resid=W.resid2, out_y=W.normed)
-            proj(W.normed, L, Qd, "w_gate_up", W.gu)
-            ops.swiglu(W.gu[:, : c.ffn_dim], W.gu[:, c.ffn_dim :], out=W.act)
+            if Qd is None:
+                # fused gate_up GEMV + SwiGLU: one launch, no gu round-trip
+                ops.gemv_gateup(W.normed, L.w_gate_up, W.act)
+            else:
+                proj(W.normed, L, Qd, "w_gate_up", W.gu)
+                ops.swiglu(W.gu[:, : c.ffn_dim], W.gu[:, c.ffn_dim :],
+                           out=W.act)
             proj(W.act, L, Qd, "w_down", W.mlp_out)
             if self.tp is not None and self.tp.size > 1:
                 self.tp.all_reduce_(W.mlp_out)

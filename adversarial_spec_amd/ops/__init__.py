@@ -189,6 +189,21 @@ def gemv(x: torch.Tensor, w: torch.Tensor,
     return x @ w.t()
 
 
+def gemv_gateup(x: torch.Tensor, w_gate_up: torch.Tensor,
+                out: torch.Tensor) -> torch.Tensor:
+    """Fused decode MLP front half: act = silu(x @ Wg^T) * (x @ Wu^T) with
+    w_gate_up = [gate rows | up rows] (the fused HF layout). One launch
+    replaces gate_up GEMV + swiglu on the kernel-count-bound decode path."""
+    if _on_gpu(x):
+        _require_hip().gemv_gateup(x, w_gate_up, out)
+        return out
+    f = w_gate_up.shape[0] // 2
+    gu = x @ w_gate_up.t()
+    y = torch_ref.swiglu(gu[..., :f], gu[..., f:])
+    out.copy_(y.reshape(out.shape))
+    return out
+
+
 def gemm(x: torch.Tensor, w: torch.Tensor) -> torch.Tensor:
     """Prefill matmul C = x @ w^T, weights ROW-MAJOR [out, in] (HF layout).
     On GPU the in-tree tiled MFMA kernel — library GEMMs (hipBLASLt/rocBLAS

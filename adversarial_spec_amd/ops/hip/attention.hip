@@ -67,16 +67,13 @@ __global__ void __launch_bounds__(256) attn_decode_split_kernel(
   const int limit = min(seq_len, min(start + pos_per_wave,
                                      (split + 1) * split_len));
 
-  // q fragments: [group][8] for this lane's dim slice
-  float qf[MG][8];
+  // q fragments: raw bf16 (the packed v_dot2 dot consumes bf16 directly —
+  // same fp32 products, no unpack pass)
+  bf16x8 qraw[MG];
 #pragma unroll
   for (int gi = 0; gi < MG; ++gi) {
     if (gi < group) {
-      const bf16x8 qv =
-          ((const bf16x8 *)(q + ((size_t)(g * group + gi)) * hd))[sl];
-      f32x8 qd = unpack8(qv);
-#pragma unroll
-      for (int j = 0; j < 8; ++j) qf[gi][j] = qd.v[j];
+      qraw[gi] = ((const bf16x8 *)(q + ((size_t)(g * group + gi)) * hd))[sl];
     }
   }
 
@@ -109,15 +106,12 @@ __global__ void __launch_bounds__(256) attn_decode_split_kernel(
       kn = ((const bf16x8 *)(kc + rown + (size_t)g * hd))[sl];
       vn = ((const bf16x8 *)(vc + rown + (size_t)g * hd))[sl];
     }
-    const f32x8 kd = unpack8(kraw);
     const f32x8 vd = unpack8(vraw);
 
 #pragma unroll
     for (int gi = 0; gi < MG; ++gi) {
       if (gi >= group) break;
-      float dot = 0.f;
-#pragma unroll
-      for (int j = 0; j < 8; ++j) dot += qf[gi][j] * kd.v[j];
+      const float dot = dot8_bf16(qraw[gi], kraw, 0.f);
       const float s = group_reduce_sum<LPP>(dot) * scale;
       const float m_new = fmaxf(m[gi], s);
       const float alpha = __expf(m[gi] - m_new);

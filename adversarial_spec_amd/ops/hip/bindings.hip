@@ -7,6 +7,7 @@
 #include <c10/hip/HIPStream.h>
 #include <hip/hip_runtime.h>
 
+#include <cstdlib>
 #include <mutex>
 #include <unordered_map>
 #include <vector>
@@ -40,10 +41,12 @@ __global__ void rope_kv_kernel(ushort_t*, ushort_t*, const ushort_t*,
                                long, int, const int*);
 __global__ void sample_kernel(const ushort_t*, int, float, uint32_t, int*);
 __global__ void sample_state_kernel(const ushort_t*, int, const float*,
-                                    uint32_t*, int*, const int*, int*);
+                                    uint32_t*, int*, const int*, long long*);
 __global__ void bump_kernel(int*, int*);
 void launch_gemv(const ushort_t*, const ushort_t*, ushort_t*, int, int,
                  hipStream_t);
+void launch_gemv_gateup(const ushort_t*, const ushort_t*, ushort_t*, int, int,
+                        hipStream_t);
 void launch_gemm(const ushort_t*, const ushort_t*, ushort_t*, int, int, int,
                  hipStream_t);
 void launch_gemm256(const ushort_t*, const ushort_t*, ushort_t*, int, int, int,
@@ -309,6 +312,23 @@ torch::Tensor mfma_probe16(torch::Tensor a, torch::Tensor b) {
   return d;
 }
 
+// Decode-attention split geometry. The split kernel's cost is dominated
+// by per-wave SERIAL iterations (shfl-reduce + online-softmax chains), not
+// KV bytes, so occupancy (total blocks) is the lever; ADVSPEC_SPLIT_BLOCKS
+// overrides the target block count for A/B tuning on hardware.
+static void split_geometry(long seq, int kh, int* n_splits, int* split_len) {
+  static const int cap_blocks = [] {
+    const char* e = getenv("ADVSPEC_SPLIT_BLOCKS");
+    const int v = e ? atoi(e) : 512;
+    return v > 0 ? v : 512;
+  }();
+  const int target = std::max(1, cap_blocks / kh);
+  int ns = std::max(1, std::min((int)((seq + 63) / 64), target));
+  int sl = (int)((seq + ns - 1) / ns + 63) / 64 * 64;
+  *n_splits = (int)((seq + sl - 1) / sl);
+  *split_len = sl;
+}
+
 torch::Tensor attn_decode_paged(torch::Tensor q, torch::Tensor kc,
                                 torch::Tensor vc, torch::Tensor page_table,
                                 int64_t seq_len, double scale) {
@@ -323,15 +343,8 @@ torch::Tensor attn_decode_paged(torch::Tensor q, torch::Tensor kc,
   TORCH_CHECK(group <= 8, "GQA group <= 8");
   TORCH_CHECK(hd == 32 || hd == 64 || hd == 128);
 
-  // Size splits for full-chip occupancy: the split kernel runs one
-  // 256-thread block per (kv_head, split); target ~1024 blocks so every CU
-  // carries ~4 blocks (16 waves) to hide the KV-read latency.
-  // ~512 blocks total: bigger splits keep the combine walk short (the
-  // combine is latency-bound at 1 thread per (head, dim) element) while
-  // the split kernel hides KV latency with in-loop prefetch.
-  int n_splits = std::max(1, std::min((int)((seq_len + 255) / 256), 512 / kh));
-  int split_len = (int)((seq_len + n_splits - 1) / n_splits + 63) / 64 * 64;
-  n_splits = (int)((seq_len + split_len - 1) / split_len);
+  int n_splits, split_len;
+  split_geometry(seq_len, kh, &n_splits, &split_len);
   auto out = torch::empty({hq, hd}, qc.options());
   const long khnsg = (long)kh * n_splits * group;
   auto stream = cur_stream();
@@ -381,6 +394,21 @@ torch::Tensor gemv(torch::Tensor x, torch::Tensor w,
                               x.options());
   launch_gemv(uptr(xc), uptr(w), uptr_mut(y), K, N, cur_stream());
   return y;
+}
+
+// Fused decode gate_up GEMV + SwiGLU: act = silu(x@Wg^T) * (x@Wu^T) with
+// W = [gate | up] rows (the fused w_gate_up layout). Allocation-free.
+void gemv_gateup(torch::Tensor x, torch::Tensor w, torch::Tensor act) {
+  CHECK_BF16_CUDA(x);
+  CHECK_BF16_CUDA(w);
+  CHECK_BF16_CUDA(act);
+  TORCH_CHECK(w.dim() == 2 && w.is_contiguous());
+  const int K = w.size(1), F2 = w.size(0);
+  TORCH_CHECK(F2 % 2 == 0 && (long)x.numel() == (long)K && K % 8 == 0);
+  TORCH_CHECK((long)act.numel() == (long)(F2 / 2));
+  auto xc = x.contiguous();
+  launch_gemv_gateup(uptr(xc), uptr(w), uptr_mut(act), K, F2 / 2,
+                     cur_stream());
 }
 
 // Tiled MFMA GEMM: C = A @ B, bf16, fp32 accumulation. Replaces library
@@ -498,9 +526,8 @@ torch::Tensor attn_decode_paged_ds(torch::Tensor q, torch::Tensor kc,
   const int hq = qc.size(0), hd = qc.size(1);
   const int page = kc.size(1), kh = kc.size(2);
   const int group = hq / kh;
-  int n_splits = std::max(1, std::min((int)((max_seq + 255) / 256), 512 / kh));
-  int split_len = (int)((max_seq + n_splits - 1) / n_splits + 63) / 64 * 64;
-  n_splits = (int)((max_seq + split_len - 1) / split_len);
+  int n_splits, split_len;
+  split_geometry(max_seq, kh, &n_splits, &split_len);
   auto out = out_opt.has_value() ? *out_opt : torch::empty({hq, hd}, qc.options());
   const long khnsg = (long)kh * n_splits * group;
   auto stream = cur_stream();
@@ -521,16 +548,18 @@ torch::Tensor attn_decode_paged_ds(torch::Tensor q, torch::Tensor kc,
 
 void sample_state(torch::Tensor logits, torch::Tensor temp_state,
                   torch::Tensor rng_state, torch::Tensor tok_hist,
-                  torch::Tensor step_state, torch::Tensor tok_slot) {
+                  torch::Tensor step_state, torch::Tensor tok_long) {
   CHECK_BF16_CUDA(logits);
   TORCH_CHECK(temp_state.scalar_type() == at::kFloat && temp_state.is_cuda(),
               "sample_state: temp_state must be a device f32 word");
+  TORCH_CHECK(tok_long.scalar_type() == at::kLong,
+              "sample_state: tok_long must be int64 (the embedding index)");
   auto lc = logits.contiguous();
   sample_state_kernel<<<1, 1024, 0, cur_stream()>>>(
       uptr(lc), (int)lc.numel(), temp_state.data_ptr<float>(),
       reinterpret_cast<uint32_t*>(rng_state.data_ptr<int>()),
       tok_hist.data_ptr<int>(), step_state.data_ptr<int>(),
-      tok_slot.data_ptr<int>());
+      reinterpret_cast<long long*>(tok_long.data_ptr<int64_t>()));
 }
 
 // Release the decode-attention scratch entries whose keys (per-layer KV
@@ -578,6 +607,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("sample_to", &sample_to, "async on-device sample into out[idx]");
   m.def("gemv", &gemv, "batch-1 decode GEMV (weight streaming)",
         py::arg("x"), py::arg("w"), py::arg("out") = py::none());
+  m.def("gemv_gateup", &gemv_gateup, "fused gate_up GEMV + SwiGLU (decode)");
   m.def("gemm", &gemm, "tiled MFMA GEMM (bf16, fp32 accum)");
   m.def("gemm_variant", &gemm_variant, "force GEMM kernel 128/256 (A/B)");
   m.def("gemm_fp8", &gemm_fp8, "fp8 e4m3 MFMA GEMM (rowwise scales)");

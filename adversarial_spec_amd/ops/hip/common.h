@@ -50,6 +50,19 @@ DEVINL bf16x8 pack8(const f32x8 &p) {
   return r;
 }
 
+// packed-bf16 dot: acc += sum_j a[j]*b[j] over one 16 B chunk (8 elems)
+// via 4 v_dot2_f32_bf16 — 4 VALU ops instead of 24 (16 unpacks + 8 FMAs).
+typedef __attribute__((__vector_size__(2 * sizeof(short)))) short bf16x2v_;
+
+DEVINL float dot8_bf16(const bf16x8 &a, const bf16x8 &b, float acc) {
+  const bf16x2v_ *ap = (const bf16x2v_ *)&a;
+  const bf16x2v_ *bp = (const bf16x2v_ *)&b;
+#pragma unroll
+  for (int q = 0; q < 4; ++q)
+    acc = __builtin_amdgcn_fdot2_f32_bf16(ap[q], bp[q], acc, false);
+  return acc;
+}
+
 // ---- wave/block reductions ---------------------------------------------
 
 DEVINL float wave_reduce_sum(float v) {

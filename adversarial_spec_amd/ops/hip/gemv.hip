@@ -14,19 +14,7 @@
 // slowdown in round 1).
 
 #include "common.h"
-
-typedef __attribute__((__vector_size__(2 * sizeof(short)))) short bf16x2v;
-
-// packed-bf16 dot: acc += sum_j a[j]*b[j] over one 16 B chunk (8 elems)
-// via 4 v_dot2_f32_bf16 — 4 VALU ops instead of 24 (16 unpacks + 8 FMAs).
-DEVINL float dot8_bf16(const bf16x8 &a, const bf16x8 &b, float acc) {
-  const bf16x2v *ap = (const bf16x2v *)&a;
-  const bf16x2v *bp = (const bf16x2v *)&b;
-#pragma unroll
-  for (int q = 0; q < 4; ++q)
-    acc = __builtin_amdgcn_fdot2_f32_bf16(ap[q], bp[q], acc, false);
-  return acc;
-}
+// dot8_bf16 (packed v_dot2_f32_bf16 row dot) comes from common.h
 
 // 16-lane-group sum (lanes p, p+1, .., p+15 with stride 1)
 DEVINL float group16_sum(float v) {
@@ -118,4 +106,64 @@ extern "C" void launch_gemv(const ushort_t *x, const ushort_t *w, ushort_t *y,
   } else {
     gemv_kernel<<<dim3((N + 15) / 16), 256, 0, stream>>>(x, w, y, K, N);
   }
+}
+
+// Fused gate_up GEMV + SwiGLU for the decode path:
+//   act[n] = silu(x @ Wg[n]) * (x @ Wu[n]),  W = [gate rows | up rows].
+// Each 32-lane group computes BOTH row dots for one n (same x chunks),
+// so the separate swiglu launch (+ the gu activation round-trip)
+// disappears from the ~330-kernel decode step. Weight bytes unchanged —
+// this is a launch/boundary fusion, not a traffic change.
+extern "C" __global__ void __launch_bounds__(256)
+gemv_gateup_kernel(const ushort_t *__restrict__ x,
+                   const ushort_t *__restrict__ w,
+                   ushort_t *__restrict__ act, int K, int F) {
+  const int lane = threadIdx.x & (WAVE - 1);
+  const int wid = threadIdx.x / WAVE;
+  const int rg = lane >> 5;       // row group within wave: 0..1
+  const int sl = lane & 31;       // k-slice lane: 0..31
+  const int n = blockIdx.x * 8 + wid * 2 + rg;
+  if (n >= F) return;
+
+  const ushort_t *wg = w + (size_t)n * K;
+  const ushort_t *wu = w + (size_t)(n + F) * K;
+  const int nc = K / 8;
+
+  float ag = 0.f, au = 0.f;
+  int c = sl;
+  for (; c + 96 < nc; c += 128) {
+    bf16x8 gv[4], uv[4], xv[4];
+#pragma unroll
+    for (int u = 0; u < 4; ++u) {
+      gv[u] = ((const bf16x8 *)wg)[c + 32 * u];
+      uv[u] = ((const bf16x8 *)wu)[c + 32 * u];
+      xv[u] = ((const bf16x8 *)x)[c + 32 * u];
+    }
+#pragma unroll
+    for (int u = 0; u < 4; ++u) {
+      ag = dot8_bf16(xv[u], gv[u], ag);
+      au = dot8_bf16(xv[u], uv[u], au);
+    }
+  }
+  for (; c < nc; c += 32) {
+    const bf16x8 xv = ((const bf16x8 *)x)[c];
+    ag = dot8_bf16(xv, ((const bf16x8 *)wg)[c], ag);
+    au = dot8_bf16(xv, ((const bf16x8 *)wu)[c], au);
+  }
+
+#pragma unroll
+  for (int off = 16; off > 0; off >>= 1) {
+    ag += __shfl_xor(ag, off, WAVE);
+    au += __shfl_xor(au, off, WAVE);
+  }
+  if (sl == 0) {
+    const float s = ag / (1.0f + __expf(-ag));  // silu(gate)
+    act[n] = f32_to_bf16(s * au);
+  }
+}
+
+extern "C" void launch_gemv_gateup(const ushort_t *x, const ushort_t *w,
+                                   ushort_t *act, int K, int F,
+                                   hipStream_t stream) {
+  gemv_gateup_kernel<<<dim3((F + 7) / 8), 256, 0, stream>>>(x, w, act, K, F);
 }

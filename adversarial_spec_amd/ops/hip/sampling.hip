@@ -86,21 +86,23 @@ sample_kernel(const ushort_t *__restrict__ logits, int vocab, float temp,
 //   rng_state:  uint32[1], advanced each call
 //   step_state: int32[1], the decode step index (bumped by bump_kernel)
 //   tok_hist:   int32[max_new] history the host polls every N tokens
-//   tok_slot:   int32[1] fixed slot feeding the next embedding lookup
+//   tok_long:   int64[1] embedding index for the next forward — written
+//               directly so the decode step drops the int32->int64 copy
+//               launch torch would otherwise add per token
 extern "C" __global__ void __launch_bounds__(1024)
 sample_state_kernel(const ushort_t *__restrict__ logits, int vocab,
                     const float *__restrict__ temp_state,
                     uint32_t *__restrict__ rng_state,
                     int *__restrict__ tok_hist,
                     const int *__restrict__ step_state,
-                    int *__restrict__ tok_slot) {
+                    long long *__restrict__ tok_long) {
   __shared__ int picked[1];
   const uint32_t seed = *rng_state;
   sample_body(logits, vocab, *temp_state, seed, picked);
   __syncthreads();
   if (threadIdx.x == 0) {
     const int tok = picked[0];
-    tok_slot[0] = tok;
+    tok_long[0] = (long long)tok;
     tok_hist[*step_state] = tok;
     *rng_state = hash_u32(seed ^ 0x6a09e667u) | 1u;  // never 0
   }

@@ -103,14 +103,17 @@ class TPContext:
 
     def all_reduce_(self, x: torch.Tensor) -> torch.Tensor:
         """In-place sum all-reduce of partial activations (RCCL on GPU,
-        gloo in CPU tests). Gloo lacks bf16: round-trip through fp32 there."""
+        gloo in CPU tests). Gloo lacks bf16 and CUDA buffers: round-trip
+        through fp32 on host there (the gloo path exists only so TP
+        numerics are testable without N GPUs — RCCL needs one device per
+        rank, so GPU-compute-with-gloo-collectives is the 1-GPU TP lane)."""
         if self.size == 1:
             return x
         backend = dist.get_backend(self.group)
-        if backend == "gloo" and x.dtype == torch.bfloat16:
-            xf = x.float()
+        if backend == "gloo" and (x.dtype == torch.bfloat16 or x.is_cuda):
+            xf = x.float().cpu()
             dist.all_reduce(xf, group=self.group)
-            x.copy_(xf.to(x.dtype))
+            x.copy_(xf.to(x.dtype).to(x.device))
         else:
             dist.all_reduce(x, group=self.group)
         return x

@@ -91,9 +91,13 @@ class TestGenerationGPU:
             return orig(ids)
 
         monkeypatch.setattr(eng.tokenizer, "decode", capture)
+        # the final decode(out_ids) is the LAST decode call of a generate
+        # (the stop scanner also calls decode per token on small windows)
         eng.generate("s", "u", max_tokens=12, temperature=0.0, timeout=300)
+        a = captured[-1]
+        captured.clear()
         eng.generate("s", "u", max_tokens=12, temperature=0.0, timeout=300)
-        a, b = captured
+        b = captured[-1]
         div = next((i for i, (p, q) in enumerate(zip(a, b)) if p != q), None)
         assert a == b, f"diverge at {div}: first={a} second={b}"
 
@@ -112,10 +116,12 @@ class TestGenerationGPU:
         monkeypatch.setenv("ADVSPEC_NO_GRAPH", "1")
         eng.generate("sys", "graph parity prompt", max_tokens=24,
                      temperature=0.0, timeout=300)
+        a = captured[-1]
+        captured.clear()
         monkeypatch.delenv("ADVSPEC_NO_GRAPH")
         eng.generate("sys", "graph parity prompt", max_tokens=24,
                      temperature=0.0, timeout=300)
-        a, b = captured
+        b = captured[-1]
         div = next((i for i, (p, q) in enumerate(zip(a, b)) if p != q), None)
         assert a == b, f"diverge at {div}: eager={a} graph={b}"
 

@@ -122,7 +122,7 @@ def main():
     tmp_r = torch.tensor([12345], dtype=torch.int32, device="cuda")
     tmp_h = torch.zeros(64, dtype=torch.int32, device="cuda")
     tmp_s = torch.zeros(1, dtype=torch.int32, device="cuda")
-    tmp_o = torch.zeros(1, dtype=torch.int32, device="cuda")
+    tmp_o = torch.zeros(1, dtype=torch.int64, device="cuda")
     comps["sample+bump"] = bench(lambda: hip.sample_state(
         W.logits.view(-1), tmp_t, tmp_r, tmp_h, tmp_s, tmp_o))
 
