@@ -86,26 +86,29 @@ __global__ void __launch_bounds__(256) attn_decode_split_kernel(
     for (int j = 0; j < 8; ++j) acc[gi][j] = 0.f;
   }
 
-  // software prefetch: next position's K/V rows are loaded while the
-  // current position's softmax update computes (the serial m/l/acc chain
-  // otherwise exposes full HBM latency every iteration).
+  // software prefetch, 2 iterations deep: the PMC profile shows this
+  // kernel 58% wave-parked on s_waitcnt with a 1-deep ring (the serial
+  // m/l/acc chain per iteration is shorter than the K/V load latency at
+  // ~1.7 blocks/CU), so keep TWO future positions' K/V rows in flight.
+  auto ldrow = [&](int pp, bf16x8 &kr, bf16x8 &vr) {
+    if (pp < limit) {
+      const int phys = page_table[pp / page];
+      const size_t row = ((size_t)phys * page + (pp % page)) * kh * hd;
+      kr = ((const bf16x8 *)(kc + row + (size_t)g * hd))[sl];
+      vr = ((const bf16x8 *)(vc + row + (size_t)g * hd))[sl];
+    }
+  };
   int p = start + sub;
-  bf16x8 kraw, vraw;
-  if (p < limit) {
-    const int phys0 = page_table[p / page];
-    const size_t row0 = ((size_t)phys0 * page + (p % page)) * kh * hd;
-    kraw = ((const bf16x8 *)(kc + row0 + (size_t)g * hd))[sl];
-    vraw = ((const bf16x8 *)(vc + row0 + (size_t)g * hd))[sl];
-  }
+  bf16x8 kring[2], vring[2];
+  ldrow(p, kring[0], vring[0]);
+  ldrow(p + subs, kring[1], vring[1]);
+  int cur = 0;
   for (; p < limit; ) {
     const int pn = p + subs;
-    bf16x8 kn, vn;
-    if (pn < limit) {
-      const int physn = page_table[pn / page];
-      const size_t rown = ((size_t)physn * page + (pn % page)) * kh * hd;
-      kn = ((const bf16x8 *)(kc + rown + (size_t)g * hd))[sl];
-      vn = ((const bf16x8 *)(vc + rown + (size_t)g * hd))[sl];
-    }
+    const bf16x8 kraw = kring[cur];
+    const bf16x8 vraw = vring[cur];
+    ldrow(p + 2 * subs, kring[cur], vring[cur]);
+    cur ^= 1;
     const f32x8 vd = unpack8(vraw);
 
 #pragma unroll
@@ -122,8 +125,6 @@ __global__ void __launch_bounds__(256) attn_decode_split_kernel(
         acc[gi][j] = acc[gi][j] * alpha + pex * vd.v[j];
       m[gi] = m_new;
     }
-    kraw = kn;
-    vraw = vn;
     p = pn;
   }
 
@@ -202,14 +203,17 @@ __global__ void __launch_bounds__(256) attn_decode_split_kernel(
 }
 
 // ---------------------------------------------------------------------------
-// Decode combine kernel: grid.x = hq, block = hd threads.
+// Decode combine kernel: grid = (hq, hd/64), block = 256 (4 waves).
 // out[h, :] = sum_splits(acc * exp(m - M)) / L_total
 // A SEPARATE kernel on purpose: a fused last-block-arrives combine needs
 // device-scope fences, which on the 8-XCD MI355X triggered cross-L2 traffic
 // that slowed the whole device ~5x (round-1 profiles).
+// Parallel walk: wave w strides splits w, w+4, ...; lane = dim. The old
+// 1-thread-per-(head,dim) single-block walk measured 10.5 us at 54 splits
+// (62% wave-parked) — as costly as the split pass itself.
 // ---------------------------------------------------------------------------
 
-extern "C" __global__ void __launch_bounds__(128)
+extern "C" __global__ void __launch_bounds__(256)
 attn_decode_combine_kernel(const float *__restrict__ ws_m,
                            const float *__restrict__ ws_l,
                            const float *__restrict__ ws_acc,
@@ -217,49 +221,55 @@ attn_decode_combine_kernel(const float *__restrict__ ws_m,
                            int group, int hd) {
   const int h = blockIdx.x;
   const int g = h / group, gi = h % group;
-  const int dd = threadIdx.x;
+  const int dd = blockIdx.y * 64 + (threadIdx.x & (WAVE - 1));
+  const int sg = threadIdx.x / WAVE;  // split stride group: 0..3
   if (dd >= hd) return;
 
-  // 8-way unrolled split walk keeps >=8 independent loads in flight
-  // (the walk is latency-bound: one thread per (head, dim) output).
-  float M = -INFINITY;
-  {
-    float m8[8] = {-INFINITY, -INFINITY, -INFINITY, -INFINITY,
-                   -INFINITY, -INFINITY, -INFINITY, -INFINITY};
-    int s = 0;
-    for (; s + 8 <= n_splits; s += 8) {
-#pragma unroll
-      for (int u = 0; u < 8; ++u)
-        m8[u] = fmaxf(m8[u], ws_m[((size_t)g * n_splits + s + u) * group + gi]);
-    }
-    for (; s < n_splits; ++s)
-      m8[0] = fmaxf(m8[0], ws_m[((size_t)g * n_splits + s) * group + gi]);
-#pragma unroll
-    for (int u = 0; u < 8; ++u) M = fmaxf(M, m8[u]);
+  __shared__ float red[4];
+  __shared__ float redL[4];
+
+  // pass 1: global max over this wave's split stride, merged in LDS
+  float m = -INFINITY;
+  for (int s = sg; s < n_splits; s += 4)
+    m = fmaxf(m, ws_m[((size_t)g * n_splits + s) * group + gi]);
+  if ((threadIdx.x & (WAVE - 1)) == 0) red[sg] = m;
+  __syncthreads();
+  const float M = fmaxf(fmaxf(red[0], red[1]), fmaxf(red[2], red[3]));
+
+  // pass 2: strided L/acc accumulation, 2 independent accumulators per
+  // wave for load ILP, merged via LDS
+  float L = 0.f, A0 = 0.f, A1 = 0.f;
+  int s = sg;
+  for (; s + 4 < n_splits; s += 8) {
+    const size_t b0 = ((size_t)g * n_splits + s) * group + gi;
+    const size_t b1 = ((size_t)g * n_splits + s + 4) * group + gi;
+    const float mw0 = ws_m[b0], mw1 = ws_m[b1];
+    const float sc0 = (mw0 == -INFINITY) ? 0.f : __expf(mw0 - M);
+    const float sc1 = (mw1 == -INFINITY) ? 0.f : __expf(mw1 - M);
+    L += ws_l[b0] * sc0 + ws_l[b1] * sc1;
+    A0 += ws_acc[b0 * hd + dd] * sc0;
+    A1 += ws_acc[b1 * hd + dd] * sc1;
   }
-  float L8[8] = {0.f}, A8[8] = {0.f};
-  int s = 0;
-  for (; s + 8 <= n_splits; s += 8) {
-#pragma unroll
-    for (int u = 0; u < 8; ++u) {
-      const size_t base = ((size_t)g * n_splits + s + u) * group + gi;
-      const float mw = ws_m[base];
-      const float sc = (mw == -INFINITY) ? 0.f : __expf(mw - M);
-      L8[u] += ws_l[base] * sc;
-      A8[u] += ws_acc[base * hd + dd] * sc;
-    }
-  }
-  for (; s < n_splits; ++s) {
-    const size_t base = ((size_t)g * n_splits + s) * group + gi;
-    const float mw = ws_m[base];
+  if (s < n_splits) {
+    const size_t b = ((size_t)g * n_splits + s) * group + gi;
+    const float mw = ws_m[b];
     const float sc = (mw == -INFINITY) ? 0.f : __expf(mw - M);
-    L8[0] += ws_l[base] * sc;
-    A8[0] += ws_acc[base * hd + dd] * sc;
+    L += ws_l[b] * sc;
+    A0 += ws_acc[b * hd + dd] * sc;
   }
-  float L = 0.f, A = 0.f;
-#pragma unroll
-  for (int u = 0; u < 8; ++u) { L += L8[u]; A += A8[u]; }
-  out[(size_t)h * hd + dd] = f32_to_bf16(L > 0.f ? A / L : 0.f);
+  float A = A0 + A1;
+
+  // merge the 4 waves: A via LDS columns, L via lane-0 scalars
+  __shared__ float accs[4][64];
+  accs[sg][threadIdx.x & (WAVE - 1)] = A;
+  if ((threadIdx.x & (WAVE - 1)) == 0) redL[sg] = L;
+  __syncthreads();
+  if (sg == 0) {
+    A = accs[0][threadIdx.x] + accs[1][threadIdx.x] + accs[2][threadIdx.x] +
+        accs[3][threadIdx.x];
+    L = redL[0] + redL[1] + redL[2] + redL[3];
+    out[(size_t)h * hd + dd] = f32_to_bf16(L > 0.f ? A / L : 0.f);
+  }
 }
 
 // ---------------------------------------------------------------------------
@@ -358,8 +368,9 @@ extern "C" void launch_attn_decode_split(
   }
 #undef DISPATCH_MG
   if (n_splits > 1) {
-    attn_decode_combine_kernel<<<dim3(kh * group), dim3(hd), 0, stream>>>(
-        ws_m, ws_l, ws_acc, out, n_splits, group, hd);
+    attn_decode_combine_kernel<<<dim3(kh * group, (hd + 63) / 64), 256, 0,
+                                 stream>>>(ws_m, ws_l, ws_acc, out, n_splits,
+                                           group, hd);
   }
 }
 
