@@ -33,13 +33,21 @@ DEVINL float group_reduce_sum(float v) {
 // ---------------------------------------------------------------------------
 // Decode split kernel.
 // grid.x = kh, grid.y = n_splits, block = 256 (4 waves).
-// Each wave covers SPLIT_LEN/4 = 64 consecutive positions of the split's
-// 256-position range; a wave runs 64/LPP positions concurrently (one per
-// 16-lane group at hd=128).
+//
+// The KV read is staged through LDS by async global_load_lds in 64-position
+// double-buffered tiles: register-ring prefetch measured latency-bound
+// (Little's law: ~1.7 KB in flight per CU vs ~9.4 KB needed at ~900-cycle
+// HBM latency; PMC: 58% wave-parked), while LDS-DMA keeps a whole tile
+// (8 glds x 1 KiB per wave) in flight behind a counted vmcnt, like the
+// prefill kernel's staging. Compute then walks the tile from LDS
+// (ds_read_b128, conflict-free: each 16-lane group reads one full 256-B
+// row) with the positions of a tile split across the block's 4 waves.
 //
 // Workspace (fp32): ws_m, ws_l: [kh, n_splits, group]
 //                   ws_acc:     [kh, n_splits, group, hd]
 // ---------------------------------------------------------------------------
+
+#define DTILE 64  // positions staged per LDS tile
 
 template <int LPP, int MG>
 __global__ void __launch_bounds__(256) attn_decode_split_kernel(
@@ -60,15 +68,18 @@ __global__ void __launch_bounds__(256) attn_decode_split_kernel(
   const int wid = threadIdx.x / WAVE;
   const int sub = lane / LPP;        // position sub-group within wave
   const int sl = lane % LPP;         // dim slice within position
-  const int pos_per_wave = split_len / 4;
   const int subs = WAVE / LPP;       // concurrent positions per wave
 
-  const int start = split * split_len + wid * pos_per_wave;
-  const int limit = min(seq_len, min(start + pos_per_wave,
-                                     (split + 1) * split_len));
+  const int start = split * split_len;
+  const int limit = min(seq_len, (split + 1) * split_len);
 
-  // q fragments: raw bf16 (the packed v_dot2 dot consumes bf16 directly —
-  // same fp32 products, no unpack pass)
+  // ONE shared array (guide §5 trap 4a); the stage buffers are reused as
+  // the wave-merge scratch after the tile loop (barrier-separated).
+  // Stage region: [2 buffers][K tile | V tile], tile = DTILE x hd bf16.
+  constexpr int TILE_E = DTILE * (LPP * 8);          // elements per tile
+  __shared__ __attribute__((aligned(16))) ushort_t lds[4 * TILE_E];
+
+  // q fragments: raw bf16 (the packed v_dot2 dot consumes bf16 directly)
   bf16x8 qraw[MG];
 #pragma unroll
   for (int gi = 0; gi < MG; ++gi) {
@@ -86,46 +97,84 @@ __global__ void __launch_bounds__(256) attn_decode_split_kernel(
     for (int j = 0; j < 8; ++j) acc[gi][j] = 0.f;
   }
 
-  // software prefetch, 2 iterations deep: the PMC profile shows this
-  // kernel 58% wave-parked on s_waitcnt with a 1-deep ring (the serial
-  // m/l/acc chain per iteration is shorter than the K/V load latency at
-  // ~1.7 blocks/CU), so keep TWO future positions' K/V rows in flight.
-  auto ldrow = [&](int pp, bf16x8 &kr, bf16x8 &vr) {
-    if (pp < limit) {
+  // stage one DTILE-position K/V tile into LDS buffer b via glds.
+  // Each wave issues LPP/4 x 1 KiB pieces for K and for V (4 waves x
+  // LPP/4 x 64/LPP rows = 64 = DTILE); a piece covers 1024/(hd*2) = subs
+  // consecutive positions; lane l handles the piece's position l/LPP,
+  // chunk l%LPP — lane-linear in LDS as glds requires. Out-of-range
+  // positions clamp to a valid row (their scores are masked in compute).
+  auto stage_tile = [&](int t0, int b) {
+    ushort_t *kimg = lds + (size_t)b * 2 * TILE_E;
+    ushort_t *vimg = kimg + TILE_E;
+#pragma unroll
+    for (int i = 0; i < LPP / 4; ++i) {  // pieces per wave per tensor
+      const int prow = (wid * (LPP / 4) + i) * subs;  // tile-local row
+      const int pp = min(t0 + prow + sub, seq_len - 1);
       const int phys = page_table[pp / page];
-      const size_t row = ((size_t)phys * page + (pp % page)) * kh * hd;
-      kr = ((const bf16x8 *)(kc + row + (size_t)g * hd))[sl];
-      vr = ((const bf16x8 *)(vc + row + (size_t)g * hd))[sl];
+      const size_t row = ((size_t)phys * page + (pp % page)) * kh * hd +
+                         (size_t)g * hd;
+      __builtin_amdgcn_global_load_lds(
+          (const __attribute__((address_space(1))) void *)(kc + row + sl * 8),
+          (__attribute__((address_space(3))) void *)(kimg +
+                                                     (size_t)prow * hd),
+          16, 0, 0);
+      __builtin_amdgcn_global_load_lds(
+          (const __attribute__((address_space(1))) void *)(vc + row + sl * 8),
+          (__attribute__((address_space(3))) void *)(vimg +
+                                                     (size_t)prow * hd),
+          16, 0, 0);
     }
   };
-  int p = start + sub;
-  bf16x8 kring[2], vring[2];
-  ldrow(p, kring[0], vring[0]);
-  ldrow(p + subs, kring[1], vring[1]);
-  int cur = 0;
-  for (; p < limit; ) {
-    const int pn = p + subs;
-    const bf16x8 kraw = kring[cur];
-    const bf16x8 vraw = vring[cur];
-    ldrow(p + 2 * subs, kring[cur], vring[cur]);
-    cur ^= 1;
-    const f32x8 vd = unpack8(vraw);
 
+  const int ntiles = (limit - start + DTILE - 1) / DTILE;
+  if (ntiles > 0) {
+    stage_tile(start, 0);
+    if (ntiles > 1) stage_tile(start + DTILE, 1);
+
+    for (int t = 0; t < ntiles; ++t) {
+      // wait for tile t's DMA (tile t+1 stays in flight: 2 tensors x
+      // LPP/4 pieces per wave = LPP/2 loads outstanding), then align.
+      if (ntiles > t + 1) {
+        if (LPP == 16) asm volatile("s_waitcnt vmcnt(8)" ::: "memory");
+        else if (LPP == 8) asm volatile("s_waitcnt vmcnt(4)" ::: "memory");
+        else asm volatile("s_waitcnt vmcnt(2)" ::: "memory");
+      } else {
+        asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+      }
+      __builtin_amdgcn_s_barrier();
+
+      const ushort_t *kimg = lds + (size_t)(t & 1) * 2 * TILE_E;
+      const ushort_t *vimg = kimg + TILE_E;
+      const int tbase = start + t * DTILE;
+      // this wave's positions within the tile: wid*DTILE/4 .. +DTILE/4
+      const int wbeg = wid * (DTILE / 4);
+      const int wend = min(wbeg + DTILE / 4, limit - tbase);
+      for (int r = wbeg + sub; r < wend; r += subs) {
+        const bf16x8 kraw = ((const bf16x8 *)(kimg + (size_t)r * hd))[sl];
+        const bf16x8 vraw = ((const bf16x8 *)(vimg + (size_t)r * hd))[sl];
+        const f32x8 vd = unpack8(vraw);
 #pragma unroll
-    for (int gi = 0; gi < MG; ++gi) {
-      if (gi >= group) break;
-      const float dot = dot8_bf16(qraw[gi], kraw, 0.f);
-      const float s = group_reduce_sum<LPP>(dot) * scale;
-      const float m_new = fmaxf(m[gi], s);
-      const float alpha = __expf(m[gi] - m_new);
-      const float pex = __expf(s - m_new);
-      l[gi] = l[gi] * alpha + pex;
+        for (int gi = 0; gi < MG; ++gi) {
+          if (gi >= group) break;
+          const float dot = dot8_bf16(qraw[gi], kraw, 0.f);
+          const float s = group_reduce_sum<LPP>(dot) * scale;
+          const float m_new = fmaxf(m[gi], s);
+          const float alpha = __expf(m[gi] - m_new);
+          const float pex = __expf(s - m_new);
+          l[gi] = l[gi] * alpha + pex;
 #pragma unroll
-      for (int j = 0; j < 8; ++j)
-        acc[gi][j] = acc[gi][j] * alpha + pex * vd.v[j];
-      m[gi] = m_new;
+          for (int j = 0; j < 8; ++j)
+            acc[gi][j] = acc[gi][j] * alpha + pex * vd.v[j];
+          m[gi] = m_new;
+        }
+      }
+      // issue tile t+2 into the buffer just consumed (all waves have
+      // finished their ds_reads of it: they run under the same wave's
+      // program order... other waves may still be reading buffer t&1, so
+      // re-align first, then issue.
+      __builtin_amdgcn_s_barrier();
+      if (t + 2 < ntiles) stage_tile(tbase + 2 * DTILE, t & 1);
     }
-    p = pn;
   }
 
   // merge position sub-groups within the wave (lanes xor LPP, 2*LPP, ...)
@@ -149,12 +198,12 @@ __global__ void __launch_bounds__(256) attn_decode_split_kernel(
     }
   }
 
-  // merge the 4 waves via LDS, then write this split's partial state.
-  // LDS: per wave: m[group], l[group], acc[group][hd]
-  extern __shared__ __attribute__((aligned(16))) char smem[];
-  float *lm = (float *)smem;                       // [4][MAXG]
-  float *ll = lm + 4 * MAXG;                       // [4][MAXG]
-  float *lacc = ll + 4 * MAXG;                     // [4][MAXG][hd]
+  // merge the 4 waves via LDS (reusing the stage region), then write this
+  // split's partial state. Layout: m[4][MAXG], l[4][MAXG], acc[4][MAXG][hd]
+  __syncthreads();  // all ds_reads of the stage image done on every wave
+  float *lm = (float *)lds;
+  float *ll = lm + 4 * MAXG;
+  float *lacc = ll + 4 * MAXG;
 
   if (lane < LPP) {  // one lane per dim slice (sub==0 lanes)
 #pragma unroll
@@ -341,7 +390,7 @@ extern "C" void launch_attn_decode_split(
     int hd, int page, int split_len, int n_splits, float *ws_m, float *ws_l,
     float *ws_acc, const int *pos_ptr, ushort_t *out, hipStream_t stream) {
   dim3 grid(kh, n_splits);
-  const int lds = (4 * MAXG * 2 + 4 * MAXG * hd) * sizeof(float);
+  const int lds = 0;  // all LDS is static in the kernel (stage + merge)
   // MG = smallest supported bound >= group keeps the per-head state arrays
   // (q fragments + online-softmax accumulators) sized to the real GQA
   // group: MG=8 cost 194 VGPR (2 waves/SIMD); MG=4 fits 4 waves/SIMD.
