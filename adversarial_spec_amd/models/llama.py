@@ -78,8 +78,13 @@ class PagedKVCache:
         # per-layer page pools: [n_layers, n_pages, page_size, kh, hd]
         self.k = torch.zeros(nl, n_pages, page_size, kh, hd, device=device, dtype=dtype)
         self.v = torch.zeros(nl, n_pages, page_size, kh, hd, device=device, dtype=dtype)
-        # identity mapping by default; kept as a real table for paged reads
+        # identity mapping by default; kept as a real table for paged reads.
+        # `identity` certifies page_table[i] == i to the kernels (pure
+        # address math, no table read) and to chunked prefill (which views
+        # the pool as contiguous [seq, kh, hd]); any future eviction logic
+        # must clear it.
         self.page_table = torch.arange(n_pages, device=device, dtype=torch.int32)
+        self.identity = True
         self.seq_len = 0
         self.max_seq = n_pages * page_size
 
@@ -386,7 +391,7 @@ class LlamaModel:
                 seq = max_seq_bound if pos_state is not None else pos0 + 1
                 attn = ops.attn_decode_paged(
                     q[0], cache.k[i], cache.v[i], cache.page_table, seq,
-                    self.scale, pos_state=pos_state,
+                    self.scale, pos_state=pos_state, identity=cache.identity,
                 ).unsqueeze(0)
             attn_out = proj(attn.reshape(t, h * hd), L, Qd, "wo")
             if self.tp is not None and self.tp.size > 1:
@@ -489,6 +494,7 @@ class LlamaModel:
             ops.attn_decode_paged(
                 q[0], cache.k[i], cache.v[i], cache.page_table,
                 max_seq_bound, self.scale, pos_state=pos_state, out=W.attn,
+                identity=cache.identity,
             )
             proj(W.attn.view(1, h * hd), L, Qd, "wo", W.attn_out)
             if self.tp is not None and self.tp.size > 1:

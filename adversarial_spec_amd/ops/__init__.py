@@ -121,7 +121,11 @@ def attn_decode_paged(
     scale: Optional[float] = None,
     pos_state: Optional[torch.Tensor] = None,
     out: Optional[torch.Tensor] = None,
+    identity: bool = False,
 ) -> torch.Tensor:
+    """identity=True promises page_table[i] == i (the engine's single-pool
+    cache): the split kernel then does pure address math — the per-lane
+    table read otherwise forces a DMA-queue drain per staged tile."""
     if _on_gpu(q):
         import math
 
@@ -131,9 +135,10 @@ def attn_decode_paged(
             # the static max bound sizing the split geometry.
             return _require_hip().attn_decode_paged_ds(
                 q, k_cache, v_cache, page_table, pos_state, seq_len, s,
-                out,
+                out, identity,
             )
-        return _require_hip().attn_decode_paged(q, k_cache, v_cache, page_table, seq_len, s)
+        return _require_hip().attn_decode_paged(
+            q, k_cache, v_cache, page_table, seq_len, s, identity)
     return torch_ref.attn_decode_paged(q, k_cache, v_cache, page_table, seq_len, scale)
 
 

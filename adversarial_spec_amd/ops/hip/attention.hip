@@ -47,9 +47,9 @@ DEVINL float group_reduce_sum(float v) {
 //                   ws_acc:     [kh, n_splits, group, hd]
 // ---------------------------------------------------------------------------
 
-#define DTILE 64  // positions staged per LDS tile
+#define DTILE 64   // positions staged per LDS tile
 
-template <int LPP, int MG>
+template <int LPP, int MG, bool IDENT>
 __global__ void __launch_bounds__(256) attn_decode_split_kernel(
     const ushort_t *__restrict__ q,   // [hq, hd]
     const ushort_t *__restrict__ kc, const ushort_t *__restrict__ vc,
@@ -75,18 +75,12 @@ __global__ void __launch_bounds__(256) attn_decode_split_kernel(
 
   // ONE shared array (guide §5 trap 4a); the stage buffers are reused as
   // the wave-merge scratch after the tile loop (barrier-separated).
-  // Stage region: [2 buffers][K tile | V tile], tile = DTILE x hd bf16.
+  // CRITICAL (.s-verified): the function must contain NO ds_write and no
+  // LDS-region type-punning — either makes hipcc order every ds_read of
+  // the stage image behind `s_waitcnt vmcnt(0)`, draining the whole DMA
+  // pipeline once per position (the 16 us/dispatch plateau of rounds 1-2).
   constexpr int TILE_E = DTILE * (LPP * 8);          // elements per tile
   __shared__ __attribute__((aligned(16))) ushort_t lds[4 * TILE_E];
-
-  // q fragments: raw bf16 (the packed v_dot2 dot consumes bf16 directly)
-  bf16x8 qraw[MG];
-#pragma unroll
-  for (int gi = 0; gi < MG; ++gi) {
-    if (gi < group) {
-      qraw[gi] = ((const bf16x8 *)(q + ((size_t)(g * group + gi)) * hd))[sl];
-    }
-  }
 
   float m[MG], l[MG], acc[MG][8];
 #pragma unroll
@@ -110,7 +104,11 @@ __global__ void __launch_bounds__(256) attn_decode_split_kernel(
     for (int i = 0; i < LPP / 4; ++i) {  // pieces per wave per tensor
       const int prow = (wid * (LPP / 4) + i) * subs;  // tile-local row
       const int pp = min(t0 + prow + sub, seq_len - 1);
-      const int phys = page_table[pp / page];
+      // IDENT: the engine's single-pool cache uses the identity page
+      // table — pure address math, no table read. The general path pays
+      // an ordinary VMEM load here (hipcc then drains the DMA queue at
+      // its use — correct but slower; only non-identity tables take it).
+      const int phys = IDENT ? (pp / page) : page_table[pp / page];
       const size_t row = ((size_t)phys * page + (pp % page)) * kh * hd +
                          (size_t)g * hd;
       __builtin_amdgcn_global_load_lds(
@@ -125,6 +123,22 @@ __global__ void __launch_bounds__(256) attn_decode_split_kernel(
           16, 0, 0);
     }
   };
+
+  // q fragments by plain loads BEFORE any glds is issued. The asm "+v"
+  // use right after forces hipcc's counted wait HERE, while the VMEM
+  // queue holds only the q loads: deferred to the first dot inside the
+  // position loop, the wait count is loop-variant (DMA in flight), so
+  // hipcc emits `s_waitcnt vmcnt(0)` INSIDE the loop and drains the DMA
+  // pipeline once per position (.s-verified failure mode).
+  typedef __attribute__((__vector_size__(8 * sizeof(short)))) short v8s;
+  v8s qraw[MG];
+#pragma unroll
+  for (int gi = 0; gi < MG; ++gi)
+    if (gi < group)
+      qraw[gi] = ((const v8s *)(q + ((size_t)(g * group + gi)) * hd))[sl];
+#pragma unroll
+  for (int gi = 0; gi < MG; ++gi)
+    asm volatile("" : "+v"(qraw[gi]));
 
   const int ntiles = (limit - start + DTILE - 1) / DTILE;
   if (ntiles > 0) {
@@ -142,7 +156,6 @@ __global__ void __launch_bounds__(256) attn_decode_split_kernel(
         asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
       }
       __builtin_amdgcn_s_barrier();
-
       const ushort_t *kimg = lds + (size_t)(t & 1) * 2 * TILE_E;
       const ushort_t *vimg = kimg + TILE_E;
       const int tbase = start + t * DTILE;
@@ -150,13 +163,16 @@ __global__ void __launch_bounds__(256) attn_decode_split_kernel(
       const int wbeg = wid * (DTILE / 4);
       const int wend = min(wbeg + DTILE / 4, limit - tbase);
       for (int r = wbeg + sub; r < wend; r += subs) {
-        const bf16x8 kraw = ((const bf16x8 *)(kimg + (size_t)r * hd))[sl];
-        const bf16x8 vraw = ((const bf16x8 *)(vimg + (size_t)r * hd))[sl];
+        const v8s kv_ = ((const v8s *)(kimg + (size_t)r * hd))[sl];
+        const v8s vv_ = ((const v8s *)(vimg + (size_t)r * hd))[sl];
+        const bf16x8 kraw = __builtin_bit_cast(bf16x8, kv_);
+        const bf16x8 vraw = __builtin_bit_cast(bf16x8, vv_);
         const f32x8 vd = unpack8(vraw);
 #pragma unroll
         for (int gi = 0; gi < MG; ++gi) {
           if (gi >= group) break;
-          const float dot = dot8_bf16(qraw[gi], kraw, 0.f);
+          const float dot =
+              dot8_bf16(__builtin_bit_cast(bf16x8, qraw[gi]), kraw, 0.f);
           const float s = group_reduce_sum<LPP>(dot) * scale;
           const float m_new = fmaxf(m[gi], s);
           const float alpha = __expf(m[gi] - m_new);
@@ -277,36 +293,49 @@ attn_decode_combine_kernel(const float *__restrict__ ws_m,
   __shared__ float red[4];
   __shared__ float redL[4];
 
-  // pass 1: global max over this wave's split stride, merged in LDS
+  // pass 1: global max over this wave's split stride (4 independent max
+  // chains — a serial walk exposed full cross-XCD latency per entry)
   float m = -INFINITY;
-  for (int s = sg; s < n_splits; s += 4)
-    m = fmaxf(m, ws_m[((size_t)g * n_splits + s) * group + gi]);
+  {
+    float m4[4] = {-INFINITY, -INFINITY, -INFINITY, -INFINITY};
+    int s = sg;
+    for (; s + 12 < n_splits; s += 16) {
+#pragma unroll
+      for (int u = 0; u < 4; ++u)
+        m4[u] = fmaxf(m4[u],
+                      ws_m[((size_t)g * n_splits + s + 4 * u) * group + gi]);
+    }
+    for (; s < n_splits; s += 4)
+      m4[0] = fmaxf(m4[0], ws_m[((size_t)g * n_splits + s) * group + gi]);
+    m = fmaxf(fmaxf(m4[0], m4[1]), fmaxf(m4[2], m4[3]));
+  }
   if ((threadIdx.x & (WAVE - 1)) == 0) red[sg] = m;
   __syncthreads();
   const float M = fmaxf(fmaxf(red[0], red[1]), fmaxf(red[2], red[3]));
 
-  // pass 2: strided L/acc accumulation, 2 independent accumulators per
-  // wave for load ILP, merged via LDS
-  float L = 0.f, A0 = 0.f, A1 = 0.f;
+  // pass 2: strided L/acc accumulation, 4 independent accumulators per
+  // wave (16 concurrent loads across the block) — the ws round-trip is
+  // cross-XCD latency-bound, so ILP depth is the lever
+  float L = 0.f, Au[4] = {0.f, 0.f, 0.f, 0.f};
   int s = sg;
-  for (; s + 4 < n_splits; s += 8) {
-    const size_t b0 = ((size_t)g * n_splits + s) * group + gi;
-    const size_t b1 = ((size_t)g * n_splits + s + 4) * group + gi;
-    const float mw0 = ws_m[b0], mw1 = ws_m[b1];
-    const float sc0 = (mw0 == -INFINITY) ? 0.f : __expf(mw0 - M);
-    const float sc1 = (mw1 == -INFINITY) ? 0.f : __expf(mw1 - M);
-    L += ws_l[b0] * sc0 + ws_l[b1] * sc1;
-    A0 += ws_acc[b0 * hd + dd] * sc0;
-    A1 += ws_acc[b1 * hd + dd] * sc1;
+  for (; s + 12 < n_splits; s += 16) {
+#pragma unroll
+    for (int u = 0; u < 4; ++u) {
+      const size_t b = ((size_t)g * n_splits + s + 4 * u) * group + gi;
+      const float mw = ws_m[b];
+      const float sc = (mw == -INFINITY) ? 0.f : __expf(mw - M);
+      L += ws_l[b] * sc;
+      Au[u] += ws_acc[b * hd + dd] * sc;
+    }
   }
-  if (s < n_splits) {
+  for (; s < n_splits; s += 4) {
     const size_t b = ((size_t)g * n_splits + s) * group + gi;
     const float mw = ws_m[b];
     const float sc = (mw == -INFINITY) ? 0.f : __expf(mw - M);
     L += ws_l[b] * sc;
-    A0 += ws_acc[b * hd + dd] * sc;
+    Au[0] += ws_acc[b * hd + dd] * sc;
   }
-  float A = A0 + A1;
+  float A = (Au[0] + Au[1]) + (Au[2] + Au[3]);
 
   // merge the 4 waves: A via LDS columns, L via lane-0 scalars
   __shared__ float accs[4][64];
@@ -388,26 +417,29 @@ extern "C" void launch_attn_decode_split(
     const ushort_t *q, const ushort_t *kc, const ushort_t *vc,
     const int *page_table, int seq_len, float scale, int kh, int group,
     int hd, int page, int split_len, int n_splits, float *ws_m, float *ws_l,
-    float *ws_acc, const int *pos_ptr, ushort_t *out, hipStream_t stream) {
+    float *ws_acc, const int *pos_ptr, ushort_t *out, int identity,
+    hipStream_t stream) {
   dim3 grid(kh, n_splits);
   const int lds = 0;  // all LDS is static in the kernel (stage + merge)
   // MG = smallest supported bound >= group keeps the per-head state arrays
   // (q fragments + online-softmax accumulators) sized to the real GQA
-  // group: MG=8 cost 194 VGPR (2 waves/SIMD); MG=4 fits 4 waves/SIMD.
-#define DISPATCH_MG(LPP)                                                       \
+  // group; IDENT elides the page-table read (identity single-pool cache).
+#define DISPATCH_ONE(LPP, MG)                                                  \
   do {                                                                         \
-    if (group <= 2)                                                            \
-      attn_decode_split_kernel<LPP, 2><<<grid, 256, lds, stream>>>(            \
-          q, kc, vc, page_table, seq_len, scale, kh, group, hd, page,          \
-          split_len, ws_m, ws_l, ws_acc, pos_ptr, out);                        \
-    else if (group <= 4)                                                       \
-      attn_decode_split_kernel<LPP, 4><<<grid, 256, lds, stream>>>(            \
+    if (identity)                                                              \
+      attn_decode_split_kernel<LPP, MG, true><<<grid, 256, lds, stream>>>(     \
           q, kc, vc, page_table, seq_len, scale, kh, group, hd, page,          \
           split_len, ws_m, ws_l, ws_acc, pos_ptr, out);                        \
     else                                                                       \
-      attn_decode_split_kernel<LPP, 8><<<grid, 256, lds, stream>>>(            \
+      attn_decode_split_kernel<LPP, MG, false><<<grid, 256, lds, stream>>>(    \
           q, kc, vc, page_table, seq_len, scale, kh, group, hd, page,          \
           split_len, ws_m, ws_l, ws_acc, pos_ptr, out);                        \
+  } while (0)
+#define DISPATCH_MG(LPP)                                                       \
+  do {                                                                         \
+    if (group <= 2) DISPATCH_ONE(LPP, 2);                                      \
+    else if (group <= 4) DISPATCH_ONE(LPP, 4);                                 \
+    else DISPATCH_ONE(LPP, 8);                                                 \
   } while (0)
 
   switch (hd / 8) {
@@ -416,6 +448,7 @@ extern "C" void launch_attn_decode_split(
     case 16: DISPATCH_MG(16); break;
   }
 #undef DISPATCH_MG
+#undef DISPATCH_ONE
   if (n_splits > 1) {
     attn_decode_combine_kernel<<<dim3(kh * group, (hd + 63) / 64), 256, 0,
                                  stream>>>(ws_m, ws_l, ws_acc, out, n_splits,

@@ -19,7 +19,7 @@ extern "C" {
 void launch_attn_decode_split(const ushort_t*, const ushort_t*, const ushort_t*,
                               const int*, int, float, int, int, int, int, int,
                               int, float*, float*, float*, const int*,
-                              ushort_t*, hipStream_t);
+                              ushort_t*, int, hipStream_t);
 void launch_attn_prefill_simple(const ushort_t*, const ushort_t*, const ushort_t*,
                                 ushort_t*, int, int, int, float, int, int, int,
                                 int, hipStream_t);
@@ -331,7 +331,8 @@ static void split_geometry(long seq, int kh, int* n_splits, int* split_len) {
 
 torch::Tensor attn_decode_paged(torch::Tensor q, torch::Tensor kc,
                                 torch::Tensor vc, torch::Tensor page_table,
-                                int64_t seq_len, double scale) {
+                                int64_t seq_len, double scale,
+                                bool identity) {
   CHECK_BF16_CUDA(q);
   CHECK_BF16_CUDA(kc);
   auto qc = q.contiguous();
@@ -358,7 +359,7 @@ torch::Tensor attn_decode_paged(torch::Tensor q, torch::Tensor kc,
                            page_table.data_ptr<int>(), (int)seq_len,
                            (float)scale, kh, group, hd, page, split_len,
                            n_splits, ws, ws + khnsg, ws + 2 * khnsg,
-                           nullptr, uptr_mut(out), stream);
+                           nullptr, uptr_mut(out), identity ? 1 : 0, stream);
   return out;
 }
 
@@ -520,7 +521,8 @@ torch::Tensor attn_decode_paged_ds(torch::Tensor q, torch::Tensor kc,
                                    torch::Tensor vc, torch::Tensor page_table,
                                    torch::Tensor pos_state, int64_t max_seq,
                                    double scale,
-                                   c10::optional<torch::Tensor> out_opt) {
+                                   c10::optional<torch::Tensor> out_opt,
+                                   bool identity) {
   CHECK_BF16_CUDA(q);
   auto qc = q.contiguous();
   const int hq = qc.size(0), hd = qc.size(1);
@@ -542,7 +544,7 @@ torch::Tensor attn_decode_paged_ds(torch::Tensor q, torch::Tensor kc,
                            (float)scale, kh, group, hd, page, split_len,
                            n_splits, ws, ws + khnsg, ws + 2 * khnsg,
                            pos_state.data_ptr<int>(), uptr_mut(out),
-                           stream);
+                           identity ? 1 : 0, stream);
   return out;
 }
 
@@ -602,7 +604,9 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("attn_prefill", &attn_prefill, "causal prefill attention");
   m.def("attn_prefill_simple", &attn_prefill_simple, "non-MFMA prefill (anchor)");
   m.def("mfma_probe16", &mfma_probe16, "MFMA 16x16x32 fragment-map probe");
-  m.def("attn_decode_paged", &attn_decode_paged, "paged decode attention");
+  m.def("attn_decode_paged", &attn_decode_paged, "paged decode attention",
+        py::arg("q"), py::arg("kc"), py::arg("vc"), py::arg("page_table"),
+        py::arg("seq_len"), py::arg("scale"), py::arg("identity") = false);
   m.def("sample", &sample, "fused temperature softmax sample");
   m.def("sample_to", &sample_to, "async on-device sample into out[idx]");
   m.def("gemv", &gemv, "batch-1 decode GEMV (weight streaming)",
@@ -623,7 +627,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "graph-mode paged decode attention (device pos)",
         py::arg("q"), py::arg("kc"), py::arg("vc"), py::arg("page_table"),
         py::arg("pos_state"), py::arg("max_seq"), py::arg("scale"),
-        py::arg("out") = py::none());
+        py::arg("out") = py::none(), py::arg("identity") = false);
   m.def("sample_state", &sample_state, "graph-mode on-device sampling");
   m.def("bump", &bump, "graph-mode pos/step bump");
   m.def("ws_release", &ws_release,

@@ -30,9 +30,15 @@ def child(blocks: str) -> None:
         pos = torch.tensor([seq - 1], dtype=torch.int32, device="cuda")
         out = torch.empty(kh * group, hd, device="cuda").bfloat16()
 
-        # correctness vs CPU fp32 reference (first call)
+        # correctness vs CPU fp32 reference (first call) — general path
         got = ops.attn_decode_paged(q, kc, vc, pt, seq, None, pos_state=pos,
                                     out=out)
+        # identity fast path must agree bitwise (the table IS identity)
+        got_id = ops.attn_decode_paged(q, kc, vc, pt, seq, None,
+                                       pos_state=pos,
+                                       out=torch.empty_like(out),
+                                       identity=True)
+        assert torch.equal(got, got_id), "identity path mismatch"
         want = torch_ref.attn_decode_paged(
             q.float().cpu(), kc.float().cpu(), vc.float().cpu(), pt.cpu(),
             seq, None)
@@ -41,7 +47,7 @@ def child(blocks: str) -> None:
 
         def run():
             ops.attn_decode_paged(q, kc, vc, pt, seq, None, pos_state=pos,
-                                  out=out)
+                                  out=out, identity=True)
 
         for _ in range(10):
             run()
