@@ -46,6 +46,7 @@ ACTIONS = [
     "sessions",
     "bedrock",
     "local",
+    "serve",
 ]
 
 
@@ -615,9 +616,65 @@ def output_results(
 # main
 # ---------------------------------------------------------------------------
 
+def handle_serve(args: argparse.Namespace) -> int:
+    """`serve [status|stop]`: persistent engine daemon (daemon.py).
+
+    Round 2+ of a skill session skips the ~40 s per-process model init:
+    `critique` transparently forwards to a live daemon whose engine cache
+    keeps opponent weights resident in HBM3E between rounds.
+    """
+    from .. import daemon
+
+    sub = args.profile_name
+    if sub == "status":
+        alive = daemon.ping()
+        print(f"daemon: {'running' if alive else 'not running'} "
+              f"({daemon.SOCKET_PATH})")
+        return EXIT_OK if alive else EXIT_ERROR
+    if sub == "stop":
+        ok = daemon.stop()
+        print("daemon: stopped" if ok else "daemon: not running")
+        return EXIT_OK
+    if sub not in (None, "start"):
+        print(f"Unknown serve subcommand: {sub}", file=sys.stderr)
+        return EXIT_ERROR
+    srv = daemon.serve()
+    print(f"adversarial-spec daemon listening on {daemon.SOCKET_PATH}",
+          file=sys.stderr)
+    try:
+        srv.serve_forever()
+    except KeyboardInterrupt:
+        pass
+    finally:
+        srv.server_close()
+    return EXIT_OK
+
+
 def main(argv: Optional[list[str]] = None) -> int:
     parser = create_parser()
     args = parser.parse_args(argv)
+
+    if args.action == "serve":
+        return handle_serve(args)
+
+    # a live daemon serves the expensive actions with warm engines; same
+    # argv + stdin forwarded, stdout/exit relayed (ADVSPEC_NO_DAEMON=1 or
+    # no daemon -> unchanged in-process path)
+    if args.action in ("critique", "export-tasks"):
+        from .. import daemon
+
+        raw_argv = list(argv) if argv is not None else sys.argv[1:]
+        if daemon.ping():
+            stdin_text = "" if args.resume else sys.stdin.read()
+            fwd = daemon.try_forward(raw_argv, stdin_text)
+            if fwd is not None:
+                code, out, err = fwd
+                sys.stdout.write(out)
+                sys.stderr.write(err)
+                return code
+            # daemon vanished mid-flight: fall through with the spec we read
+            if not args.resume:
+                sys.stdin = __import__("io").StringIO(stdin_text)
 
     code = handle_info_command(args)
     if code is not None:

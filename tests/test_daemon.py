@@ -1,0 +1,82 @@
+"""Engine daemon round-trip (debate.py serve): warm-process forwarding."""
+
+from __future__ import annotations
+
+import json
+import threading
+
+import pytest
+
+from adversarial_spec_amd import daemon
+from adversarial_spec_amd.cli import debate as cli
+
+SPEC = "# Payments API\n\nA spec that needs critique."
+
+
+@pytest.fixture
+def live_daemon(tmp_path, monkeypatch):
+    sock = tmp_path / "d.sock"
+    monkeypatch.setattr(daemon, "SOCKET_PATH", sock)
+    srv = daemon.serve(sock)
+    t = threading.Thread(target=srv.serve_forever, daemon=True)
+    t.start()
+    yield sock, srv
+    srv.shutdown()
+    srv.server_close()
+
+
+class TestDaemon:
+    def test_ping(self, live_daemon):
+        sock, _srv = live_daemon
+        assert daemon.ping(sock)
+        assert not daemon.ping(sock.parent / "missing.sock")
+
+    def test_forward_critique_round_trip(self, live_daemon, monkeypatch):
+        sock, srv = live_daemon
+        monkeypatch.delenv("ADVSPEC_IN_DAEMON", raising=False)
+        argv = ["critique", "--models", "stub/agree", "--json"]
+        for i in range(2):  # second request hits the WARM daemon process
+            fwd = daemon.try_forward(argv, SPEC, sock)
+            assert fwd is not None, "daemon did not answer"
+            code, out, err = fwd
+            assert code == 0, err
+            payload = json.loads(out)
+            assert payload["all_agreed"] is True
+            assert payload["results"][0]["model"] == "stub/agree"
+        assert srv.requests_served == 2
+
+    def test_no_forward_inside_daemon(self, live_daemon, monkeypatch):
+        sock, _ = live_daemon
+        monkeypatch.setenv("ADVSPEC_IN_DAEMON", "1")
+        assert daemon.try_forward(["critique"], SPEC, sock) is None
+        monkeypatch.delenv("ADVSPEC_IN_DAEMON")
+        monkeypatch.setenv("ADVSPEC_NO_DAEMON", "1")
+        assert daemon.try_forward(["critique"], SPEC, sock) is None
+
+    def test_cli_serve_status_and_stop(self, live_daemon, monkeypatch, capsys):
+        sock, srv = live_daemon
+        assert cli.main(["serve", "status"]) == 0
+        assert "running" in capsys.readouterr().out
+        assert daemon.stop(sock)
+        # server thread shuts down; give it a beat
+        import time
+
+        for _ in range(50):
+            if not daemon.ping(sock):
+                break
+            time.sleep(0.05)
+
+    def test_cli_critique_uses_daemon(self, live_daemon, monkeypatch, capsys):
+        """debate.py critique transparently forwards when a daemon runs."""
+        sock, srv = live_daemon
+        monkeypatch.delenv("ADVSPEC_IN_DAEMON", raising=False)
+        monkeypatch.delenv("ADVSPEC_NO_DAEMON", raising=False)
+        import io
+        import sys as _sys
+
+        monkeypatch.setattr(_sys, "stdin", io.StringIO(SPEC))
+        code = cli.main(["critique", "--models", "stub/agree", "--json"])
+        assert code == 0
+        payload = json.loads(capsys.readouterr().out)
+        assert payload["all_agreed"] is True
+        assert srv.requests_served == 1

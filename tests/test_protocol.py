@@ -163,3 +163,52 @@ class TestCostTracker:
         ct.add("gpt-4o", 100, 200)
         s = ct.summary()
         assert "Cost Summary" in s and "gpt-4o" in s
+
+
+class TestMutationKillers:
+    """Kill-the-mutant tests for survivors of tools/mutation_check.py
+    (mirrors the reference's mutmut-driven test additions,
+    test_models.py:856-872 class of tests)."""
+
+    def test_model_response_agreed_defaults_false(self):
+        from adversarial_spec_amd.protocol import ModelResponse
+
+        assert ModelResponse(model="m").agreed is False
+
+    def test_summary_truncates_at_exactly_300_default(self):
+        # 300 chars pass through untouched; 301 gains the ellipsis
+        assert get_critique_summary("x" * 300) == "x" * 300
+        assert get_critique_summary("x" * 301) == "x" * 300 + "..."
+
+    def test_summary_spec_tag_at_position_zero_keeps_response(self):
+        # spec_start == 0 must NOT slice to the empty prefix
+        r = "[SPEC]\nbody\n[/SPEC]"
+        assert get_critique_summary(r) == r
+
+    def test_summary_spec_tag_mid_response(self):
+        r = "critique text\n[SPEC]\nbody\n[/SPEC]"
+        assert get_critique_summary(r) == "critique text"
+
+    def test_task_single_line_value_is_verbatim(self):
+        # len(buf) == 1 path: value taken as-is (no join/strip rewrite)
+        t = extract_tasks(
+            "[TASK]\ntitle: Exact Title\ntype: feature\n"
+            "description: one liner\nacceptance_criteria:\n- a\n[/TASK]"
+        )
+        assert t[0]["title"] == "Exact Title"
+        assert t[0]["description"] == "one liner"
+
+    def test_task_two_line_value_joins_both(self):
+        # len(buf) == 2 crosses the >1 boundary: both lines must survive
+        t = extract_tasks(
+            "[TASK]\ntitle: T\ndescription: first\nsecond\n"
+            "type: chore\n[/TASK]"
+        )
+        assert t[0]["description"] == "first\nsecond"
+
+    def test_task_field_value_offset_exact(self):
+        # rest = line[len(key)+1:]: off-by-one would eat or keep the colon
+        t = extract_tasks("[TASK]\ntitle:NoSpace\ntype: bug\n[/TASK]")
+        assert t[0]["title"] == "NoSpace"
+        t2 = extract_tasks("[TASK]\ntitle:  padded\ntype: bug\n[/TASK]")
+        assert t2[0]["title"] == "padded"

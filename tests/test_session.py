@@ -78,3 +78,57 @@ class TestCheckpoints:
         monkeypatch.setattr(mod, "CHECKPOINTS_DIR", isolated_paths / "ckpt")
         with pytest.raises(ValueError):
             save_checkpoint("s", 1, "../../evil")
+
+
+class TestMutationKillers:
+    """Killers for tools/mutation_check.py survivors (session store)."""
+
+    def test_defaults(self, isolated_paths):
+        from adversarial_spec_amd.session import SessionState
+
+        st = SessionState(session_id="k1", spec="s")
+        assert st.round == 1
+        assert st.preserve_intent is False
+
+    def test_save_twice_is_idempotent(self, isolated_paths):
+        """mkdir(exist_ok=True): the second save hits an existing dir."""
+        from adversarial_spec_amd.session import SessionState
+
+        st = SessionState(session_id="k2", spec="s")
+        st.save()
+        st.round = 2
+        st.save()  # exist_ok=False mutant raises FileExistsError here
+        from adversarial_spec_amd.session import SessionState as S
+
+        assert S.load("k2").round == 2
+
+    def test_checkpoint_twice_same_dir(self, isolated_paths):
+        from adversarial_spec_amd.session import save_checkpoint
+
+        p1 = save_checkpoint("spec v1", 1, None)
+        p2 = save_checkpoint("spec v2", 2, None)
+        assert p1.exists() and p2.exists()
+
+    def test_session_file_indent_format(self, isolated_paths):
+        """Session JSON is the reference's 2-space-indented format."""
+        from adversarial_spec_amd import session as sess
+        from adversarial_spec_amd.session import SessionState
+
+        st = SessionState(session_id="k3", spec="s")
+        st.save()
+        text = (sess.SESSIONS_DIR / "k3.json").read_text()
+        assert '\n  "spec"' in text
+        assert '\n   "spec"' not in text
+
+    def test_save_creates_nested_parents(self, tmp_path, monkeypatch):
+        """mkdir(parents=True): the sessions/checkpoints dirs may live
+        under a not-yet-created config tree."""
+        from adversarial_spec_amd import session as sess
+        from adversarial_spec_amd.session import SessionState, save_checkpoint
+
+        monkeypatch.setattr(sess, "SESSIONS_DIR", tmp_path / "a" / "b" / "sessions")
+        monkeypatch.setattr(sess, "CHECKPOINTS_DIR", tmp_path / "c" / "d" / "ckpt")
+        SessionState(session_id="k9", spec="s").save()
+        assert (sess.SESSIONS_DIR / "k9.json").exists()
+        p = save_checkpoint("spec", 1, None)
+        assert p.exists()
