@@ -424,15 +424,35 @@ class TestFp8:
 
 class TestFp8Engine:
     def test_fp8_generate_close_to_bf16(self):
-        """fp8 opponent produces a similar logits trajectory: compare the
-        first greedy token with the bf16 engine (same seed/name)."""
+        """fp8 vs bf16 end-to-end numerics on the SAME random-init weights:
+        prefill logits stay close (cosine + top-1 agreement) and the greedy
+        decode trajectories match for the first steps (round-1 verdict:
+        the old version only checked that both engines emitted tokens)."""
+        import torch.nn.functional as F
+
         from adversarial_spec_amd.engine.local import LocalEngine
 
         e_bf = LocalEngine({"name": "f8cmp", "arch": "debug-1b"}, device=DEV)
         e_f8 = LocalEngine({"name": "f8cmp", "arch": "debug-1b",
                             "dtype": "fp8"}, device=DEV)
-        a = e_bf.generate("s", "fp8 parity prompt", max_tokens=8,
-                          temperature=0.0, timeout=300)
-        b = e_f8.generate("s", "fp8 parity prompt", max_tokens=8,
-                          temperature=0.0, timeout=300)
-        assert a[2] > 0 and b[2] > 0  # both decoded something
+
+        ids = e_bf.tokenizer.render_chat("s", "fp8 parity prompt")
+        toks = torch.tensor(ids, device=DEV, dtype=torch.long)
+        c_bf = e_bf.model.new_cache(256)
+        c_f8 = e_f8.model.new_cache(256)
+        l_bf = e_bf.model.prefill(toks, c_bf).float()
+        l_f8 = e_f8.model.prefill(toks, c_f8).float()
+
+        cos = F.cosine_similarity(l_bf.unsqueeze(0), l_f8.unsqueeze(0)).item()
+        assert cos > 0.98, f"fp8 prefill logits cosine {cos}"
+        assert int(l_bf.argmax()) == int(l_f8.argmax()), "top-1 diverged"
+
+        # greedy decode trajectory: identical token ids for the first steps
+        tb = tf = int(l_bf.argmax().item())
+        for step in range(6):
+            lb = e_bf.model.decode_one(tb, c_bf).float()
+            lf = e_f8.model.decode_one(tf, c_f8).float()
+            c = F.cosine_similarity(lb.unsqueeze(0), lf.unsqueeze(0)).item()
+            assert c > 0.97, f"step {step}: logits cosine {c}"
+            tb, tf = int(lb.argmax().item()), int(lf.argmax().item())
+            assert tb == tf, f"greedy trajectory diverged at step {step}"
