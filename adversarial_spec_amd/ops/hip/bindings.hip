@@ -56,6 +56,7 @@ void launch_gemm_fp8(const uint8_t*, const float*, const uint8_t*, const float*,
 void launch_gemv_fp8(const uint8_t*, const float*, const uint8_t*, const float*,
                      ushort_t*, int, int, hipStream_t);
 void launch_quant_fp8(const ushort_t*, uint8_t*, float*, int, int, hipStream_t);
+void launch_mfma_rate(const ushort_t*, float*, int, int, hipStream_t);
 }
 
 static hipStream_t cur_stream() {
@@ -454,6 +455,15 @@ torch::Tensor gemm_variant(torch::Tensor a, torch::Tensor b, int64_t which) {
   return c;
 }
 
+// MFMA issue-rate microbench (bf16 vs non-scaled fp8 16x16x32): the
+// config-5 fp8-attention decision evidence (see mfma_rate.hip header).
+void mfma_rate(torch::Tensor seed, torch::Tensor out, int64_t which,
+               int64_t blocks) {
+  CHECK_BF16_CUDA(seed);
+  launch_mfma_rate(uptr(seed), out.data_ptr<float>(), (int)which,
+                   (int)blocks, cur_stream());
+}
+
 // Row-quantize bf16 [M,K] to OCP e4m3 + per-row scale (scale = rowmax/448).
 void quant_fp8(torch::Tensor x, torch::Tensor q, torch::Tensor scale) {
   CHECK_BF16_CUDA(x);
@@ -617,6 +627,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("gemm_fp8", &gemm_fp8, "fp8 e4m3 MFMA GEMM (rowwise scales)");
   m.def("gemv_fp8", &gemv_fp8, "fp8 decode GEMV (rowwise scales)");
   m.def("quant_fp8", &quant_fp8, "rowwise bf16 -> e4m3 quantizer");
+  m.def("mfma_rate", &mfma_rate, "MFMA issue-rate microbench (bf16/fp8)");
   m.def("rope_inplace_ds", &rope_inplace_ds, "graph-mode RoPE (device pos)");
   m.def("rope_kv", &rope_kv, "fused RoPE + paged KV scatter",
         py::arg("q"), py::arg("k"), py::arg("v"), py::arg("cost"),

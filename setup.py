@@ -29,6 +29,7 @@ ext = CUDAExtension(
         os.path.join(HIP_DIR, "gemm256.hip"),
         os.path.join(HIP_DIR, "gemm_fp8.hip"),
         os.path.join(HIP_DIR, "sampling.hip"),
+        os.path.join(HIP_DIR, "mfma_rate.hip"),
     ],
     extra_compile_args={
         "cxx": ["-O3", "-std=c++17"],
