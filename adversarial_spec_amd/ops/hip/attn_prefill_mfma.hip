@@ -34,7 +34,12 @@ typedef __attribute__((__vector_size__(4 * sizeof(float)))) float f32x4v;
 #define HD 128     // head dim (this kernel is hd=128 only)
 #define VPITCH 72  // padded row length (elements) of the Vt and P images
 
-extern "C" __global__ void __launch_bounds__(512, 1)
+// ABL: perf-ablation variants (guide §5 mistake #8 — ablate before
+// optimizing; rule 17 keeps stubbed values live via asm). 0 = full
+// kernel (the production path); 1 = no softmax (raw scores feed PV);
+// 2 = no P LDS bounce (reuse a fixed pfrag); 3 = no PV MFMAs.
+template <int ABL>
+__global__ void __launch_bounds__(512, 1)
 attn_prefill_mfma_kernel(const ushort_t *__restrict__ q,
                          const ushort_t *__restrict__ k,
                          const ushort_t *__restrict__ v,
@@ -158,6 +163,45 @@ attn_prefill_mfma_kernel(const ushort_t *__restrict__ q,
     }
     __builtin_amdgcn_s_setprio(0);
 
+    if (ABL >= 1) {  // skip softmax; keep scores live (rule 17)
+#pragma unroll
+      for (int n = 0; n < 4; ++n)
+        asm volatile("" :: "v"(s[n][0]), "v"(s[n][1]), "v"(s[n][2]),
+                     "v"(s[n][3]));
+      if (ABL < 3) {
+        if (ABL < 2) {
+#pragma unroll
+          for (int n = 0; n < 4; ++n)
+#pragma unroll
+            for (int r = 0; r < 4; ++r)
+              ldsP[(lhi * 4 + r) * VPITCH + n * 16 + lrow] =
+                  f32_to_bf16(s[n][r]);
+        }
+        __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+        for (int half = 0; half < 2; ++half) {
+          const bf16x8v pfrag =
+              *(const bf16x8v *)(ldsP + lrow * VPITCH + half * 32 + lhi * 8);
+#pragma unroll
+          for (int d = 0; d < 8; ++d) {
+            const int vdim = d * 16 + lrow;
+            const int vrot = (((unsigned)vdim >> 3) & 7) * 8;
+            const bf16x8v vfr = *(const bf16x8v *)(
+                ldsVt + (size_t)vdim * VPITCH +
+                ((half * 32 + lhi * 8 + vrot) & 63));
+            o[d] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(pfrag, vfr, o[d],
+                                                           0, 0, 0);
+          }
+        }
+        __builtin_amdgcn_s_setprio(0);
+      } else {
+#pragma unroll
+        for (int d = 0; d < 8; ++d)
+          asm volatile("" :: "v"(o[d][0]), "v"(o[d][3]));
+      }
+      continue;
+    }
+
     // ---- online softmax update (4 q rows per lane: row=lhi*4+r) ----
     float rmax[4];
 #pragma unroll
@@ -252,9 +296,29 @@ extern "C" void launch_attn_prefill_mfma(const ushort_t *q, const ushort_t *k,
     return;
   }
   dim3 grid(hq, (tq + NWAVE * QROWS - 1) / (NWAVE * QROWS));
-  attn_prefill_mfma_kernel<<<grid, NWAVE * WAVE, 0, stream>>>(
+  attn_prefill_mfma_kernel<0><<<grid, NWAVE * WAVE, 0, stream>>>(
       q, k, v, out, tq, tk, kv_offset, scale, hq, kh, 1);
   *ok = 1;
+}
+
+// perf-ablation entry (tools/attn_ablate probe; never on the hot path)
+extern "C" void launch_attn_prefill_mfma_abl(
+    const ushort_t *q, const ushort_t *k, const ushort_t *v, ushort_t *out,
+    int tq, int tk, float scale, int hq, int kh, int abl,
+    hipStream_t stream) {
+  dim3 grid(hq, (tq + NWAVE * QROWS - 1) / (NWAVE * QROWS));
+#define ABL_CASE(N)                                                           \
+  case N:                                                                     \
+    attn_prefill_mfma_kernel<N><<<grid, NWAVE * WAVE, 0, stream>>>(           \
+        q, k, v, out, tq, tk, 0, scale, hq, kh, 1);                           \
+    break;
+  switch (abl) {
+    ABL_CASE(0)
+    ABL_CASE(1)
+    ABL_CASE(2)
+    ABL_CASE(3)
+  }
+#undef ABL_CASE
 }
 
 // ---------------------------------------------------------------------------

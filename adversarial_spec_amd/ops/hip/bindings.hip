@@ -26,6 +26,9 @@ void launch_attn_prefill_simple(const ushort_t*, const ushort_t*, const ushort_t
 void launch_attn_prefill_mfma(const ushort_t*, const ushort_t*, const ushort_t*,
                               ushort_t*, int, int, int, float, int, int, int,
                               hipStream_t, int* ok);
+void launch_attn_prefill_mfma_abl(const ushort_t*, const ushort_t*,
+                                  const ushort_t*, ushort_t*, int, int, float,
+                                  int, int, int, hipStream_t);
 __global__ void mfma_probe_16x16x32(const ushort_t*, const ushort_t*, float*);
 __global__ void rmsnorm_kernel(const ushort_t*, const ushort_t*, ushort_t*, int, float);
 __global__ void add_rmsnorm_kernel(const ushort_t*, const ushort_t*, const ushort_t*,
@@ -294,6 +297,24 @@ torch::Tensor attn_prefill_simple(torch::Tensor q, torch::Tensor k,
   launch_attn_prefill_simple(uptr(qc), uptr(kc), uptr(vc), uptr_mut(out), tq,
                              tk, (int)kv_offset, (float)scale, hq, kh, hd,
                              causal ? 1 : 0, cur_stream());
+  return out;
+}
+
+// prefill-attention ablation launcher (perf diagnosis only)
+torch::Tensor attn_prefill_ablate(torch::Tensor q, torch::Tensor k,
+                                  torch::Tensor v, double scale,
+                                  int64_t abl) {
+  CHECK_BF16_CUDA(q);
+  auto qc = q.contiguous();
+  auto kc = k.contiguous();
+  auto vc = v.contiguous();
+  const int tq = qc.size(0), hq = qc.size(1), hd = qc.size(2);
+  const int tk = kc.size(0), kh = kc.size(1);
+  TORCH_CHECK(hd == 128, "ablation probe is hd=128 only");
+  auto out = torch::empty_like(qc);
+  launch_attn_prefill_mfma_abl(uptr(qc), uptr(kc), uptr(vc), uptr_mut(out),
+                               tq, tk, (float)scale, hq, kh, (int)abl,
+                               cur_stream());
   return out;
 }
 
@@ -613,6 +634,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("kv_write", &kv_write, "paged KV scatter");
   m.def("attn_prefill", &attn_prefill, "causal prefill attention");
   m.def("attn_prefill_simple", &attn_prefill_simple, "non-MFMA prefill (anchor)");
+  m.def("attn_prefill_ablate", &attn_prefill_ablate,
+        "prefill attention perf-ablation variants");
   m.def("mfma_probe16", &mfma_probe16, "MFMA 16x16x32 fragment-map probe");
   m.def("attn_decode_paged", &attn_decode_paged, "paged decode attention",
         py::arg("q"), py::arg("kc"), py::arg("vc"), py::arg("page_table"),
