@@ -82,10 +82,13 @@ attn_prefill_mfma_kernel(const ushort_t *__restrict__ q,
   }
 
   // ---- state ----
-  float m[4], lsum[4];
-  f32x4v o[8];  // O accumulators: 8 dim-chunks of 16
-#pragma unroll
-  for (int r = 0; r < 4; ++r) { m[r] = -INFINITY; lsum[r] = 0.f; }
+  // SWAPPED-QK^T layout (guide T12 idea): S^T = mfma(K, Q) puts the
+  // whole softmax row of THIS LANE's q-row (= lrow) in registers, so the
+  // online-softmax state is one scalar pair per lane and the row reduce
+  // is an in-lane tree + 2 shfls (the row-major form needed 8 chains of
+  // 4 shfls per tile — the ablation showed softmax at 43% of the kernel).
+  float m1 = -INFINITY, l1 = 0.f;  // row = lrow
+  f32x4v o[8];  // O accumulators: 8 dim-chunks of 16 (rows lhi*4+r)
 #pragma unroll
   for (int d = 0; d < 8; ++d) o[d] = (f32x4v){0.f, 0.f, 0.f, 0.f};
 
@@ -145,7 +148,9 @@ attn_prefill_mfma_kernel(const ushort_t *__restrict__ q,
     __syncthreads();
     if (kt + KVBLK < kmax) stage_load(kt + KVBLK);  // hide under compute
 
-    // ---- S = Q K^T for four 16-key subtiles ----
+    // ---- S^T = K Q^T for four 16-key subtiles (swapped operands: the
+    // 16x16x32 A and B fragment layouts coincide, so the SAME fragments
+    // serve; C becomes S^T[key][q-row] with the row lane-local) ----
     f32x4v s[4];
     __builtin_amdgcn_s_setprio(1);
 #pragma unroll
@@ -154,11 +159,11 @@ attn_prefill_mfma_kernel(const ushort_t *__restrict__ q,
       const int keyr = n * 16 + lrow;
 #pragma unroll
       for (int c = 0; c < 4; ++c) {
-        // B frag: B[kd][col] = K[keyr][c*32+kd8] from the swizzled image
+        // A frag: A[key][kd] = K[keyr][c*32+kd8] from the swizzled image
         unsigned kbyte = (unsigned)keyr * 256u + (unsigned)(c * 4 + lhi) * 16u;
         kbyte ^= ((unsigned)(keyr & 7)) << 4;
-        const bf16x8v bfr = *(const bf16x8v *)((const char *)ldsK + kbyte);
-        s[n] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(qfrag[c], bfr, s[n], 0, 0, 0);
+        const bf16x8v afr = *(const bf16x8v *)((const char *)ldsK + kbyte);
+        s[n] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(afr, qfrag[c], s[n], 0, 0, 0);
       }
     }
     __builtin_amdgcn_s_setprio(0);
@@ -202,53 +207,73 @@ attn_prefill_mfma_kernel(const ushort_t *__restrict__ q,
       continue;
     }
 
-    // ---- online softmax update (4 q rows per lane: row=lhi*4+r) ----
-    float rmax[4];
+    // ---- online softmax, lane-local row (row = lrow) ----
+    // Interior tiles of a causal 8k+ walk are FULLY unmasked for the
+    // whole wave (uniform predicate): skip the 32 mask compares there.
+    const bool full_tile =
+        (kt + KVBLK <= tk) &&
+        (!causal || kt + KVBLK <= kv_offset + q0 + 1);
+    if (full_tile) {
 #pragma unroll
-    for (int r = 0; r < 4; ++r) {
-      const int qrow_abs = kv_offset + q0 + lhi * 4 + r;
+      for (int n = 0; n < 4; ++n)
 #pragma unroll
-      for (int n = 0; n < 4; ++n) {
-        const int key = kt + n * 16 + lrow;
-        float sv = s[n][r] * scale;
-        if (key >= tk || (causal && key > qrow_abs)) sv = -INFINITY;
-        s[n][r] = sv;
-      }
-      float mx = fmaxf(fmaxf(s[0][r], s[1][r]), fmaxf(s[2][r], s[3][r]));
-#pragma unroll
-      for (int off = 8; off > 0; off >>= 1)
-        mx = fmaxf(mx, __shfl_xor(mx, off, WAVE));
-      rmax[r] = mx;
-    }
-
-#pragma unroll
-    for (int r = 0; r < 4; ++r) {
-      const float mn = fmaxf(m[r], rmax[r]);
-      const float alpha =
-          (m[r] == -INFINITY && mn == -INFINITY) ? 0.f : __expf(m[r] - mn);
-      float psum = 0.f;
+        for (int r = 0; r < 4; ++r) s[n][r] *= scale;
+    } else {
+      const int qrow_abs = kv_offset + q0 + lrow;
 #pragma unroll
       for (int n = 0; n < 4; ++n) {
-        const float p = (s[n][r] == -INFINITY) ? 0.f : __expf(s[n][r] - mn);
-        s[n][r] = p;
-        psum += p;
+#pragma unroll
+        for (int r = 0; r < 4; ++r) {
+          const int key = kt + n * 16 + lhi * 4 + r;
+          float sv = s[n][r] * scale;
+          if (key >= tk || (causal && key > qrow_abs)) sv = -INFINITY;
+          s[n][r] = sv;
+        }
       }
-#pragma unroll
-      for (int off = 8; off > 0; off >>= 1)
-        psum += __shfl_xor(psum, off, WAVE);
-      lsum[r] = lsum[r] * alpha + psum;
-      m[r] = mn;
-#pragma unroll
-      for (int d = 0; d < 8; ++d) o[d][r] *= alpha;
     }
-
-    // ---- P -> LDS (C layout -> A layout bounce; padded pitch) ----
+    // in-lane 16-value max tree + 2 cross-lane levels (lanes lrow,
+    // lrow+16, lrow+32, lrow+48 hold the rest of this row's keys)
+    float mx = fmaxf(fmaxf(s[0][0], s[0][1]), fmaxf(s[0][2], s[0][3]));
+#pragma unroll
+    for (int n = 1; n < 4; ++n)
+      mx = fmaxf(mx, fmaxf(fmaxf(s[n][0], s[n][1]),
+                           fmaxf(s[n][2], s[n][3])));
+    mx = fmaxf(mx, __shfl_xor(mx, 16, WAVE));
+    mx = fmaxf(mx, __shfl_xor(mx, 32, WAVE));
+    const float mn = fmaxf(m1, mx);
+    const float alpha = (mn == -INFINITY) ? 0.f : __expf(m1 - mn);
+    float psum = 0.f;
 #pragma unroll
     for (int n = 0; n < 4; ++n) {
 #pragma unroll
       for (int r = 0; r < 4; ++r) {
-        ldsP[(lhi * 4 + r) * VPITCH + n * 16 + lrow] = f32_to_bf16(s[n][r]);
+        const float p =
+            (s[n][r] == -INFINITY) ? 0.f : __expf(s[n][r] - mn);
+        s[n][r] = p;
+        psum += p;
       }
+    }
+    psum += __shfl_xor(psum, 16, WAVE);
+    psum += __shfl_xor(psum, 32, WAVE);
+    l1 = l1 * alpha + psum;
+    m1 = mn;
+    // O rows are lhi*4+r: fetch those rows' alphas (any lane with
+    // lrow == row has the value; pick the same lhi quartile)
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+      const float oa = __shfl(alpha, (lhi * 4 + r) | (lhi << 4), WAVE);
+#pragma unroll
+      for (int d = 0; d < 8; ++d) o[d][r] *= oa;
+    }
+
+    // ---- P -> LDS: lane-local row, 4 keys packed per 8-B store ----
+#pragma unroll
+    for (int n = 0; n < 4; ++n) {
+      union { uint32_t u32[2]; ushort_t u16[4]; } pk;
+#pragma unroll
+      for (int r = 0; r < 4; ++r) pk.u16[r] = f32_to_bf16(s[n][r]);
+      *(uint2 *)(ldsP + (size_t)lrow * VPITCH + n * 16 + lhi * 4) =
+          make_uint2(pk.u32[0], pk.u32[1]);
     }
 
     // ---- O += P V (two 32-key halves) ----
@@ -272,12 +297,13 @@ attn_prefill_mfma_kernel(const ushort_t *__restrict__ q,
     __builtin_amdgcn_s_setprio(0);
   }
 
-  // ---- epilogue: O / l ----
+  // ---- epilogue: O / l (l lives on the lane whose lrow == row) ----
 #pragma unroll
   for (int r = 0; r < 4; ++r) {
     const int qrow = q0 + lhi * 4 + r;
     if (qrow >= tq) continue;
-    const float inv = (lsum[r] > 0.f) ? 1.0f / lsum[r] : 0.f;
+    const float lr = __shfl(l1, (lhi * 4 + r) | (lhi << 4), WAVE);
+    const float inv = (lr > 0.f) ? 1.0f / lr : 0.f;
     ushort_t *orow = out + ((size_t)qrow * hq + h) * HD;
 #pragma unroll
     for (int d = 0; d < 8; ++d) {
