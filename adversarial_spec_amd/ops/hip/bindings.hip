@@ -56,6 +56,8 @@ void launch_gemm256(const ushort_t*, const ushort_t*, ushort_t*, int, int, int,
                     hipStream_t);
 void launch_gemm_fp8(const uint8_t*, const float*, const uint8_t*, const float*,
                      ushort_t*, int, int, int, hipStream_t);
+void launch_gemm_fp8_256(const uint8_t*, const float*, const uint8_t*,
+                         const float*, ushort_t*, int, int, int, hipStream_t);
 void launch_gemv_fp8(const uint8_t*, const float*, const uint8_t*, const float*,
                      ushort_t*, int, int, hipStream_t);
 void launch_quant_fp8(const ushort_t*, uint8_t*, float*, int, int, hipStream_t);
@@ -511,9 +513,17 @@ torch::Tensor gemm_fp8(torch::Tensor x, torch::Tensor w8, torch::Tensor wsc) {
   launch_quant_fp8(uptr(xc), x8.data_ptr<uint8_t>(), xs.data_ptr<float>(),
                    M, K, cur_stream());
   auto c = torch::empty({M, N}, xc.options());
-  launch_gemm_fp8(x8.data_ptr<uint8_t>(), xs.data_ptr<float>(),
-                  w8.data_ptr<uint8_t>(), wsc.data_ptr<float>(), uptr_mut(c),
-                  M, N, K, cur_stream());
+  // 8-phase 256^2 kernel for prefill-sized shapes (same dispatch rule as
+  // the bf16 gemm); 128^2 covers edges and small shapes.
+  if (M > 128 && N >= 256 && K >= 512 && (K % 128) == 0) {
+    launch_gemm_fp8_256(x8.data_ptr<uint8_t>(), xs.data_ptr<float>(),
+                        w8.data_ptr<uint8_t>(), wsc.data_ptr<float>(),
+                        uptr_mut(c), M, N, K, cur_stream());
+  } else {
+    launch_gemm_fp8(x8.data_ptr<uint8_t>(), xs.data_ptr<float>(),
+                    w8.data_ptr<uint8_t>(), wsc.data_ptr<float>(),
+                    uptr_mut(c), M, N, K, cur_stream());
+  }
   return c;
 }
 

@@ -28,6 +28,7 @@ ext = CUDAExtension(
         os.path.join(HIP_DIR, "gemm.hip"),
         os.path.join(HIP_DIR, "gemm256.hip"),
         os.path.join(HIP_DIR, "gemm_fp8.hip"),
+        os.path.join(HIP_DIR, "gemm256_fp8.hip"),
         os.path.join(HIP_DIR, "sampling.hip"),
         os.path.join(HIP_DIR, "mfma_rate.hip"),
     ],

@@ -369,6 +369,20 @@ class TestGemm256:
                 again = hip.gemm_variant(a, b, 256)
                 assert torch.equal(again, first), "gemm256 nondeterminism"
 
+    def test_fp8_256_vs_dequant_ref_and_race(self):
+        """fp8 8-phase variant (gemm256_fp8.hip): CPU dequant parity at a
+        256-dispatch shape + bitwise rerun stability."""
+        m, n, k = 512, 768, 640
+        x = _bf(torch.randn(m, k)).to(DEV)
+        w = _bf(torch.randn(n, k) * 0.05).to(DEV)
+        wq, wsc = ops.quantize_fp8_rowwise(w)
+        got = ops.gemm_fp8(x, wq.to(DEV), wsc.to(DEV))
+        want = ops.gemm_fp8(x.cpu(), wq, wsc)
+        _assert_close(got, want.float(), atol=8e-2, name="fp8-256 dequant")
+        for _ in range(8):
+            again = ops.gemm_fp8(x, wq.to(DEV), wsc.to(DEV))
+            assert torch.equal(again, got), "fp8-256 nondeterminism"
+
 
 class TestGemv:
     @pytest.mark.parametrize("n,k", [(1024, 512), (6144, 4096), (1000, 264)])
