@@ -10,9 +10,15 @@
 // glds piece per wave per half-tile, so the steady-state wait is
 // vmcnt(2) (two half-tiles in flight).
 //
-// LDS swizzle: 16-B chunks XOR'd with (row & 3) within the 64-B row
-// (4 chunks/row — the same involution the 128^2 fp8 kernel verified);
-// fragments are 8-B reads of the low/high half of a swizzled chunk.
+// LDS swizzle: 16-B chunks XOR'd with ((row >> 2) & 3) within the 64-B
+// row. NOT (row & 3): 64-B rows repeat a 256-B bank row every 4 rows, so
+// rows {r, r+4, r+8, r+12} of a 16-lane read group alias the same banks
+// at the same intra-row chunk — the low-bit XOR (which the 2-barrier
+// 128^2 fp8 kernel gets away with, its stage stall hiding LDS reads)
+// leaves that 4-way conflict in place, and in the 8-phase schedule the
+// LDS read IS the critical path (guide §5.5 T2 regime gate). XORing with
+// row bits 2-3 gives the four aliasing rows four different chunks.
+// Fragments are 8-B reads of the low/high half of a swizzled chunk.
 
 #include "common.h"
 
@@ -49,7 +55,7 @@ DEVINL void q8_tile(const uint8_t *__restrict__ imA,
   // fragment read: 8 fp8 at k-slice ks*32 + lhi*8 of the 64-B row,
   // chunk-swizzled (chunk ^ (row & 3)), low/high half by lhi parity
   auto frag = [&](const uint8_t *im, int row, int ks) -> long {
-    const int ch = ((ks * 2) + (lhi >> 1)) ^ (row & 3);
+    const int ch = ((ks * 2) + (lhi >> 1)) ^ ((row >> 2) & 3);
     return *(const long *)(im + (size_t)row * Q8_BK + ch * 16 +
                            (lhi & 1) * 8);
   };
@@ -179,7 +185,7 @@ gemm_fp8_256_kernel(const uint8_t *__restrict__ a,
     const int rl = wid * 16 + (lane >> 2);
     const int grow = min(row0 + rl, rmax);
     const int c = lane & 3;
-    sbase[h][0] = src + (size_t)grow * K + (size_t)(c ^ (rl & 3)) * 16;
+    sbase[h][0] = src + (size_t)grow * K + (size_t)(c ^ ((rl >> 2) & 3)) * 16;
     sbase[h][1] = sbase[h][0];  // single piece per wave (layout parity)
   }
 
