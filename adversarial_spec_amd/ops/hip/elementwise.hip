@@ -16,7 +16,7 @@
 // d must be a multiple of 8. Each thread covers d/256/8 vectors (or strided).
 // ---------------------------------------------------------------------------
 
-extern "C" __global__ void __launch_bounds__(256)
+extern "C" __global__ void __launch_bounds__(1024)
 rmsnorm_kernel(const ushort_t *__restrict__ x, const ushort_t *__restrict__ w,
                ushort_t *__restrict__ y, int d, float eps) {
   __shared__ float scratch[16];
@@ -50,7 +50,7 @@ rmsnorm_kernel(const ushort_t *__restrict__ x, const ushort_t *__restrict__ w,
 //   y  = rmsnorm(r') * w
 // ---------------------------------------------------------------------------
 
-extern "C" __global__ void __launch_bounds__(256)
+extern "C" __global__ void __launch_bounds__(1024)
 add_rmsnorm_kernel(const ushort_t *__restrict__ resid,
                    const ushort_t *__restrict__ delta,
                    const ushort_t *__restrict__ w,
