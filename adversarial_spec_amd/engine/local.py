@@ -207,14 +207,18 @@ class LocalEngine:
 
         out_ids: list[int] = []
         with timer.phase("decode"):
-            # TP decode uses the host-stepped loop (decode_one, covered by
-            # the gloo TP=2 parity tests): capturing RCCL collectives in
-            # HIP graphs is not supported reliably on this stack, and the
-            # per-layer all-reduce already bounds decode latency there.
+            # TP decode runs the SAME allocation-free workspace loop as
+            # single-GPU decode, just ungraphed: RCCL collectives cannot be
+            # captured in HIP graphs on this stack, and per-segment graphs
+            # between the all-reduces would not pay — the measured graph
+            # replay floor (~10-16 us host per replay) matches the eager
+            # launch cost of the ~5 kernels a segment holds. Ranks sample
+            # identically (deterministic kernel + identical all-reduced
+            # activations), and the stop decision is rank-coordinated in
+            # _decode_graphed so no rank leaves a collective early.
             use_async = (
                 self.device.type == "cuda"
                 and top_p >= 1.0
-                and self.model.tp is None
                 and ops.hip_available()
             )
             if use_async:
@@ -222,6 +226,7 @@ class LocalEngine:
                 # for longer generations (ADVSPEC_NO_GRAPH forces eager —
                 # same kernels, bitwise-identical tokens)
                 use_graph = (max_new >= 8
+                             and self.model.tp is None
                              and not os.environ.get("ADVSPEC_NO_GRAPH"))
                 out_ids = self._decode_graphed(
                     logits, cache, max_new, temperature, stop_ids, deadline,
@@ -416,6 +421,7 @@ class LocalEngine:
                     return True
             return False
 
+        tp = self.model.tp
         while i < max_new and not done:
             n = min(CHECK, max_new - i)
             if graph is not None:
@@ -432,6 +438,13 @@ class LocalEngine:
             done = scan_until(min(i, max_new))
             if time.monotonic() > deadline:
                 done = True
+            if tp is not None and tp.size > 1:
+                # rank-coordinate the stop decision: token content is
+                # identical across ranks (deterministic sampling on
+                # identical all-reduced activations) but the wall-clock
+                # deadline is not — an uncoordinated early exit would
+                # leave the other ranks blocked in the next all-reduce
+                done = tp.any_flag(done)
         if not done:
             scan_until(min(i, max_new))
 

@@ -118,6 +118,19 @@ class TPContext:
             dist.all_reduce(x, group=self.group)
         return x
 
+    def any_flag(self, flag: bool) -> bool:
+        """Logical OR of a host flag across the TP group (MAX reduce on a
+        1-element tensor): coordinates decode stop decisions so no rank
+        exits a collective loop early. CPU tensor on gloo, device tensor
+        on nccl."""
+        if self.size == 1:
+            return flag
+        backend = dist.get_backend(self.group)
+        dev = "cpu" if backend == "gloo" else "cuda"
+        t = torch.tensor([1 if flag else 0], dtype=torch.int32, device=dev)
+        dist.all_reduce(t, op=dist.ReduceOp.MAX, group=self.group)
+        return bool(t.item())
+
     @classmethod
     def from_default_group(cls) -> "TPContext":
         return cls(dist.get_world_size(), dist.get_rank(), None)

@@ -82,7 +82,7 @@ def run_tp(rank: int, world: int) -> None:
     scale = l_full.abs().max().item() + 1e-6
     assert err / scale < 5e-2, f"TP parity: max err {err} vs scale {scale}"
 
-    # decode parity for a few tokens (host-stepped TP decode loop)
+    # decode parity for a few tokens (eager TP decode)
     tok_tp = tok_full = int(l_full.argmax().item())
     for _ in range(4):
         lt = m_tp.decode_one(tok_tp, c_tp).float()
@@ -92,6 +92,35 @@ def run_tp(rank: int, world: int) -> None:
         tok_tp = int(lt.argmax().item())
         tok_full = int(lf.argmax().item())
         assert tok_tp == tok_full
+
+    # engine generate() over the allocation-free TP workspace loop: every
+    # rank must emit the SAME token ids (deterministic sampling on
+    # identical all-reduced activations + rank-coordinated stop)
+    from adversarial_spec_amd.engine.local import LocalEngine
+    from adversarial_spec_amd.models.config import PRESETS
+
+    PRESETS["tp-gpu-test"] = cfg  # registry entry for the engine
+    tp2 = TPContext(world, rank, None)
+    eng = LocalEngine({"name": "tp-gen", "arch": "tp-gpu-test"},
+                      device=dev, tp=tp2)
+    text, _in, out_tok, _tm = eng.generate("s", "tp generate parity",
+                                           max_tokens=12, temperature=0.7,
+                                           timeout=300)
+    ids = eng.tokenizer.encode(text)[:32]
+    buf = torch.full((32,), -1, dtype=torch.int64, device="cpu")
+    buf[: len(ids)] = torch.tensor(ids, dtype=torch.int64)
+    gathered = [torch.empty_like(buf) for _ in range(world)]
+    # gather on gloo-compatible CPU tensors regardless of backend
+    if dist.get_backend() == "nccl":
+        bufc = buf.cuda()
+        gath = [torch.empty_like(bufc) for _ in range(world)]
+        dist.all_gather(gath, bufc)
+        gathered = [g.cpu() for g in gath]
+    else:
+        dist.all_gather(gathered, buf)
+    for g in gathered:
+        assert torch.equal(g, gathered[0]), "TP ranks emitted different tokens"
+    assert out_tok > 0
 
     if rank == 0:
         print(f"RCCL_TP_OK backend={dist.get_backend()} world={world}")
