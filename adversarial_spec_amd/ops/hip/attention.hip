@@ -35,13 +35,20 @@ DEVINL float group_reduce_sum(float v) {
 // grid.x = kh, grid.y = n_splits, block = 256 (4 waves).
 //
 // The KV read is staged through LDS by async global_load_lds in 64-position
-// double-buffered tiles: register-ring prefetch measured latency-bound
-// (Little's law: ~1.7 KB in flight per CU vs ~9.4 KB needed at ~900-cycle
-// HBM latency; PMC: 58% wave-parked), while LDS-DMA keeps a whole tile
-// (8 glds x 1 KiB per wave) in flight behind a counted vmcnt, like the
-// prefill kernel's staging. Compute then walks the tile from LDS
-// (ds_read_b128, conflict-free: each 16-lane group reads one full 256-B
-// row) with the positions of a tile split across the block's 4 waves.
+// double-buffered tiles (LDS-DMA keeps a whole tile in flight behind a
+// counted vmcnt — register-ring prefetch measured latency-bound).
+//
+// Compute is GROUP-PER-WAVE, LANE-PER-POSITION: wave w owns GQA head
+// group w (and w+4 at group 8); in the QK pass lane p holds position
+// tbase+p's full dot (the K image is chunk-XOR-swizzled so 16-lane b128
+// groups reading the same chunk index of 16 different rows stay
+// conflict-free), so the online-softmax reduce is ONE 6-level wave
+// reduction per 64-position tile — the old dim-sliced layout paid 4
+// shfl-chain levels per position per group (64+ chained DS ops per
+// tile). P values reach the dim-sliced PV pass by shfl broadcast: no
+// ds_write anywhere in the kernel (an LDS store makes hipcc order every
+// stage-image ds_read behind vmcnt(0), draining the DMA pipeline —
+// .s-verified round-2 failure mode).
 //
 // Workspace (fp32): ws_m, ws_l: [kh, n_splits, group]
 //                   ws_acc:     [kh, n_splits, group, hd]
@@ -66,25 +73,30 @@ __global__ void __launch_bounds__(256) attn_decode_split_kernel(
   const int n_splits = gridDim.y;
   const int lane = threadIdx.x & (WAVE - 1);
   const int wid = threadIdx.x / WAVE;
-  const int sub = lane / LPP;        // position sub-group within wave
-  const int sl = lane % LPP;         // dim slice within position
-  const int subs = WAVE / LPP;       // concurrent positions per wave
+  const int sub = lane / LPP;        // PV pass: position sub-group
+  const int sl = lane % LPP;         // PV pass: dim slice within position
+  const int subs = WAVE / LPP;       // concurrent positions per PV step
+  constexpr int NG = (MG + 3) / 4;   // GQA heads owned per wave
 
   const int start = split * split_len;
   const int limit = min(seq_len, (split + 1) * split_len);
 
-  // ONE shared array (guide §5 trap 4a); the stage buffers are reused as
-  // the wave-merge scratch after the tile loop (barrier-separated).
-  // CRITICAL (.s-verified): the function must contain NO ds_write and no
-  // LDS-region type-punning — either makes hipcc order every ds_read of
-  // the stage image behind `s_waitcnt vmcnt(0)`, draining the whole DMA
-  // pipeline once per position (the 16 us/dispatch plateau of rounds 1-2).
+  // ONE shared array (guide §5 trap 4a): [2 stage buffers][K | V] + the
+  // q image. K images are chunk-swizzled; V and q are linear.
   constexpr int TILE_E = DTILE * (LPP * 8);          // elements per tile
-  __shared__ __attribute__((aligned(16))) ushort_t lds[4 * TILE_E];
+  constexpr int QIMG_E = MG * LPP * 8;
+  __shared__ __attribute__((aligned(16))) ushort_t lds[4 * TILE_E + QIMG_E];
+  ushort_t *qlds = lds + 4 * TILE_E;
 
-  float m[MG], l[MG], acc[MG][8];
+  // K chunk swizzle: a 256-B LDS bank row holds 32/LPP image rows, so
+  // same-chunk reads of rows p and p' collide when p ≡ p' (mod 16/LPP
+  // ... spread by XORing the chunk with the row's bank-alias class.
+  constexpr int SWZ_DIV = 16 / LPP;  // rows sharing a bank-row offset
+  auto kswz = [&](int row) { return (row / SWZ_DIV) & (LPP - 1); };
+
+  float m[NG], l[NG], acc[NG][8];
 #pragma unroll
-  for (int gi = 0; gi < MG; ++gi) {
+  for (int gi = 0; gi < NG; ++gi) {
     m[gi] = -INFINITY;
     l[gi] = 0.f;
 #pragma unroll
@@ -92,10 +104,11 @@ __global__ void __launch_bounds__(256) attn_decode_split_kernel(
   }
 
   // stage one DTILE-position K/V tile into LDS buffer b via glds.
-  // Each wave issues LPP/4 x 1 KiB pieces for K and for V (4 waves x
-  // LPP/4 x 64/LPP rows = 64 = DTILE); a piece covers 1024/(hd*2) = subs
-  // consecutive positions; lane l handles the piece's position l/LPP,
-  // chunk l%LPP — lane-linear in LDS as glds requires. Out-of-range
+  // Each wave issues LPP/4 x 1 KiB pieces for K and for V; a piece covers
+  // `subs` consecutive positions; lane l handles the piece's position
+  // l/LPP, chunk l%LPP — lane-linear in LDS as glds requires. The K
+  // SOURCE chunk is pre-swizzled (kswz involution) so the lane-linear
+  // LDS image realizes the swizzled layout; V stays linear. Out-of-range
   // positions clamp to a valid row (their scores are masked in compute).
   auto stage_tile = [&](int t0, int b) {
     ushort_t *kimg = lds + (size_t)b * 2 * TILE_E;
@@ -103,7 +116,8 @@ __global__ void __launch_bounds__(256) attn_decode_split_kernel(
 #pragma unroll
     for (int i = 0; i < LPP / 4; ++i) {  // pieces per wave per tensor
       const int prow = (wid * (LPP / 4) + i) * subs;  // tile-local row
-      const int pp = min(t0 + prow + sub, seq_len - 1);
+      const int myrow = prow + sub;
+      const int pp = min(t0 + myrow, seq_len - 1);
       // IDENT: the engine's single-pool cache uses the identity page
       // table — pure address math, no table read. The general path pays
       // an ordinary VMEM load here (hipcc then drains the DMA queue at
@@ -112,7 +126,8 @@ __global__ void __launch_bounds__(256) attn_decode_split_kernel(
       const size_t row = ((size_t)phys * page + (pp % page)) * kh * hd +
                          (size_t)g * hd;
       __builtin_amdgcn_global_load_lds(
-          (const __attribute__((address_space(1))) void *)(kc + row + sl * 8),
+          (const __attribute__((address_space(1))) void
+               *)(kc + row + (size_t)(sl ^ kswz(myrow)) * 8),
           (__attribute__((address_space(3))) void *)(kimg +
                                                      (size_t)prow * hd),
           16, 0, 0);
@@ -124,23 +139,28 @@ __global__ void __launch_bounds__(256) attn_decode_split_kernel(
     }
   };
 
-  // q fragments by plain loads BEFORE any glds is issued. The asm "+v"
-  // use right after forces hipcc's counted wait HERE, while the VMEM
-  // queue holds only the q loads: deferred to the first dot inside the
-  // position loop, the wait count is loop-variant (DMA in flight), so
-  // hipcc emits `s_waitcnt vmcnt(0)` INSIDE the loop and drains the DMA
-  // pipeline once per position (.s-verified failure mode).
+  // stage q (group*hd elements) by glds from wave 0 — staged, not plain
+  // loads: an ordinary VMEM load whose wait lands inside the tile loop
+  // forces hipcc to a loop-variant count, i.e. `vmcnt(0)` per iteration,
+  // draining the DMA pipeline (.s-verified failure mode). The q pieces
+  // are issued BEFORE the K/V prologue, so wave 0's first counted wait
+  // covers them and the barrier publishes the image.
   typedef __attribute__((__vector_size__(8 * sizeof(short)))) short v8s;
-  v8s qraw[MG];
+  if (wid == 0) {
 #pragma unroll
-  for (int gi = 0; gi < MG; ++gi)
-    if (gi < group)
-      qraw[gi] = ((const v8s *)(q + ((size_t)(g * group + gi)) * hd))[sl];
-#pragma unroll
-  for (int gi = 0; gi < MG; ++gi)
-    asm volatile("" : "+v"(qraw[gi]));
+    for (int i = 0; i < (QIMG_E * 2 + 1023) / 1024; ++i) {
+      if (i * 512 >= group * hd) break;  // uniform: MG may exceed group
+      const int off8 = min(i * 512 + lane * 8, group * hd - 8);
+      __builtin_amdgcn_global_load_lds(
+          (const __attribute__((address_space(1))) void
+               *)(q + (size_t)(g * group) * hd + off8),
+          (__attribute__((address_space(3))) void *)(qlds + i * 512), 16, 0,
+          0);
+    }
+  }
 
   const int ntiles = (limit - start + DTILE - 1) / DTILE;
+  float pex[NG];
   if (ntiles > 0) {
     stage_tile(start, 0);
     if (ntiles > 1) stage_tile(start + DTILE, 1);
@@ -148,6 +168,8 @@ __global__ void __launch_bounds__(256) attn_decode_split_kernel(
     for (int t = 0; t < ntiles; ++t) {
       // wait for tile t's DMA (tile t+1 stays in flight: 2 tensors x
       // LPP/4 pieces per wave = LPP/2 loads outstanding), then align.
+      // Wave 0's q pieces are OLDER than tile 0's, so the same counted
+      // wait covers them.
       if (ntiles > t + 1) {
         if (LPP == 16) asm volatile("s_waitcnt vmcnt(8)" ::: "memory");
         else if (LPP == 8) asm volatile("s_waitcnt vmcnt(4)" ::: "memory");
@@ -159,111 +181,101 @@ __global__ void __launch_bounds__(256) attn_decode_split_kernel(
       const ushort_t *kimg = lds + (size_t)(t & 1) * 2 * TILE_E;
       const ushort_t *vimg = kimg + TILE_E;
       const int tbase = start + t * DTILE;
-      // this wave's positions within the tile: wid*DTILE/4 .. +DTILE/4
-      const int wbeg = wid * (DTILE / 4);
-      const int wend = min(wbeg + DTILE / 4, limit - tbase);
-      for (int r = wbeg + sub; r < wend; r += subs) {
-        const v8s kv_ = ((const v8s *)(kimg + (size_t)r * hd))[sl];
-        const v8s vv_ = ((const v8s *)(vimg + (size_t)r * hd))[sl];
-        const bf16x8 kraw = __builtin_bit_cast(bf16x8, kv_);
-        const bf16x8 vraw = __builtin_bit_cast(bf16x8, vv_);
-        const f32x8 vd = unpack8(vraw);
+
+      // ---- QK: lane-per-position full dot (4 chunk-residue
+      // accumulators cover the dependent v_dot2 latency) ----
+      const bool valid = tbase + lane < limit;
+      const int myswz = kswz(lane);
 #pragma unroll
-        for (int gi = 0; gi < MG; ++gi) {
-          if (gi >= group) break;
-          const float dot =
-              dot8_bf16(__builtin_bit_cast(bf16x8, qraw[gi]), kraw, 0.f);
-          const float s = group_reduce_sum<LPP>(dot) * scale;
-          const float m_new = fmaxf(m[gi], s);
-          const float alpha = __expf(m[gi] - m_new);
-          const float pex = __expf(s - m_new);
-          l[gi] = l[gi] * alpha + pex;
+      for (int gi = 0; gi < NG; ++gi) {
+        const int gq = wid + 4 * gi;
+        if (gq >= group) break;  // wave-uniform
+        float a4[4] = {0.f, 0.f, 0.f, 0.f};
+#pragma unroll
+        for (int c = 0; c < LPP; ++c) {
+          const v8s kv = ((const v8s *)(kimg + (size_t)lane * hd))[c];
+          const v8s qv =
+              ((const v8s *)(qlds + (size_t)gq * hd))[c ^ myswz];
+          a4[c & 3] = dot8_bf16(__builtin_bit_cast(bf16x8, qv),
+                                __builtin_bit_cast(bf16x8, kv), a4[c & 3]);
+        }
+        const float s =
+            ((a4[0] + a4[1]) + (a4[2] + a4[3])) * scale;
+
+        // ---- online softmax for this wave's head: ONE wave-wide
+        // max/sum pair per tile ----
+        float mx = valid ? s : -INFINITY;
+        mx = wave_reduce_max(mx);
+        const float mn = fmaxf(m[gi], mx);
+        const float alpha = (mn == -INFINITY) ? 0.f : __expf(m[gi] - mn);
+        const float p = (valid && mn != -INFINITY) ? __expf(s - mn) : 0.f;
+        const float ps = wave_reduce_sum(p);
+        l[gi] = l[gi] * alpha + ps;
+        m[gi] = mn;
+        pex[gi] = p;
+#pragma unroll
+        for (int j = 0; j < 8; ++j) acc[gi][j] *= alpha;
+      }
+
+      // ---- PV: dim-sliced; this tile's P values arrive by shfl from
+      // the lane that owns the position (no LDS write — see header) ----
+#pragma unroll 4
+      for (int it = 0; it < LPP; ++it) {
+        const int p = it * subs + sub;
+        const v8s vv = ((const v8s *)(vimg + (size_t)p * hd))[sl];
+        const f32x8 vd = unpack8(__builtin_bit_cast(bf16x8, vv));
+#pragma unroll
+        for (int gi = 0; gi < NG; ++gi) {
+          if (wid + 4 * gi >= group) break;
+          const float pw = __shfl(pex[gi], p, WAVE);
 #pragma unroll
           for (int j = 0; j < 8; ++j)
-            acc[gi][j] = acc[gi][j] * alpha + pex * vd.v[j];
-          m[gi] = m_new;
+            acc[gi][j] = fmaf(pw, vd.v[j], acc[gi][j]);
         }
       }
-      // issue tile t+2 into the buffer just consumed (all waves have
-      // finished their ds_reads of it: they run under the same wave's
-      // program order... other waves may still be reading buffer t&1, so
-      // re-align first, then issue.
+
+      // issue tile t+2 into the buffer just consumed — after re-aligning
+      // so no wave still reads it
       __builtin_amdgcn_s_barrier();
       if (t + 2 < ntiles) stage_tile(tbase + 2 * DTILE, t & 1);
     }
   }
 
-  // merge position sub-groups within the wave (lanes xor LPP, 2*LPP, ...)
+  // ---- merge the PV sub-partitions (lanes xor LPP, 2*LPP, ...) and
+  // write this split's state. m/l are already wave-uniform; each wave
+  // owns its head(s), so there is NO cross-wave merge at all. ----
 #pragma unroll
-  for (int off = LPP; off < WAVE; off <<= 1) {
+  for (int off = LPP; off < WAVE; off <<= 1)
 #pragma unroll
-    for (int gi = 0; gi < MG; ++gi) {
-      if (gi >= group) break;
-      const float mo = __shfl_xor(m[gi], off, WAVE);
-      const float lo = __shfl_xor(l[gi], off, WAVE);
-      float ao[8];
+    for (int gi = 0; gi < NG; ++gi)
 #pragma unroll
-      for (int j = 0; j < 8; ++j) ao[j] = __shfl_xor(acc[gi][j], off, WAVE);
-      const float mn = fmaxf(m[gi], mo);
-      const float a = (m[gi] == -INFINITY && mo == -INFINITY) ? 0.f : __expf(m[gi] - mn);
-      const float b = (m[gi] == -INFINITY && mo == -INFINITY) ? 0.f : __expf(mo - mn);
-      l[gi] = l[gi] * a + lo * b;
-#pragma unroll
-      for (int j = 0; j < 8; ++j) acc[gi][j] = acc[gi][j] * a + ao[j] * b;
-      m[gi] = mn;
-    }
-  }
+      for (int j = 0; j < 8; ++j) acc[gi][j] += __shfl_xor(acc[gi][j], off, WAVE);
 
-  // merge the 4 waves via LDS (reusing the stage region), then write this
-  // split's partial state. Layout: m[4][MAXG], l[4][MAXG], acc[4][MAXG][hd]
-  __syncthreads();  // all ds_reads of the stage image done on every wave
-  float *lm = (float *)lds;
-  float *ll = lm + 4 * MAXG;
-  float *lacc = ll + 4 * MAXG;
-
-  if (lane < LPP) {  // one lane per dim slice (sub==0 lanes)
 #pragma unroll
-    for (int gi = 0; gi < MG; ++gi) {
-      if (gi >= group) break;
-      if (lane == 0) {
-        lm[wid * MAXG + gi] = m[gi];
-        ll[wid * MAXG + gi] = l[gi];
-      }
-#pragma unroll
-      for (int j = 0; j < 8; ++j)
-        lacc[(wid * MAXG + gi) * hd + sl * 8 + j] = acc[gi][j];
-    }
-  }
-  __syncthreads();
-
-  // threads combine the 4 wave-states, one (gi, dim) pair each; group*hd
-  // can exceed the block size (e.g. 8*128), so stride the block over it.
-  // n_splits == 1: this block holds the whole softmax state — write the
-  // output directly, no workspace round-trip, no cross-block combine.
-  for (int tid = threadIdx.x; tid < group * hd; tid += blockDim.x) {
-    const int gi = tid / hd, dd = tid % hd;
-    float M = -INFINITY;
-#pragma unroll
-    for (int w = 0; w < 4; ++w) M = fmaxf(M, lm[w * MAXG + gi]);
-    float L = 0.f, A = 0.f;
-#pragma unroll
-    for (int w = 0; w < 4; ++w) {
-      const float mw = lm[w * MAXG + gi];
-      const float sc = (mw == -INFINITY) ? 0.f : __expf(mw - M);
-      L += ll[w * MAXG + gi] * sc;
-      A += lacc[(w * MAXG + gi) * hd + dd] * sc;
-    }
+  for (int gi = 0; gi < NG; ++gi) {
+    const int gq = wid + 4 * gi;
+    if (gq >= group) break;
     if (n_splits == 1) {
-      out[(size_t)(g * group + gi) * hd + dd] =
-          f32_to_bf16(L > 0.f ? A / L : 0.f);
-      continue;
+      // whole softmax state in this block: write the output directly
+      if (lane < LPP) {
+        const float inv = (l[gi] > 0.f) ? 1.0f / l[gi] : 0.f;
+        bf16x8 o;
+#pragma unroll
+        for (int j = 0; j < 8; ++j) o.u[j] = f32_to_bf16(acc[gi][j] * inv);
+        ((bf16x8 *)(out + (size_t)(g * group + gq) * hd))[sl] = o;
+      }
+    } else {
+      const size_t base = ((size_t)g * n_splits + split) * group + gq;
+      if (lane == 0) {
+        ws_m[base] = m[gi];
+        ws_l[base] = l[gi];
+      }
+      if (lane < LPP) {
+#pragma unroll
+        for (int j = 0; j < 8; ++j)
+          ws_acc[base * hd + sl * 8 + j] = acc[gi][j];
+      }
     }
-    const size_t base = ((size_t)g * n_splits + split) * group + gi;
-    if (dd == 0) {
-      ws_m[base] = M;
-      ws_l[base] = L;
-    }
-    ws_acc[base * hd + dd] = A;
   }
 }
 
