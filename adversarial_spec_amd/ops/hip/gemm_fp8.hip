@@ -212,6 +212,58 @@ DEVINL float fp8_to_f32(uint8_t v) {
   return __builtin_bit_cast(float, bits);
 }
 
+// MFMA-based fp8 decode GEMV: y[n0..n0+15] computed by ONE block.
+// The cvt_pk dot path above is VALU/issue-bound (24 ops per 16 weight
+// bytes -> 2.4-3.6 TB/s effective, round-2 profile); one
+// mfma_f32_16x16x32_fp8_fp8 instead consumes 512 weight bytes per
+// ~20-cycle instruction, pushing the kernel back to the weight-streaming
+// bound. A operand = the x chunk broadcast to all 16 rows (only C row 0
+// is real and read); the 4 waves split K and combine partials in LDS.
+typedef __attribute__((__vector_size__(4 * sizeof(float)))) float f32x4v_;
+
+extern "C" __global__ void __launch_bounds__(256)
+gemv_fp8_mfma_kernel(const uint8_t *__restrict__ x,
+                     const float *__restrict__ xs,
+                     const uint8_t *__restrict__ w,
+                     const float *__restrict__ wsc,
+                     ushort_t *__restrict__ y, int K, int N) {
+  const int lane = threadIdx.x & (WAVE - 1);
+  const int wid = threadIdx.x / WAVE;
+  const int col = lane & 15;          // output column within the tile
+  const int ksl = lane >> 4;          // k-slice quarter: 0..3
+  const int n0 = blockIdx.x * 16;
+  const int n = n0 + col;
+
+  // wave k-range: quarter of K, multiple of 32 (K % 128 == 0 dispatch)
+  const int kq = K / 4;
+  const int k0 = wid * kq;
+
+  const uint8_t *wr = w + (size_t)min(n, N - 1) * K;  // clamp: tail cols
+  f32x4v_ acc0 = {0.f, 0.f, 0.f, 0.f};
+  f32x4v_ acc1 = {0.f, 0.f, 0.f, 0.f};
+  // 2-step unroll: two independent accumulator chains cover the MFMA
+  // dependent latency; loads for both steps issue together.
+  for (int k = k0; k < k0 + kq; k += 64) {
+    const long wv0 = *(const long *)(wr + (size_t)k + ksl * 8);
+    const long xv0 = *(const long *)(x + (size_t)k + ksl * 8);
+    const long wv1 = *(const long *)(wr + (size_t)k + 32 + ksl * 8);
+    const long xv1 = *(const long *)(x + (size_t)k + 32 + ksl * 8);
+    acc0 = __builtin_amdgcn_mfma_f32_16x16x32_fp8_fp8(xv0, wv0, acc0, 0, 0, 0);
+    acc1 = __builtin_amdgcn_mfma_f32_16x16x32_fp8_fp8(xv1, wv1, acc1, 0, 0, 0);
+  }
+
+  // C row 0 lives in lanes 0-15, register 0
+  __shared__ float part[4][16];
+  if (lane < 16) part[wid][lane] = acc0[0] + acc1[0];
+  __syncthreads();
+  if (threadIdx.x < 16 && n0 + (int)threadIdx.x < N) {
+    const float s = part[0][threadIdx.x] + part[1][threadIdx.x] +
+                    part[2][threadIdx.x] + part[3][threadIdx.x];
+    y[n0 + threadIdx.x] =
+        f32_to_bf16(s * xs[0] * wsc[n0 + threadIdx.x]);
+  }
+}
+
 extern "C" __global__ void __launch_bounds__(256)
 gemv_fp8_kernel(const uint8_t *__restrict__ x, const float *__restrict__ xs,
                 const uint8_t *__restrict__ w, const float *__restrict__ wsc,
@@ -356,7 +408,11 @@ extern "C" void launch_gemv_fp8(const uint8_t *x, const float *xs,
                                 const uint8_t *w, const float *wsc,
                                 ushort_t *y, int K, int N,
                                 hipStream_t stream) {
-  if (N <= 8192) {
+  if ((K % 128) == 0) {
+    // weight-streaming-bound MFMA path (the cvt_pk dots are issue-bound)
+    gemv_fp8_mfma_kernel<<<dim3((N + 15) / 16), 256, 0, stream>>>(
+        x, xs, w, wsc, y, K, N);
+  } else if (N <= 8192) {
     gemv_fp8_kernel_w32<<<dim3((N + 7) / 8), 256, 0, stream>>>(x, xs, w, wsc,
                                                                y, K, N);
   } else {
