@@ -408,11 +408,15 @@ extern "C" void launch_gemv_fp8(const uint8_t *x, const float *xs,
                                 const uint8_t *w, const float *wsc,
                                 ushort_t *y, int K, int N,
                                 hipStream_t stream) {
-  if ((K % 128) == 0) {
-    // weight-streaming-bound MFMA path (the cvt_pk dots are issue-bound)
-    gemv_fp8_mfma_kernel<<<dim3((N + 15) / 16), 256, 0, stream>>>(
-        x, xs, w, wsc, y, K, N);
-  } else if (N <= 8192) {
+  // NOTE measured-and-rejected: an MFMA-based variant
+  // (gemv_fp8_mfma_kernel below, fragment-layout direct loads) ran
+  // SLOWER than these cvt_pk dot kernels at every decode shape
+  // (1.3-3.5 TB/s vs 1.7-3.6): the B-fragment's 8-byte per-lane loads
+  // scatter across 16 weight rows and the load ISSUE dominates. A
+  // viable MFMA route needs the guide's M=256-projection pattern
+  // (stage W and x through LDS in full 128-B lines via glds, 8-B
+  // fragment reads from LDS) — future work, not a drop-in.
+  if (N <= 8192) {
     gemv_fp8_kernel_w32<<<dim3((N + 7) / 8), 256, 0, stream>>>(x, xs, w, wsc,
                                                                y, K, N);
   } else {
