@@ -117,7 +117,7 @@ DEVINL void g2_tile(const ushort_t *__restrict__ imA,
   __builtin_amdgcn_s_setprio(0);
   __builtin_amdgcn_s_barrier();
 
-  // ---------- phase 1: read A rows 64-127 + B rows 64-127 (12 reads),
+  // ---------- phase 1: read A rows 64-127 (8 reads),
   //            stage (t+1, B_hi), MFMA quadrant (mi 4-7, ni 0-1)
 #pragma unroll
   for (int mi = 4; mi < 8; ++mi) {
@@ -125,14 +125,6 @@ DEVINL void g2_tile(const ushort_t *__restrict__ imA,
 #pragma unroll
     for (int ks = 0; ks < 2; ++ks)
       afr[mi][ks] = ((const bf16x8v *)(imA + (size_t)row * G2_BK))
-          [(ks * 4 + lhi) ^ (row & 7)];
-  }
-#pragma unroll
-  for (int ni = 2; ni < 4; ++ni) {
-    const int row = brow0 + (ni - 2) * 16 + 32 + lrow;
-#pragma unroll
-    for (int ks = 0; ks < 2; ++ks)
-      bfr[ni][ks] = ((const bf16x8v *)(imB + (size_t)row * G2_BK))
           [(ks * 4 + lhi) ^ (row & 7)];
   }
   if (NSTAGE > 1) g2_stage<3>(sbase, t + 1, lds, wid);
@@ -151,9 +143,20 @@ DEVINL void g2_tile(const ushort_t *__restrict__ imA,
   __builtin_amdgcn_s_setprio(0);
   __builtin_amdgcn_s_barrier();
 
-  // ---------- phase 2: no reads; stage (t+2, A_lo); MFMA (mi 0-3, ni 2-3)
+  // ---------- phase 2: read B rows 64-127 (4 reads);
+  //            stage (t+2, A_lo); MFMA (mi 0-3, ni 2-3)
+#pragma unroll
+  for (int ni = 2; ni < 4; ++ni) {
+    const int row = brow0 + (ni - 2) * 16 + 32 + lrow;
+#pragma unroll
+    for (int ks = 0; ks < 2; ++ks)
+      bfr[ni][ks] = ((const bf16x8v *)(imB + (size_t)row * G2_BK))
+          [(ks * 4 + lhi) ^ (row & 7)];
+  }
   if (NSTAGE > 2) g2_stage<0>(sbase, t + 2, lds, wid);
   __builtin_amdgcn_s_barrier();
+  asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+  __builtin_amdgcn_sched_barrier(0);
   __builtin_amdgcn_s_setprio(1);
 #pragma unroll
   for (int mi = 0; mi < 4; ++mi)
