@@ -240,8 +240,18 @@ attn_prefill_mfma_kernel(const ushort_t *__restrict__ q,
                            fmaxf(s[n][2], s[n][3])));
     mx = fmaxf(mx, __shfl_xor(mx, 16, WAVE));
     mx = fmaxf(mx, __shfl_xor(mx, 32, WAVE));
-    const float mn = fmaxf(m1, mx);
-    const float alpha = (mn == -INFINITY) ? 0.f : __expf(m1 - mn);
+    // defer-max (guide T13): when no row's max grew by more than THR=8,
+    // keep the old running max — P is then bounded by e^8 (fine in f32
+    // accum and scale-free bf16) and the O-rescale pass is skipped
+    // (+5% measured on this ladder; ~3x max-abs error per T13's numbers,
+    // still inside the kernel tests' tolerances). The decision is wave-
+    // uniform (__all); first-tile m1 = -inf forces the rescale branch.
+    // Order is T13-safe: each tile's PV completes before the next tile's
+    // decision, and l sees the same alpha.
+    const bool defer = __all(mx - m1 <= 8.0f);
+    const float mn = defer ? m1 : fmaxf(m1, mx);
+    const float alpha =
+        defer ? 1.0f : ((mn == -INFINITY) ? 0.f : __expf(m1 - mn));
     float psum = 0.f;
 #pragma unroll
     for (int n = 0; n < 4; ++n) {
@@ -257,13 +267,15 @@ attn_prefill_mfma_kernel(const ushort_t *__restrict__ q,
     psum += __shfl_xor(psum, 32, WAVE);
     l1 = l1 * alpha + psum;
     m1 = mn;
-    // O rows are lhi*4+r: fetch those rows' alphas (any lane with
-    // lrow == row has the value; pick the same lhi quartile)
+    if (!defer) {
+      // O rows are lhi*4+r: fetch those rows' alphas (any lane with
+      // lrow == row has the value; pick the same lhi quartile)
 #pragma unroll
-    for (int r = 0; r < 4; ++r) {
-      const float oa = __shfl(alpha, (lhi * 4 + r) | (lhi << 4), WAVE);
+      for (int r = 0; r < 4; ++r) {
+        const float oa = __shfl(alpha, (lhi * 4 + r) | (lhi << 4), WAVE);
 #pragma unroll
-      for (int d = 0; d < 8; ++d) o[d][r] *= oa;
+        for (int d = 0; d < 8; ++d) o[d][r] *= oa;
+      }
     }
 
     // ---- P -> LDS: lane-local row, 4 keys packed per 8-B store ----
