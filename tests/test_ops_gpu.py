@@ -480,3 +480,26 @@ class TestFp8Engine:
             agree += int(lb.argmax().item()) == int(lf.argmax().item())
             tok = int(lb.argmax().item())
         assert agree >= 4, f"fp8 greedy agreed on only {agree}/6 steps"
+
+
+class TestGemvGateup:
+    def test_fused_vs_ref(self):
+        """gemv_gateup == silu(x@Wg^T) * (x@Wu^T) on the fused layout."""
+        k, f = 1024, 768
+        x = _bf(torch.randn(1, k)).to(DEV)
+        w = _bf(torch.randn(2 * f, k) * 0.05).to(DEV)
+        act = torch.empty(1, f, dtype=torch.bfloat16, device=DEV)
+        ops.gemv_gateup(x, w, act)
+        gu = x.float().cpu() @ w.float().cpu().t()
+        want = torch_ref.swiglu(gu[:, :f], gu[:, f:])
+        _assert_close(act, want, atol=3e-2, name="gemv_gateup")
+
+    def test_matches_unfused_ops(self):
+        k, f = 4096, 14336  # llama-3-8b decode shape
+        x = _bf(torch.randn(1, k)).to(DEV)
+        w = _bf(torch.randn(2 * f, k) * 0.02).to(DEV)
+        act = torch.empty(1, f, dtype=torch.bfloat16, device=DEV)
+        ops.gemv_gateup(x, w, act)
+        gu = ops.gemv(x, w)
+        want = ops.swiglu(gu[:, :f], gu[:, f:])
+        _assert_close(act, want.float().cpu(), atol=3e-2, name="fused vs ops")
