@@ -1,0 +1,45 @@
+"""Daemon warm-engine evidence on MI355X (round-2 verdict item 6):
+round 2+ of a session through `debate.py serve` skips model init."""
+
+from __future__ import annotations
+
+import threading
+import time
+
+import pytest
+import torch
+
+pytestmark = pytest.mark.gpu
+
+SPEC = "# Spec\n\nA small spec to critique."
+
+
+@pytest.mark.skipif(not torch.cuda.is_available(), reason="needs GPU")
+def test_daemon_round2_skips_engine_init(tmp_path, monkeypatch):
+    from adversarial_spec_amd import daemon
+
+    sock = tmp_path / "d.sock"
+    monkeypatch.setattr(daemon, "SOCKET_PATH", sock)
+    srv = daemon.serve(sock)
+    t = threading.Thread(target=srv.serve_forever, daemon=True)
+    t.start()
+    try:
+        argv = ["critique", "--models", "local/debug-1b", "--json",
+                "--timeout", "120"]
+        t0 = time.monotonic()
+        code, out, err = daemon.try_forward(argv, SPEC, sock)
+        cold = time.monotonic() - t0
+        assert code == 0, err
+        assert '"results"' in out
+
+        t0 = time.monotonic()
+        code, out, err = daemon.try_forward(argv, SPEC, sock)
+        warm = time.monotonic() - t0
+        assert code == 0, err
+        # round 2 reuses the resident engine: no init_random, no graph
+        # capture — it must be dramatically faster than round 1
+        assert warm < cold * 0.6, (cold, warm)
+        print(f"daemon rounds: cold {cold:.1f}s warm {warm:.1f}s")
+    finally:
+        srv.shutdown()
+        srv.server_close()
