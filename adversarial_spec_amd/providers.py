@@ -128,7 +128,8 @@ LOCAL_MODEL_MAP: dict[str, str] = {
     "llama-3.1-8b": "llama-3-8b",
     "llama-3.1-70b": "llama-3-70b",
     "mistral-7b": "mistral-7b",
-    "tiny": "tiny",  # 4-layer test model
+    "tiny": "tiny",          # 4-layer CPU test model
+    "debug-1b": "debug-1b",  # 16-layer GPU smoke/test model
 }
 
 # ---------------------------------------------------------------------------
