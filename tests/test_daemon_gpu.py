@@ -24,8 +24,11 @@ def test_daemon_round2_skips_engine_init(tmp_path, monkeypatch):
     t = threading.Thread(target=srv.serve_forever, daemon=True)
     t.start()
     try:
+        # --timeout 5 caps the decode deadline so both rounds decode for
+        # ~5 s and the round-1-only costs (engine init, graph capture)
+        # dominate the difference
         argv = ["critique", "--models", "local/debug-1b", "--json",
-                "--timeout", "120"]
+                "--timeout", "5"]
         t0 = time.monotonic()
         code, out, err = daemon.try_forward(argv, SPEC, sock)
         cold = time.monotonic() - t0
@@ -37,8 +40,8 @@ def test_daemon_round2_skips_engine_init(tmp_path, monkeypatch):
         warm = time.monotonic() - t0
         assert code == 0, err
         # round 2 reuses the resident engine: no init_random, no graph
-        # capture — it must be dramatically faster than round 1
-        assert warm < cold * 0.6, (cold, warm)
+        # capture — at least a second of round-1-only work disappears
+        assert warm < cold - 1.0, (cold, warm)
         print(f"daemon rounds: cold {cold:.1f}s warm {warm:.1f}s")
     finally:
         srv.shutdown()
