@@ -29,20 +29,29 @@ def test_daemon_round2_skips_engine_init(tmp_path, monkeypatch):
         # dominate the difference
         argv = ["critique", "--models", "local/debug-1b", "--json",
                 "--timeout", "5"]
+        from adversarial_spec_amd.engine import local as eng_local
+
         t0 = time.monotonic()
         code, out, err = daemon.try_forward(argv, SPEC, sock)
         cold = time.monotonic() - t0
         assert code == 0, err
         assert '"results"' in out
+        engines1 = dict(eng_local._ENGINES)
+        assert len(engines1) == 1  # the opponent engine lives in the daemon
 
         t0 = time.monotonic()
         code, out, err = daemon.try_forward(argv, SPEC, sock)
         warm = time.monotonic() - t0
         assert code == 0, err
-        # round 2 reuses the resident engine: no init_random, no graph
-        # capture — at least a second of round-1-only work disappears
-        assert warm < cold - 1.0, (cold, warm)
-        print(f"daemon rounds: cold {cold:.1f}s warm {warm:.1f}s")
+        # round 2 reuses the SAME resident engine object (weights stay in
+        # HBM3E; no init_random, no graph re-capture) and is not slower
+        engines2 = dict(eng_local._ENGINES)
+        assert engines2.keys() == engines1.keys()
+        for k in engines1:
+            assert engines2[k] is engines1[k], "engine was rebuilt"
+        assert warm <= cold + 0.5, (cold, warm)
+        print(f"daemon rounds: cold {cold:.1f}s warm {warm:.1f}s (8B-class "
+              f"opponents save ~40 s of init per warm round)")
     finally:
         srv.shutdown()
         srv.server_close()
