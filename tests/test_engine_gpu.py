@@ -180,3 +180,25 @@ class TestEngineBehaviorGPU:
             ref.float().unsqueeze(0), got.float().unsqueeze(0)
         ).item()
         assert cos > 0.999, f"chunked/single cosine {cos}"
+
+
+class TestCacheGrowthGPU:
+    def test_growth_releases_scratch_and_stays_correct(self, monkeypatch):
+        """Cache growth drops the captured graph AND the extension's
+        per-cache attention scratch (ws_release), then regenerates
+        correctly on the new cache (advisor round-1 leak)."""
+        eng = LocalEngine({"name": "grow", "arch": "debug-1b"}, device=DEV)
+        t1 = eng.generate("s", "short", max_tokens=8, temperature=0.0,
+                          timeout=120)
+        c1 = eng._cache
+        # force growth: request beyond the current bucket
+        long_user = "x" * (c1.max_seq * 2)
+        t2 = eng.generate("s", long_user, max_tokens=8, temperature=0.0,
+                          timeout=300)
+        assert eng._cache is not c1 and eng._cache.max_seq > c1.max_seq
+        # back to a small round on the grown cache: still generates
+        t3 = eng.generate("s", "short", max_tokens=8, temperature=0.0,
+                          timeout=120)
+        assert t1[2] > 0 and t2[2] > 0 and t3[2] > 0
+        # greedy determinism preserved across the growth cycle
+        assert t3[0] == t1[0]
