@@ -346,10 +346,15 @@ torch::Tensor mfma_probe16(torch::Tensor a, torch::Tensor b) {
 // KV bytes, so occupancy (total blocks) is the lever; ADVSPEC_SPLIT_BLOCKS
 // overrides the target block count for A/B tuning on hardware.
 static void split_geometry(long seq, int kh, int* n_splits, int* split_len) {
+  // Default 256: the production regime is SEVERAL co-resident opponents
+  // (bench A/B under 3-way concurrency: 256-target 0.594 critiques/s vs
+  // 512-target 0.581 — each opponent needs less self-occupancy when the
+  // chip is shared, and the split/combine tails shrink). Solo-latency
+  // workloads can raise it (solo sweep optimum was 512).
   static const int cap_blocks = [] {
     const char* e = getenv("ADVSPEC_SPLIT_BLOCKS");
-    const int v = e ? atoi(e) : 512;
-    return v > 0 ? v : 512;
+    const int v = e ? atoi(e) : 256;
+    return v > 0 ? v : 256;
   }();
   const int target = std::max(1, cap_blocks / kh);
   int ns = std::max(1, std::min((int)((seq + 63) / 64), target));
