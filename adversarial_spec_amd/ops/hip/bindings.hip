@@ -204,9 +204,13 @@ void rope_kv(torch::Tensor q, torch::Tensor k, torch::Tensor v,
   const int page = kc.size(1);
   TORCH_CHECK(hd % 16 == 0, "rope_kv: hd % 16 == 0");
   const int waves = t * (hq + 2 * hk);
-  const int blocks = (waves * 64 + 255) / 256;
   const int* pp = pos_state.has_value() ? pos_state->data_ptr<int>() : nullptr;
-  rope_kv_kernel<<<blocks, 256, 0, cur_stream()>>>(
+  // decode (t==1): one wave per 64-thread block spreads the ~48
+  // latency-bound slot-waves over 4x the CUs (the 12-block launch
+  // measured 11.6 us avg under 3-opponent concurrency)
+  const int thr = t <= 2 ? 64 : 256;
+  const int blocks = (waves * 64 + thr - 1) / thr;
+  rope_kv_kernel<<<blocks, thr, 0, cur_stream()>>>(
       uptr_mut(q), uptr_mut(k), uptr(v), cost.data_ptr<float>(),
       sint.data_ptr<float>(), uptr_mut(kc), uptr_mut(vc),
       page_table.data_ptr<int>(), t, hq, hk, hd, (int)pos0, q.stride(0),
