@@ -199,16 +199,22 @@ __global__ void __launch_bounds__(256) attn_decode_split_kernel(
           a4[c & 3] = dot8_bf16(__builtin_bit_cast(bf16x8, qv),
                                 __builtin_bit_cast(bf16x8, kv), a4[c & 3]);
         }
-        const float s =
-            ((a4[0] + a4[1]) + (a4[2] + a4[3])) * scale;
+        const float s = (a4[0] + a4[1]) + (a4[2] + a4[3]);  // raw dot
 
-        // ---- online softmax for this wave's head: ONE wave-wide
-        // max/sum pair per tile ----
+        // ---- online softmax for this wave's head in the LOG2-SCALED
+        // domain (m tracks m2 = max*scale*log2e; p = exp2(fma(s, scale2,
+        // -m2)) — one fma+exp per element, no scale/sub passes); ONE
+        // wave-wide max/sum pair per tile ----
+        const float scale2 = scale * 1.4426950408889634f;
         float mx = valid ? s : -INFINITY;
         mx = wave_reduce_max(mx);
-        const float mn = fmaxf(m[gi], mx);
-        const float alpha = (mn == -INFINITY) ? 0.f : __expf(m[gi] - mn);
-        const float p = (valid && mn != -INFINITY) ? __expf(s - mn) : 0.f;
+        const float mx2 = mx * scale2;
+        const float mn = fmaxf(m[gi], mx2);
+        const float alpha = (mn == -INFINITY) ? 0.f : exp2f(m[gi] - mn);
+        const float p =
+            (valid && mn != -INFINITY)
+                ? exp2f(__builtin_fmaf(s, scale2, -mn))
+                : 0.f;
         const float ps = wave_reduce_sum(p);
         l[gi] = l[gi] * alpha + ps;
         m[gi] = mn;

@@ -213,24 +213,24 @@ attn_prefill_mfma_kernel(const ushort_t *__restrict__ q,
     const bool full_tile =
         (kt + KVBLK <= tk) &&
         (!causal || kt + KVBLK <= kv_offset + q0 + 1);
-    if (full_tile) {
-#pragma unroll
-      for (int n = 0; n < 4; ++n)
-#pragma unroll
-        for (int r = 0; r < 4; ++r) s[n][r] *= scale;
-    } else {
+    // softmax runs in the LOG2-SCALED domain (exp2+fma fold, ladder
+    // item): raw scores stay unscaled; m/l track m2 = m*scale*log2e and
+    // p = exp2(fma(s_raw, scale2, -m2)) — one v_fma + v_exp per element
+    // instead of mul + sub + (mul+exp). Masked keys set s_raw = -inf
+    // (the fma propagates it).
+    if (!full_tile) {
       const int qrow_abs = kv_offset + q0 + lrow;
 #pragma unroll
       for (int n = 0; n < 4; ++n) {
 #pragma unroll
         for (int r = 0; r < 4; ++r) {
           const int key = kt + n * 16 + lhi * 4 + r;
-          float sv = s[n][r] * scale;
-          if (key >= tk || (causal && key > qrow_abs)) sv = -INFINITY;
-          s[n][r] = sv;
+          if (key >= tk || (causal && key > qrow_abs))
+            s[n][r] = -INFINITY;
         }
       }
     }
+    const float scale2 = scale * 1.4426950408889634f;
     // in-lane 16-value max tree + 2 cross-lane levels (lanes lrow,
     // lrow+16, lrow+32, lrow+48 hold the rest of this row's keys)
     float mx = fmaxf(fmaxf(s[0][0], s[0][1]), fmaxf(s[0][2], s[0][3]));
@@ -240,25 +240,25 @@ attn_prefill_mfma_kernel(const ushort_t *__restrict__ q,
                            fmaxf(s[n][2], s[n][3])));
     mx = fmaxf(mx, __shfl_xor(mx, 16, WAVE));
     mx = fmaxf(mx, __shfl_xor(mx, 32, WAVE));
-    // defer-max (guide T13): when no row's max grew by more than THR=8,
-    // keep the old running max — P is then bounded by e^8 (fine in f32
-    // accum and scale-free bf16) and the O-rescale pass is skipped
-    // (+5% measured on this ladder; ~3x max-abs error per T13's numbers,
-    // still inside the kernel tests' tolerances). The decision is wave-
-    // uniform (__all); first-tile m1 = -inf forces the rescale branch.
-    // Order is T13-safe: each tile's PV completes before the next tile's
-    // decision, and l sees the same alpha.
-    const bool defer = __all(mx - m1 <= 8.0f);
-    const float mn = defer ? m1 : fmaxf(m1, mx);
+    const float mx2 = mx * scale2;  // log2-scaled row max (one op)
+    // defer-max (guide T13): when no row's max grew by more than THR=8
+    // (= 8*log2e in this domain), keep the old running max — P is then
+    // bounded by e^8 (fine in f32 accum and scale-free bf16) and the
+    // O-rescale pass is skipped (+5% measured on this ladder; ~3x
+    // max-abs error per T13's numbers, inside the kernel tests'
+    // tolerances). Wave-uniform (__all); first-tile m1 = -inf forces
+    // the rescale branch. Order is T13-safe: each tile's PV completes
+    // before the next tile's decision, and l sees the same alpha.
+    const bool defer = __all(mx2 - m1 <= 11.5416f);
+    const float mn = defer ? m1 : fmaxf(m1, mx2);
     const float alpha =
-        defer ? 1.0f : ((mn == -INFINITY) ? 0.f : __expf(m1 - mn));
+        defer ? 1.0f : ((mn == -INFINITY) ? 0.f : exp2f(m1 - mn));
     float psum = 0.f;
 #pragma unroll
     for (int n = 0; n < 4; ++n) {
 #pragma unroll
       for (int r = 0; r < 4; ++r) {
-        const float p =
-            (s[n][r] == -INFINITY) ? 0.f : __expf(s[n][r] - mn);
+        const float p = exp2f(__builtin_fmaf(s[n][r], scale2, -mn));
         s[n][r] = p;
         psum += p;
       }
