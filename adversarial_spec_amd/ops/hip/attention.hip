@@ -334,6 +334,8 @@ attn_decode_combine_kernel(const float *__restrict__ ws_m,
   // pass 2: strided L/acc accumulation, 4 independent accumulators per
   // wave (16 concurrent loads across the block) — the ws round-trip is
   // cross-XCD latency-bound, so ILP depth is the lever
+  // ws_m holds the split kernel's LOG2-SCALED running max (m2 domain):
+  // rescale factors are exp2, not exp
   float L = 0.f, Au[4] = {0.f, 0.f, 0.f, 0.f};
   int s = sg;
   for (; s + 12 < n_splits; s += 16) {
@@ -341,7 +343,7 @@ attn_decode_combine_kernel(const float *__restrict__ ws_m,
     for (int u = 0; u < 4; ++u) {
       const size_t b = ((size_t)g * n_splits + s + 4 * u) * group + gi;
       const float mw = ws_m[b];
-      const float sc = (mw == -INFINITY) ? 0.f : __expf(mw - M);
+      const float sc = (mw == -INFINITY) ? 0.f : exp2f(mw - M);
       L += ws_l[b] * sc;
       Au[u] += ws_acc[b * hd + dd] * sc;
     }
@@ -349,7 +351,7 @@ attn_decode_combine_kernel(const float *__restrict__ ws_m,
   for (; s < n_splits; s += 4) {
     const size_t b = ((size_t)g * n_splits + s) * group + gi;
     const float mw = ws_m[b];
-    const float sc = (mw == -INFINITY) ? 0.f : __expf(mw - M);
+    const float sc = (mw == -INFINITY) ? 0.f : exp2f(mw - M);
     L += ws_l[b] * sc;
     Au[0] += ws_acc[b * hd + dd] * sc;
   }
