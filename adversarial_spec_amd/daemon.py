@@ -122,8 +122,18 @@ def run_request(argv: list, stdin_text: str) -> tuple[int, str, str]:
 
 
 def serve(socket_path: Optional[Path] = None) -> _Server:
-    """Bind the unix socket and serve forever (foreground)."""
+    """Bind the unix socket and return the server (caller serves forever).
+
+    Refuses to displace a LIVE daemon (a second `serve` would silently
+    steal the socket and strand the first process's warm engines); a
+    stale socket file from a dead daemon is cleaned up.
+    """
     path = Path(socket_path or SOCKET_PATH)
+    if ping(path):
+        raise RuntimeError(
+            f"an adversarial-spec daemon is already serving on {path} "
+            "(stop it with: debate.py serve stop)"
+        )
     path.parent.mkdir(parents=True, exist_ok=True)
     with contextlib.suppress(FileNotFoundError):
         path.unlink()

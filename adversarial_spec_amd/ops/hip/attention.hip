@@ -89,8 +89,9 @@ __global__ void __launch_bounds__(256) attn_decode_split_kernel(
   ushort_t *qlds = lds + 4 * TILE_E;
 
   // K chunk swizzle: a 256-B LDS bank row holds 32/LPP image rows, so
-  // same-chunk reads of rows p and p' collide when p ≡ p' (mod 16/LPP
-  // ... spread by XORing the chunk with the row's bank-alias class.
+  // same-chunk reads of two rows collide whenever the rows share a
+  // bank-row offset (p ≡ p' mod 16/LPP); XOR the chunk index with the
+  // row's bank-alias class to spread them.
   constexpr int SWZ_DIV = 16 / LPP;  // rows sharing a bank-row offset
   auto kswz = [&](int row) { return (row / SWZ_DIV) & (LPP - 1); };
 

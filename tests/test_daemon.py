@@ -80,3 +80,34 @@ class TestDaemon:
         payload = json.loads(capsys.readouterr().out)
         assert payload["all_agreed"] is True
         assert srv.requests_served == 1
+
+
+class TestDaemonGuards:
+    def test_second_serve_refused_while_live(self, live_daemon):
+        sock, _srv = live_daemon
+        with pytest.raises(RuntimeError, match="already serving"):
+            daemon.serve(sock)
+
+    def test_stale_socket_cleaned_up(self, tmp_path):
+        sock = tmp_path / "stale.sock"
+        sock.touch()  # dead socket file, nothing listening
+        srv = daemon.serve(sock)
+        try:
+            assert sock.exists()
+        finally:
+            srv.server_close()
+
+    def test_concurrent_forwards(self, live_daemon):
+        """Two critiques in flight at once (the threaded server + per-
+        engine locks must not deadlock or cross wires)."""
+        import concurrent.futures as cf
+
+        sock, srv = live_daemon
+        argv = ["critique", "--models", "stub/agree", "--json"]
+        with cf.ThreadPoolExecutor(2) as pool:
+            futs = [pool.submit(daemon.try_forward, argv, SPEC, sock)
+                    for _ in range(2)]
+            results = [f.result(timeout=60) for f in futs]
+        for r in results:
+            assert r is not None and r[0] == 0
+        assert srv.requests_served == 2
