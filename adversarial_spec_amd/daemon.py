@@ -45,6 +45,13 @@ SOCKET_PATH = Path(
 # stop forwarding loops: set in the daemon process before re-entering main()
 _IN_DAEMON_ENV = "ADVSPEC_IN_DAEMON"
 
+# one CLI invocation at a time: run_request redirects the PROCESS-global
+# stdio and environment, so concurrent requests must queue (they would
+# interleave output otherwise — caught by the concurrency test). The
+# expensive inner work is engine inference, which the per-engine generate
+# locks serialize per opponent anyway.
+_REQUEST_LOCK = threading.Lock()
+
 
 def _recv_line(sock: socket.socket, limit: int = 64 * 1024 * 1024) -> bytes:
     chunks = []
@@ -92,9 +99,15 @@ class _Server(socketserver.ThreadingUnixStreamServer):
 
 
 def run_request(argv: list, stdin_text: str) -> tuple[int, str, str]:
-    """Execute one CLI invocation inside the daemon process."""
+    """Execute one CLI invocation inside the daemon process (serialized:
+    stdio/env redirection is process-global)."""
     from .cli import debate as cli
 
+    with _REQUEST_LOCK:
+        return _run_request_locked(cli, argv, stdin_text)
+
+
+def _run_request_locked(cli, argv: list, stdin_text: str) -> tuple[int, str, str]:
     prev_env = os.environ.get(_IN_DAEMON_ENV)
     os.environ[_IN_DAEMON_ENV] = "1"
     out_buf, err_buf = io.StringIO(), io.StringIO()
