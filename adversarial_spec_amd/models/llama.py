@@ -490,12 +490,25 @@ class LlamaModel:
                 return out
             return ops.gemv(x, getattr(L_, name), out=out)
 
+        # norm/residual GEMV fusion (bf16, non-TP): the rmsnorm folds into
+        # the consuming GEMV's prologue and the residual add into the
+        # producing GEMV's epilogue — 65 fewer launches per step (the
+        # decode anatomy's add_rmsnorm x2 row, 0.30 ms/tok, was pure
+        # launch-bound small kernels). TP keeps the unfused sequence: the
+        # all-reduce must see the raw partial projection before the add.
+        fused = (not fp8) and (self.tp is None or self.tp.size == 1)
+
         torch.index_select(self.embed, 0, W.tok_long, out=W.resid)
-        ops.rmsnorm(W.resid, self.layers[0].attn_norm, c.norm_eps,
-                    out=W.normed)
+        if not fused:
+            ops.rmsnorm(W.resid, self.layers[0].attn_norm, c.norm_eps,
+                        out=W.normed)
         for i, L in enumerate(self.layers):
             Qd = self.layers_q[i] if fp8 else None
-            proj(W.normed, L, Qd, "wqkv", W.qkv)
+            if fused:
+                ops.gemv_norm(W.resid, L.attn_norm, L.wqkv, c.norm_eps,
+                              out=W.qkv)
+            else:
+                proj(W.normed, L, Qd, "wqkv", W.qkv)
             q = W.qkv[:, : h * hd].view(1, h, hd)
             k = W.qkv[:, h * hd : (h + kh) * hd].view(1, kh, hd)
             v = W.qkv[:, (h + kh) * hd :].view(1, kh, hd)
@@ -506,6 +519,12 @@ class LlamaModel:
                 max_seq_bound, self.scale, pos_state=pos_state, out=W.attn,
                 identity=cache.identity,
             )
+            if fused:
+                ops.gemv_res(W.attn.view(1, h * hd), L.wo, W.resid)
+                ops.gemv_gateup_norm(W.resid, L.mlp_norm, L.w_gate_up,
+                                     c.norm_eps, W.act)
+                ops.gemv_res(W.act, L.w_down, W.resid)
+                continue
             proj(W.attn.view(1, h * hd), L, Qd, "wo", W.attn_out)
             if self.tp is not None and self.tp.size > 1:
                 self.tp.all_reduce_(W.attn_out)
@@ -525,7 +544,10 @@ class LlamaModel:
                    else self.final_norm)
             ops.add_rmsnorm(W.resid2, W.mlp_out, nxt, c.norm_eps,
                             out_resid=W.resid, out_y=W.normed)
-        if self.lm_head_q is not None:
+        if fused:
+            ops.gemv_norm(W.resid, self.final_norm, self.lm_head, c.norm_eps,
+                          out=W.logits)
+        elif self.lm_head_q is not None:
             ops.gemv_fp8(W.normed, self.lm_head_q.q, self.lm_head_q.s,
                          W.x8, W.xs, W.logits)
         else:

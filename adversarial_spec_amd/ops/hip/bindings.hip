@@ -50,6 +50,12 @@ void launch_gemv(const ushort_t*, const ushort_t*, ushort_t*, int, int,
                  hipStream_t);
 void launch_gemv_gateup(const ushort_t*, const ushort_t*, ushort_t*, int, int,
                         hipStream_t);
+void launch_gemv_norm(const ushort_t*, const ushort_t*, const ushort_t*,
+                      ushort_t*, int, int, float, hipStream_t);
+void launch_gemv_res(const ushort_t*, const ushort_t*, ushort_t*, int, int,
+                     hipStream_t);
+void launch_gemv_gateup_norm(const ushort_t*, const ushort_t*, const ushort_t*,
+                             ushort_t*, int, int, float, hipStream_t);
 void launch_gemm(const ushort_t*, const ushort_t*, ushort_t*, int, int, int,
                  hipStream_t);
 void launch_gemm256(const ushort_t*, const ushort_t*, ushort_t*, int, int, int,
@@ -450,6 +456,66 @@ void gemv_gateup(torch::Tensor x, torch::Tensor w, torch::Tensor act) {
                      cur_stream());
 }
 
+// rmsnorm fused into the GEMV prologue: y = rms(x) * ((x*wln) @ w^T).
+// Replaces the separate (add_)rmsnorm launch before every consuming decode
+// GEMV (see gemv.hip fusion comment).
+torch::Tensor gemv_norm(torch::Tensor x, torch::Tensor wln, torch::Tensor w,
+                        double eps, c10::optional<torch::Tensor> out_opt) {
+  CHECK_BF16_CUDA(x);
+  CHECK_BF16_CUDA(wln);
+  CHECK_BF16_CUDA(w);
+  TORCH_CHECK(w.dim() == 2 && w.is_contiguous());
+  const int N = w.size(0), K = w.size(1);
+  TORCH_CHECK((long)x.numel() == (long)K, "gemv_norm: x numel == K");
+  TORCH_CHECK((long)wln.numel() == (long)K, "gemv_norm: wln numel == K");
+  TORCH_CHECK(K % 8 == 0, "gemv_norm: K % 8 == 0");
+  auto xc = x.contiguous();
+  auto lc = wln.contiguous();
+  auto y = out_opt.has_value()
+               ? *out_opt
+               : torch::empty(x.dim() == 2 ? std::vector<int64_t>{1, N}
+                                           : std::vector<int64_t>{N},
+                              x.options());
+  TORCH_CHECK((long)y.numel() == (long)N, "gemv_norm: out numel == N");
+  launch_gemv_norm(uptr(xc), uptr(lc), uptr(w), uptr_mut(y), K, N,
+                   (float)eps, cur_stream());
+  return y;
+}
+
+// residual-add fused into the GEMV epilogue: resid += x @ w^T (in place).
+// NOT valid under TP (the all-reduce needs the raw partial product).
+void gemv_res(torch::Tensor x, torch::Tensor w, torch::Tensor resid) {
+  CHECK_BF16_CUDA(x);
+  CHECK_BF16_CUDA(w);
+  CHECK_BF16_CUDA(resid);
+  TORCH_CHECK(w.dim() == 2 && w.is_contiguous());
+  const int N = w.size(0), K = w.size(1);
+  TORCH_CHECK((long)x.numel() == (long)K && K % 8 == 0);
+  TORCH_CHECK((long)resid.numel() == (long)N, "gemv_res: resid numel == N");
+  TORCH_CHECK(resid.is_contiguous());
+  TORCH_CHECK(N <= 8192, "gemv_res: w32 kernel only (hidden <= 8192)");
+  auto xc = x.contiguous();
+  launch_gemv_res(uptr(xc), uptr(w), uptr_mut(resid), K, N, cur_stream());
+}
+
+// rmsnorm + gate_up GEMV + SwiGLU in one launch (decode MLP front half).
+void gemv_gateup_norm(torch::Tensor x, torch::Tensor wln, torch::Tensor w,
+                      double eps, torch::Tensor act) {
+  CHECK_BF16_CUDA(x);
+  CHECK_BF16_CUDA(wln);
+  CHECK_BF16_CUDA(w);
+  CHECK_BF16_CUDA(act);
+  TORCH_CHECK(w.dim() == 2 && w.is_contiguous());
+  const int K = w.size(1), F2 = w.size(0);
+  TORCH_CHECK(F2 % 2 == 0 && (long)x.numel() == (long)K && K % 8 == 0);
+  TORCH_CHECK((long)wln.numel() == (long)K);
+  TORCH_CHECK((long)act.numel() == (long)(F2 / 2));
+  auto xc = x.contiguous();
+  auto lc = wln.contiguous();
+  launch_gemv_gateup_norm(uptr(xc), uptr(lc), uptr(w), uptr_mut(act), K,
+                          F2 / 2, (float)eps, cur_stream());
+}
+
 // Tiled MFMA GEMM: C = A @ B, bf16, fp32 accumulation. Replaces library
 // GEMMs on the prefill path (deterministic, workspace-free; see gemm.hip).
 torch::Tensor gemm(torch::Tensor a, torch::Tensor b) {
@@ -669,6 +735,12 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("gemv", &gemv, "batch-1 decode GEMV (weight streaming)",
         py::arg("x"), py::arg("w"), py::arg("out") = py::none());
   m.def("gemv_gateup", &gemv_gateup, "fused gate_up GEMV + SwiGLU (decode)");
+  m.def("gemv_norm", &gemv_norm, "rmsnorm-prologue GEMV (decode fusion)",
+        py::arg("x"), py::arg("wln"), py::arg("w"), py::arg("eps"),
+        py::arg("out") = py::none());
+  m.def("gemv_res", &gemv_res, "residual-add-epilogue GEMV (decode fusion)");
+  m.def("gemv_gateup_norm", &gemv_gateup_norm,
+        "rmsnorm + gate_up GEMV + SwiGLU (decode fusion)");
   m.def("gemm", &gemm, "tiled MFMA GEMM (bf16, fp32 accum)");
   m.def("gemm_variant", &gemm_variant, "force GEMM kernel 128/256 (A/B)");
   m.def("gemm_fp8", &gemm_fp8, "fp8 e4m3 MFMA GEMM (rowwise scales)");

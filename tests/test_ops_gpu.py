@@ -506,3 +506,82 @@ class TestGemvGateup:
         gu = ops.gemv(x, w)
         want = ops.swiglu(gu[:, :f], gu[:, f:])
         _assert_close(act, want.float().cpu(), atol=3e-2, name="fused vs ops")
+
+
+class TestGemvNormResFusion:
+    """Norm-prologue / residual-epilogue GEMV fusion (decode launch-count
+    reduction): each fused kernel vs the plain fp32 torch reference."""
+
+    EPS = 1e-5
+
+    def _norm_ref(self, x, wln):
+        xf = x.float().cpu()
+        rms = torch.rsqrt(xf.pow(2).mean(-1, keepdim=True) + self.EPS)
+        return xf * rms * wln.float().cpu()
+
+    def test_gemv_norm_w32(self):
+        k, n = 4096, 6144  # qkv shape (w32 kernel)
+        x = _bf(torch.randn(1, k)).to(DEV)
+        wln = _bf(torch.rand(k) + 0.5).to(DEV)
+        w = _bf(torch.randn(n, k) * 0.02).to(DEV)
+        y = ops.gemv_norm(x, wln, w, self.EPS)
+        want = self._norm_ref(x, wln) @ w.float().cpu().t()
+        _assert_close(y, want, atol=8e-2, name="gemv_norm w32")
+
+    def test_gemv_norm_large_n(self):
+        k, n = 512, 16384  # 16-lane kernel path (N > 8192, lm_head-like)
+        x = _bf(torch.randn(1, k)).to(DEV)
+        wln = _bf(torch.rand(k) + 0.5).to(DEV)
+        w = _bf(torch.randn(n, k) * 0.05).to(DEV)
+        y = ops.gemv_norm(x, wln, w, self.EPS)
+        want = self._norm_ref(x, wln) @ w.float().cpu().t()
+        _assert_close(y, want, atol=8e-2, name="gemv_norm 16-lane")
+
+    def test_gemv_res_inplace(self):
+        k, n = 4096, 4096  # o/down shape
+        x = _bf(torch.randn(1, k)).to(DEV)
+        w = _bf(torch.randn(n, k) * 0.02).to(DEV)
+        resid = _bf(torch.randn(1, n)).to(DEV)
+        want = resid.float().cpu() + x.float().cpu() @ w.float().cpu().t()
+        ops.gemv_res(x, w, resid)
+        _assert_close(resid, want, atol=8e-2, name="gemv_res")
+
+    def test_gemv_gateup_norm(self):
+        k, f = 4096, 14336
+        x = _bf(torch.randn(1, k)).to(DEV)
+        wln = _bf(torch.rand(k) + 0.5).to(DEV)
+        w = _bf(torch.randn(2 * f, k) * 0.02).to(DEV)
+        act = torch.empty(1, f, dtype=torch.bfloat16, device=DEV)
+        ops.gemv_gateup_norm(x, wln, w, self.EPS, act)
+        gu = self._norm_ref(x, wln) @ w.float().cpu().t()
+        want = torch_ref.swiglu(gu[:, :f], gu[:, f:])
+        _assert_close(act, want, atol=5e-2, name="gemv_gateup_norm")
+
+    def test_fused_decode_step_matches_unfused_forward(self):
+        """decode_step_ws (fused GEMV path) vs decode_one (unfused kernel
+        sequence) on the same prefilled cache: same logits direction."""
+        from adversarial_spec_amd.models import LlamaModel
+        from adversarial_spec_amd.models.config import LlamaConfig
+
+        cfg = LlamaConfig(
+            name="fuse-t", dim=512, n_layers=3, n_heads=4, n_kv_heads=2,
+            ffn_dim=1024, vocab_size=1024, max_seq_len=512,
+            rope_theta=10000.0,
+        )
+        m = LlamaModel(cfg, device=DEV, dtype=torch.bfloat16, seed=3).init_random()
+        toks = torch.arange(1, 33, device=DEV)
+        c1 = m.new_cache(256)
+        m.prefill(toks[:-1], c1)
+        ref = m.decode_one(int(toks[-1]), c1).float()
+
+        c2 = m.new_cache(256)
+        m.prefill(toks[:-1], c2)
+        W = m.new_decode_ws()
+        W.tok_long.fill_(int(toks[-1]))
+        pos_state = torch.tensor([c2.seq_len], dtype=torch.int32, device=DEV)
+        got = m.decode_step_ws(c2, pos_state, 256, W)[0].float()
+        cos = torch.nn.functional.cosine_similarity(
+            ref.unsqueeze(0), got.unsqueeze(0)
+        ).item()
+        assert cos > 0.995, f"fused/unfused decode cosine {cos}"
+        assert int(ref.argmax()) == int(got.argmax())
