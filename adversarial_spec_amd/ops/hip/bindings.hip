@@ -469,6 +469,7 @@ torch::Tensor gemv_norm(torch::Tensor x, torch::Tensor wln, torch::Tensor w,
   TORCH_CHECK((long)x.numel() == (long)K, "gemv_norm: x numel == K");
   TORCH_CHECK((long)wln.numel() == (long)K, "gemv_norm: wln numel == K");
   TORCH_CHECK(K % 8 == 0, "gemv_norm: K % 8 == 0");
+  TORCH_CHECK(K <= 32768, "gemv_norm: LDS xl staging caps K at 32768");
   auto xc = x.contiguous();
   auto lc = wln.contiguous();
   auto y = out_opt.has_value()
@@ -510,6 +511,7 @@ void gemv_gateup_norm(torch::Tensor x, torch::Tensor wln, torch::Tensor w,
   TORCH_CHECK(F2 % 2 == 0 && (long)x.numel() == (long)K && K % 8 == 0);
   TORCH_CHECK((long)wln.numel() == (long)K);
   TORCH_CHECK((long)act.numel() == (long)(F2 / 2));
+  TORCH_CHECK(K <= 32768, "gemv_gateup_norm: LDS xl staging caps K at 32768");
   auto xc = x.contiguous();
   auto lc = wln.contiguous();
   launch_gemv_gateup_norm(uptr(xc), uptr(lc), uptr(w), uptr_mut(act), K,

@@ -187,17 +187,33 @@ extern "C" void launch_gemv_gateup(const ushort_t *x, const ushort_t *w,
 // the TP decode path keeps the unfused sequence (the all-reduce must see
 // the RAW partial projection before the residual add).
 
-// one 16 B chunk of the wln-weighted dot + sum(x^2), f32 math. (The
-// two-kernel sequence rounds the normed activation to bf16 between the
-// kernels; the fused form skips that round — slightly MORE accurate.)
-DEVINL void dot8_norm(const bf16x8 &x, const bf16x8 &l, const bf16x8 &w,
-                      float &acc, float &s2) {
-  const f32x8 xf = unpack8(x), lf = unpack8(l), wf = unpack8(w);
+// Cooperative prologue shared by the norm-fused kernels: the whole block
+// computes xl = x*wln into LDS (bf16) and block-reduces sum(x^2), so the
+// weight-streaming main loop is EXACTLY the plain GEMV's 8-deep dot8_bf16
+// loop with its x reads retargeted at LDS. A first cut that interleaved
+// x/wln VMEM loads + f32 unpack math into the weight loop measured the
+// step ~0.4 ms SLOWER than unfused (the mixed-stream loop loses the
+// weight-load latency hiding; ds_read does not occupy a vmcnt slot, LDS
+// staging restores it). Returns rms; xl[] is barrier-visible on return.
+DEVINL float norm_stage_lds(const ushort_t *__restrict__ x,
+                            const ushort_t *__restrict__ wln,
+                            ushort_t *xl, float *red, int K, float eps) {
+  float s2 = 0.f;
+  const int nc = K / 8;
+  for (int i = threadIdx.x; i < nc; i += 256) {
+    const f32x8 xf = unpack8(((const bf16x8 *)x)[i]);
+    const f32x8 lf = unpack8(((const bf16x8 *)wln)[i]);
+    f32x8 p;
 #pragma unroll
-  for (int i = 0; i < 8; ++i) {
-    acc = fmaf(xf.v[i] * lf.v[i], wf.v[i], acc);
-    s2 = fmaf(xf.v[i], xf.v[i], s2);
+    for (int j = 0; j < 8; ++j) {
+      p.v[j] = xf.v[j] * lf.v[j];
+      s2 = fmaf(xf.v[j], xf.v[j], s2);
+    }
+    ((bf16x8 *)xl)[i] = pack8(p);
   }
+  // block_reduce_sum's first barrier also makes the xl writes visible
+  s2 = block_reduce_sum(s2, red);
+  return rsqrtf(s2 / K + eps);
 }
 
 extern "C" __global__ void __launch_bounds__(256)
@@ -205,40 +221,36 @@ gemv_norm_w32_kernel(const ushort_t *__restrict__ x,
                      const ushort_t *__restrict__ wln,
                      const ushort_t *__restrict__ w,
                      ushort_t *__restrict__ y, int K, int N, float eps) {
+  extern __shared__ ushort_t xl[];  // K bf16: x*wln (unnormalized)
+  __shared__ float red[16];
   const int lane = threadIdx.x & (WAVE - 1);
   const int wid = threadIdx.x / WAVE;
   const int rg = lane >> 5;
   const int sl = lane & 31;
-  const int n = blockIdx.x * 8 + wid * 2 + rg;
-  if (n >= N) return;
+  const int n0 = blockIdx.x * 8 + wid * 2 + rg;
+  // no early return before the barrier: clamp the row, guard the write
+  const int n = n0 < N ? n0 : N - 1;
+
+  const float rms = norm_stage_lds(x, wln, xl, red, K, eps);
 
   const ushort_t *wr = w + (size_t)n * K;
   const int nc = K / 8;
-
-  float acc = 0.f, s2 = 0.f;
+  float acc = 0.f;
   int c = sl;
-  // 4-deep (not 8): three streams in flight (w + L2-resident x, wln)
-  for (; c + 96 < nc; c += 128) {
-    bf16x8 wv[4], xv[4], lv[4];
+  for (; c + 224 < nc; c += 256) {
+    bf16x8 wv[8];
 #pragma unroll
-    for (int u = 0; u < 4; ++u) {
-      wv[u] = ((const bf16x8 *)wr)[c + 32 * u];
-      xv[u] = ((const bf16x8 *)x)[c + 32 * u];
-      lv[u] = ((const bf16x8 *)wln)[c + 32 * u];
-    }
+    for (int u = 0; u < 8; ++u) wv[u] = ((const bf16x8 *)wr)[c + 32 * u];
 #pragma unroll
-    for (int u = 0; u < 4; ++u) dot8_norm(xv[u], lv[u], wv[u], acc, s2);
+    for (int u = 0; u < 8; ++u)
+      acc = dot8_bf16(((const bf16x8 *)xl)[c + 32 * u], wv[u], acc);
   }
   for (; c < nc; c += 32)
-    dot8_norm(((const bf16x8 *)x)[c], ((const bf16x8 *)wln)[c],
-              ((const bf16x8 *)wr)[c], acc, s2);
+    acc = dot8_bf16(((const bf16x8 *)xl)[c], ((const bf16x8 *)wr)[c], acc);
 
 #pragma unroll
-  for (int off = 16; off > 0; off >>= 1) {
-    acc += __shfl_xor(acc, off, WAVE);
-    s2 += __shfl_xor(s2, off, WAVE);
-  }
-  if (sl == 0) y[n] = f32_to_bf16(acc * rsqrtf(s2 / K + eps));
+  for (int off = 16; off > 0; off >>= 1) acc += __shfl_xor(acc, off, WAVE);
+  if (sl == 0 && n0 < N) y[n0] = f32_to_bf16(acc * rms);
 }
 
 // 16-lane-group norm variant for LARGE N (lm_head: final_norm fused)
@@ -247,39 +259,35 @@ gemv_norm_kernel(const ushort_t *__restrict__ x,
                  const ushort_t *__restrict__ wln,
                  const ushort_t *__restrict__ w,
                  ushort_t *__restrict__ y, int K, int N, float eps) {
+  extern __shared__ ushort_t xl[];
+  __shared__ float red[16];
   const int lane = threadIdx.x & (WAVE - 1);
   const int wid = threadIdx.x / WAVE;
   const int rg = lane >> 4;
   const int sl = lane & 15;
-  const int n = blockIdx.x * 16 + wid * 4 + rg;
-  if (n >= N) return;
+  const int n0 = blockIdx.x * 16 + wid * 4 + rg;
+  const int n = n0 < N ? n0 : N - 1;
+
+  const float rms = norm_stage_lds(x, wln, xl, red, K, eps);
 
   const ushort_t *wr = w + (size_t)n * K;
   const int nc = K / 8;
-
-  float acc = 0.f, s2 = 0.f;
+  float acc = 0.f;
   int c = sl;
   for (; c + 48 < nc; c += 64) {
-    bf16x8 wv[4], xv[4], lv[4];
+    bf16x8 wv[4];
 #pragma unroll
-    for (int u = 0; u < 4; ++u) {
-      wv[u] = ((const bf16x8 *)wr)[c + 16 * u];
-      xv[u] = ((const bf16x8 *)x)[c + 16 * u];
-      lv[u] = ((const bf16x8 *)wln)[c + 16 * u];
-    }
+    for (int u = 0; u < 4; ++u) wv[u] = ((const bf16x8 *)wr)[c + 16 * u];
 #pragma unroll
-    for (int u = 0; u < 4; ++u) dot8_norm(xv[u], lv[u], wv[u], acc, s2);
+    for (int u = 0; u < 4; ++u)
+      acc = dot8_bf16(((const bf16x8 *)xl)[c + 16 * u], wv[u], acc);
   }
   for (; c < nc; c += 16)
-    dot8_norm(((const bf16x8 *)x)[c], ((const bf16x8 *)wln)[c],
-              ((const bf16x8 *)wr)[c], acc, s2);
+    acc = dot8_bf16(((const bf16x8 *)xl)[c], ((const bf16x8 *)wr)[c], acc);
 
 #pragma unroll
-  for (int off = 8; off > 0; off >>= 1) {
-    acc += __shfl_xor(acc, off, WAVE);
-    s2 += __shfl_xor(s2, off, WAVE);
-  }
-  if (sl == 0) y[n] = f32_to_bf16(acc * rsqrtf(s2 / K + eps));
+  for (int off = 8; off > 0; off >>= 1) acc += __shfl_xor(acc, off, WAVE);
+  if (sl == 0 && n0 < N) y[n0] = f32_to_bf16(acc * rms);
 }
 
 // residual-add epilogue variant: resid[n] += x @ W[n] (in place).
@@ -317,78 +325,71 @@ gemv_res_w32_kernel(const ushort_t *__restrict__ x,
   if (sl == 0) resid[n] = f32_to_bf16(bf16_to_f32(resid[n]) + acc);
 }
 
-// one 16 B chunk for the norm-fused gate_up pair
-DEVINL void dot8_norm2(const bf16x8 &x, const bf16x8 &l, const bf16x8 &wg,
-                       const bf16x8 &wu, float &ag, float &au, float &s2) {
-  const f32x8 xf = unpack8(x), lf = unpack8(l);
-  const f32x8 gf = unpack8(wg), uf = unpack8(wu);
-#pragma unroll
-  for (int i = 0; i < 8; ++i) {
-    const float xl = xf.v[i] * lf.v[i];
-    ag = fmaf(xl, gf.v[i], ag);
-    au = fmaf(xl, uf.v[i], au);
-    s2 = fmaf(xf.v[i], xf.v[i], s2);
-  }
-}
-
 extern "C" __global__ void __launch_bounds__(256)
 gemv_gateup_norm_kernel(const ushort_t *__restrict__ x,
                         const ushort_t *__restrict__ wln,
                         const ushort_t *__restrict__ w,
                         ushort_t *__restrict__ act, int K, int F, float eps) {
+  extern __shared__ ushort_t xl[];
+  __shared__ float red[16];
   const int lane = threadIdx.x & (WAVE - 1);
   const int wid = threadIdx.x / WAVE;
   const int rg = lane >> 5;
   const int sl = lane & 31;
-  const int n = blockIdx.x * 8 + wid * 2 + rg;
-  if (n >= F) return;
+  const int n0 = blockIdx.x * 8 + wid * 2 + rg;
+  const int n = n0 < F ? n0 : F - 1;
+
+  const float rms = norm_stage_lds(x, wln, xl, red, K, eps);
 
   const ushort_t *wg = w + (size_t)n * K;
   const ushort_t *wu = w + (size_t)(n + F) * K;
   const int nc = K / 8;
 
-  float ag = 0.f, au = 0.f, s2 = 0.f;
+  float ag = 0.f, au = 0.f;
   int c = sl;
   for (; c + 96 < nc; c += 128) {
-    bf16x8 gv[4], uv[4], xv[4], lv[4];
+    bf16x8 gv[4], uv[4];
 #pragma unroll
     for (int u = 0; u < 4; ++u) {
       gv[u] = ((const bf16x8 *)wg)[c + 32 * u];
       uv[u] = ((const bf16x8 *)wu)[c + 32 * u];
-      xv[u] = ((const bf16x8 *)x)[c + 32 * u];
-      lv[u] = ((const bf16x8 *)wln)[c + 32 * u];
     }
 #pragma unroll
-    for (int u = 0; u < 4; ++u)
-      dot8_norm2(xv[u], lv[u], gv[u], uv[u], ag, au, s2);
+    for (int u = 0; u < 4; ++u) {
+      const bf16x8 xv = ((const bf16x8 *)xl)[c + 32 * u];
+      ag = dot8_bf16(xv, gv[u], ag);
+      au = dot8_bf16(xv, uv[u], au);
+    }
   }
-  for (; c < nc; c += 32)
-    dot8_norm2(((const bf16x8 *)x)[c], ((const bf16x8 *)wln)[c],
-               ((const bf16x8 *)wg)[c], ((const bf16x8 *)wu)[c], ag, au, s2);
+  for (; c < nc; c += 32) {
+    const bf16x8 xv = ((const bf16x8 *)xl)[c];
+    ag = dot8_bf16(xv, ((const bf16x8 *)wg)[c], ag);
+    au = dot8_bf16(xv, ((const bf16x8 *)wu)[c], au);
+  }
 
 #pragma unroll
   for (int off = 16; off > 0; off >>= 1) {
     ag += __shfl_xor(ag, off, WAVE);
     au += __shfl_xor(au, off, WAVE);
-    s2 += __shfl_xor(s2, off, WAVE);
   }
-  if (sl == 0) {
-    const float rms = rsqrtf(s2 / K + eps);
+  if (sl == 0 && n0 < F) {
     const float g = ag * rms;
     const float s = g / (1.0f + __expf(-g));  // silu(gate)
-    act[n] = f32_to_bf16(s * (au * rms));
+    act[n0] = f32_to_bf16(s * (au * rms));
   }
 }
 
 extern "C" void launch_gemv_norm(const ushort_t *x, const ushort_t *wln,
                                  const ushort_t *w, ushort_t *y, int K, int N,
                                  float eps, hipStream_t stream) {
+  const size_t lds = (size_t)K * 2;  // xl staging (x*wln as bf16)
   if (N <= 8192)
-    gemv_norm_w32_kernel<<<dim3((N + 7) / 8), 256, 0, stream>>>(x, wln, w, y,
-                                                                K, N, eps);
+    gemv_norm_w32_kernel<<<dim3((N + 7) / 8), 256, lds, stream>>>(x, wln, w,
+                                                                  y, K, N,
+                                                                  eps);
   else
-    gemv_norm_kernel<<<dim3((N + 15) / 16), 256, 0, stream>>>(x, wln, w, y,
-                                                              K, N, eps);
+    gemv_norm_kernel<<<dim3((N + 15) / 16), 256, lds, stream>>>(x, wln, w, y,
+                                                                K, N, eps);
 }
 
 extern "C" void launch_gemv_res(const ushort_t *x, const ushort_t *w,
@@ -401,7 +402,6 @@ extern "C" void launch_gemv_gateup_norm(const ushort_t *x, const ushort_t *wln,
                                         const ushort_t *w, ushort_t *act,
                                         int K, int F, float eps,
                                         hipStream_t stream) {
-  gemv_gateup_norm_kernel<<<dim3((F + 7) / 8), 256, 0, stream>>>(x, wln, w,
-                                                                 act, K, F,
-                                                                 eps);
+  gemv_gateup_norm_kernel<<<dim3((F + 7) / 8), 256, (size_t)K * 2, stream>>>(
+      x, wln, w, act, K, F, eps);
 }
