@@ -163,6 +163,12 @@ class LlamaModel:
         self.lm_head: Optional[torch.Tensor] = None  # [vocab, d]
         self.layers: list[LayerWeights] = []
         self.fp8 = False  # set by quantize_fp8()
+        # decode-attention split-grid block target; 0 = kernel default
+        # (256). The engine sets this from its live-opponent count before
+        # capturing decode graphs: solo decode wants ~512 blocks of
+        # self-occupancy, co-resident opponents share the chip and want
+        # 256 (bf16) / 512 (fp8) — all A/B-measured (profiles).
+        self.split_blocks = 0
         self.layers_q: list = []  # per-layer QuantW mirrors when fp8
         self.lm_head_q = None
 
@@ -402,6 +408,7 @@ class LlamaModel:
                 attn = ops.attn_decode_paged(
                     q[0], cache.k[i], cache.v[i], cache.page_table, seq,
                     self.scale, pos_state=pos_state, identity=cache.identity,
+                    split_blocks=self.split_blocks,
                 ).unsqueeze(0)
             attn_out = proj(attn.reshape(t, h * hd), L, Qd, "wo")
             if self.tp is not None and self.tp.size > 1:
@@ -517,7 +524,7 @@ class LlamaModel:
             ops.attn_decode_paged(
                 q[0], cache.k[i], cache.v[i], cache.page_table,
                 max_seq_bound, self.scale, pos_state=pos_state, out=W.attn,
-                identity=cache.identity,
+                identity=cache.identity, split_blocks=self.split_blocks,
             )
             if fused:
                 ops.gemv_res(W.attn.view(1, h * hd), L.wo, W.resid)

@@ -122,10 +122,13 @@ def attn_decode_paged(
     pos_state: Optional[torch.Tensor] = None,
     out: Optional[torch.Tensor] = None,
     identity: bool = False,
+    split_blocks: int = 0,
 ) -> torch.Tensor:
     """identity=True promises page_table[i] == i (the engine's single-pool
     cache): the split kernel then does pure address math — the per-lane
-    table read otherwise forces a DMA-queue drain per staged tile."""
+    table read otherwise forces a DMA-queue drain per staged tile.
+    split_blocks>0 sets the split-grid block target (the engine passes its
+    concurrency-aware pick; ADVSPEC_SPLIT_BLOCKS still overrides)."""
     if _on_gpu(q):
         import math
 
@@ -135,10 +138,11 @@ def attn_decode_paged(
             # the static max bound sizing the split geometry.
             return _require_hip().attn_decode_paged_ds(
                 q, k_cache, v_cache, page_table, pos_state, seq_len, s,
-                out, identity,
+                out, identity, split_blocks,
             )
         return _require_hip().attn_decode_paged(
-            q, k_cache, v_cache, page_table, seq_len, s, identity)
+            q, k_cache, v_cache, page_table, seq_len, s, identity,
+            split_blocks)
     return torch_ref.attn_decode_paged(q, k_cache, v_cache, page_table, seq_len, scale)
 
 

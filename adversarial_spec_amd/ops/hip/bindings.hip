@@ -355,17 +355,20 @@ torch::Tensor mfma_probe16(torch::Tensor a, torch::Tensor b) {
 // by per-wave SERIAL iterations (shfl-reduce + online-softmax chains), not
 // KV bytes, so occupancy (total blocks) is the lever; ADVSPEC_SPLIT_BLOCKS
 // overrides the target block count for A/B tuning on hardware.
-static void split_geometry(long seq, int kh, int* n_splits, int* split_len) {
-  // Default 256: the production regime is SEVERAL co-resident opponents
-  // (bench A/B under 3-way concurrency: 256-target 0.594 critiques/s vs
-  // 512-target 0.581 — each opponent needs less self-occupancy when the
-  // chip is shared, and the split/combine tails shrink). Solo-latency
-  // workloads can raise it (solo sweep optimum was 512).
-  static const int cap_blocks = [] {
+static void split_geometry(long seq, int kh, int target_blocks,
+                           int* n_splits, int* split_len) {
+  // Target-block precedence: ADVSPEC_SPLIT_BLOCKS env (explicit A/B
+  // override) > per-call target_blocks (the engine picks by its live
+  // concurrency: solo decode wants ~512 blocks of self-occupancy, several
+  // co-resident opponents want 256 — bench A/B under 3-way bf16
+  // concurrency: 256-target 0.594 critiques/s vs 512-target 0.581; fp8
+  // prefers 512) > compiled default 256.
+  static const int env_blocks = [] {
     const char* e = getenv("ADVSPEC_SPLIT_BLOCKS");
-    const int v = e ? atoi(e) : 256;
-    return v > 0 ? v : 256;
+    return e ? atoi(e) : 0;
   }();
+  const int cap_blocks =
+      env_blocks > 0 ? env_blocks : (target_blocks > 0 ? target_blocks : 256);
   const int target = std::max(1, cap_blocks / kh);
   int ns = std::max(1, std::min((int)((seq + 63) / 64), target));
   int sl = (int)((seq + ns - 1) / ns + 63) / 64 * 64;
@@ -376,7 +379,7 @@ static void split_geometry(long seq, int kh, int* n_splits, int* split_len) {
 torch::Tensor attn_decode_paged(torch::Tensor q, torch::Tensor kc,
                                 torch::Tensor vc, torch::Tensor page_table,
                                 int64_t seq_len, double scale,
-                                bool identity) {
+                                bool identity, int64_t split_blocks) {
   CHECK_BF16_CUDA(q);
   CHECK_BF16_CUDA(kc);
   auto qc = q.contiguous();
@@ -389,7 +392,7 @@ torch::Tensor attn_decode_paged(torch::Tensor q, torch::Tensor kc,
   TORCH_CHECK(hd == 32 || hd == 64 || hd == 128);
 
   int n_splits, split_len;
-  split_geometry(seq_len, kh, &n_splits, &split_len);
+  split_geometry(seq_len, kh, (int)split_blocks, &n_splits, &split_len);
   auto out = torch::empty({hq, hd}, qc.options());
   const long khnsg = (long)kh * n_splits * group;
   auto stream = cur_stream();
@@ -645,14 +648,14 @@ torch::Tensor attn_decode_paged_ds(torch::Tensor q, torch::Tensor kc,
                                    torch::Tensor pos_state, int64_t max_seq,
                                    double scale,
                                    c10::optional<torch::Tensor> out_opt,
-                                   bool identity) {
+                                   bool identity, int64_t split_blocks) {
   CHECK_BF16_CUDA(q);
   auto qc = q.contiguous();
   const int hq = qc.size(0), hd = qc.size(1);
   const int page = kc.size(1), kh = kc.size(2);
   const int group = hq / kh;
   int n_splits, split_len;
-  split_geometry(max_seq, kh, &n_splits, &split_len);
+  split_geometry(max_seq, kh, (int)split_blocks, &n_splits, &split_len);
   auto out = out_opt.has_value() ? *out_opt : torch::empty({hq, hd}, qc.options());
   const long khnsg = (long)kh * n_splits * group;
   auto stream = cur_stream();
@@ -731,7 +734,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("mfma_probe16", &mfma_probe16, "MFMA 16x16x32 fragment-map probe");
   m.def("attn_decode_paged", &attn_decode_paged, "paged decode attention",
         py::arg("q"), py::arg("kc"), py::arg("vc"), py::arg("page_table"),
-        py::arg("seq_len"), py::arg("scale"), py::arg("identity") = false);
+        py::arg("seq_len"), py::arg("scale"), py::arg("identity") = false,
+        py::arg("split_blocks") = 0);
   m.def("sample", &sample, "fused temperature softmax sample");
   m.def("sample_to", &sample_to, "async on-device sample into out[idx]");
   m.def("gemv", &gemv, "batch-1 decode GEMV (weight streaming)",
@@ -759,7 +763,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "graph-mode paged decode attention (device pos)",
         py::arg("q"), py::arg("kc"), py::arg("vc"), py::arg("page_table"),
         py::arg("pos_state"), py::arg("max_seq"), py::arg("scale"),
-        py::arg("out") = py::none(), py::arg("identity") = false);
+        py::arg("out") = py::none(), py::arg("identity") = false,
+        py::arg("split_blocks") = 0);
   m.def("sample_state", &sample_state, "graph-mode on-device sampling");
   m.def("bump", &bump, "graph-mode pos/step bump");
   m.def("ws_release", &ws_release,

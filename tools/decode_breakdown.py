@@ -52,6 +52,9 @@ def main():
     W.tok_long.fill_(42)
     pos = torch.tensor([8192], dtype=torch.int32, device="cuda")
     max_seq = cache.max_seq
+    # mirror the engine's concurrency-aware split pick (solo here -> 512)
+    m.split_blocks = eng._pick_split_blocks()
+    print(f"split_blocks target: {m.split_blocks}")
 
     # whole step, eager
     t_step = bench(lambda: m.decode_step_ws(cache, pos, max_seq, W), iters=30)
