@@ -15,7 +15,6 @@ from __future__ import annotations
 import os
 import threading
 import time
-import weakref
 from typing import Any, Optional
 
 import torch
@@ -68,12 +67,6 @@ def _seed_from_name(name: str) -> int:
     return h
 
 
-# Every live LocalEngine on this process (weak: dropped engines leave as
-# soon as they are collected). Drives the concurrency-aware decode-attention
-# split target (_pick_split_blocks).
-_LIVE_ENGINES: "weakref.WeakSet[LocalEngine]" = weakref.WeakSet()
-
-
 class LocalEngine:
     def __init__(self, spec: dict[str, Any], device: Optional[str] = None,
                  tp=None) -> None:
@@ -122,23 +115,18 @@ class LocalEngine:
         self._seed_counter = _seed_from_name(self.name) ^ 0x5EED
         self._cache = None  # persistent KV cache (see _get_cache)
         self._graph_state: Optional[dict] = None  # captured decode graph + state
-        _LIVE_ENGINES.add(self)
 
     def _pick_split_blocks(self) -> int:
         """Decode-attention split-grid target from live concurrency.
 
         The split kernel is occupancy/latency-bound, not bandwidth-bound,
-        so the right block count depends on who else is on the chip
-        (all A/B-measured, profiles/r02_measurements.md): a SOLO decode
-        wants ~512 blocks of self-occupancy; co-resident opponents share
-        the chip and want 256 (bf16) / 512 (fp8 — its faster GEMVs raise
-        the attention share). Baked into the decode graph at capture;
-        ADVSPEC_SPLIT_BLOCKS still overrides inside the kernel dispatch."""
-        n = sum(1 for e in _LIVE_ENGINES
-                if getattr(e, "device", None) == self.device)
-        if n <= 1 or self.model.fp8:
-            return 512
-        return 256
+        so the right block count depends on the decode regime (all
+        A/B-measured, profiles/r02_measurements.md): bf16 wants 256 (solo
+        step 3.772 ms at 256 vs 3.789 at 512; 3-opponent rounds 0.594 vs
+        0.581 critiques/s), fp8 wants 512 (0.779 vs 0.763 — its faster
+        GEMVs raise the attention share). Baked into the decode graph at
+        capture; ADVSPEC_SPLIT_BLOCKS still overrides in the dispatch."""
+        return 512 if self.model.fp8 else 256
 
     def _fit_prompt(self, ids: list[int], reserve: int) -> list[int]:
         """Clamp a prompt into the context window, dropping the middle."""
