@@ -91,12 +91,9 @@ def main() -> int:
         dev_idx = local_rank % torch.cuda.device_count()
         torch.cuda.set_device(dev_idx)
         device = f"cuda:{dev_idx}"
-        # decode-attention split-block target by regime (A/B-measured):
-        # bf16 3-opponent rounds prefer 256 (0.594 vs 0.581), fp8 rounds
-        # prefer 512 (0.779 vs 0.763 — fp8's faster GEMVs raise the
-        # attention share, which wants more self-occupancy)
-        os.environ.setdefault(
-            "ADVSPEC_SPLIT_BLOCKS", "512" if args.dtype == "fp8" else "256")
+        # decode-attention split-block target is picked by the ENGINE per
+        # dtype regime (LocalEngine._pick_split_blocks, A/B-measured);
+        # ADVSPEC_SPLIT_BLOCKS env remains the explicit override.
     else:
         device = "cpu"
         args.model = "tiny"
