@@ -497,21 +497,39 @@ class LlamaModel:
                 return out
             return ops.gemv(x, getattr(L_, name), out=out)
 
-        # norm/residual GEMV fusion (bf16, non-TP): the rmsnorm folds into
-        # the consuming GEMV's prologue and the residual add into the
-        # producing GEMV's epilogue — 65 fewer launches per step (the
-        # decode anatomy's add_rmsnorm x2 row, 0.30 ms/tok, was pure
-        # launch-bound small kernels). TP keeps the unfused sequence: the
-        # all-reduce must see the raw partial projection before the add.
-        fused = (not fp8) and (self.tp is None or self.tp.size == 1)
+        # norm/residual GEMV fusion (non-TP): the rmsnorm folds into the
+        # consuming GEMV's prologue (bf16: LDS-staged x*wln; fp8: fused
+        # quant_norm producing x8/xs) and the residual add into the
+        # producing GEMV's epilogue — 65 (bf16) / ~128 (fp8, incl. the
+        # separate quantize launches) fewer launches per step (the decode
+        # anatomy's add_rmsnorm x2 row was pure launch-bound small
+        # kernels). TP keeps the unfused sequence: the all-reduce must see
+        # the raw partial projection before the add.
+        fused = self.tp is None or self.tp.size == 1
+        # the fp8 fused path needs every projection + lm_head quantized
+        # (mixed schemes from quantize_fp8(projections=...) fall back)
+        fused_f8 = (
+            fused and fp8 and self.lm_head_q is not None
+            and all(
+                q.get(f) is not None for q in self.layers_q
+                for f in self.FP8_PROJECTIONS
+            )
+        )
+
+        fused_bf = fused and not fp8
 
         torch.index_select(self.embed, 0, W.tok_long, out=W.resid)
-        if not fused:
+        if not (fused_bf or fused_f8):
             ops.rmsnorm(W.resid, self.layers[0].attn_norm, c.norm_eps,
                         out=W.normed)
         for i, L in enumerate(self.layers):
             Qd = self.layers_q[i] if fp8 else None
-            if fused:
+            if fused_f8:
+                qw = Qd.get("wqkv")
+                ops.quant_norm_fp8(W.resid, L.attn_norm, W.x8, W.xs,
+                                   c.norm_eps)
+                ops.gemv_fp8_q(W.x8, W.xs, qw.q, qw.s, W.qkv)
+            elif fused_bf:
                 ops.gemv_norm(W.resid, L.attn_norm, L.wqkv, c.norm_eps,
                               out=W.qkv)
             else:
@@ -526,7 +544,17 @@ class LlamaModel:
                 max_seq_bound, self.scale, pos_state=pos_state, out=W.attn,
                 identity=cache.identity, split_blocks=self.split_blocks,
             )
-            if fused:
+            if fused_f8:
+                ops.gemv_fp8_res(W.attn.view(1, h * hd), Qd.get("wo").q,
+                                 Qd.get("wo").s, W.x8, W.xs, W.resid)
+                gw = Qd.get("w_gate_up")
+                ops.quant_norm_fp8(W.resid, L.mlp_norm, W.x8, W.xs,
+                                   c.norm_eps)
+                ops.gemv_fp8_gateup(W.x8, W.xs, gw.q, gw.s, W.act)
+                ops.gemv_fp8_res(W.act, Qd.get("w_down").q,
+                                 Qd.get("w_down").s, W.x8, W.xs, W.resid)
+                continue
+            if fused_bf:
                 ops.gemv_res(W.attn.view(1, h * hd), L.wo, W.resid)
                 ops.gemv_gateup_norm(W.resid, L.mlp_norm, L.w_gate_up,
                                      c.norm_eps, W.act)
@@ -551,7 +579,12 @@ class LlamaModel:
                    else self.final_norm)
             ops.add_rmsnorm(W.resid2, W.mlp_out, nxt, c.norm_eps,
                             out_resid=W.resid, out_y=W.normed)
-        if fused:
+        if fused_f8:
+            ops.quant_norm_fp8(W.resid, self.final_norm, W.x8, W.xs,
+                               c.norm_eps)
+            ops.gemv_fp8_q(W.x8, W.xs, self.lm_head_q.q, self.lm_head_q.s,
+                           W.logits)
+        elif fused_bf:
             ops.gemv_norm(W.resid, self.final_norm, self.lm_head, c.norm_eps,
                           out=W.logits)
         elif self.lm_head_q is not None:

@@ -300,6 +300,64 @@ def gemv_fp8(x: torch.Tensor, w_q: torch.Tensor, w_scale: torch.Tensor,
     return out
 
 
+# ---- fp8 decode fusion (mirror of the bf16 gemv_norm/gemv_res fusion) ----
+
+def quant_norm_fp8(x: torch.Tensor, wln: torch.Tensor, x8: torch.Tensor,
+                   xs: torch.Tensor, eps: float) -> None:
+    """x8, xs[0] = rowwise-e4m3(rmsnorm(x, wln, eps)) in ONE launch: the
+    rms scalar commutes, so one pass yields sum(x^2) AND amax(x*wln)."""
+    if _on_gpu(x):
+        _require_hip().quant_norm_fp8(x, wln, x8, xs, eps)
+        return
+    normed = (x.float() * _rms_ref(x, eps) * wln.float())
+    s = normed.abs().amax().clamp_min(1e-12).item() / 448.0
+    k = x.numel()
+    q = (normed.flatten() / s).to(torch.float8_e4m3fn).view(torch.uint8)
+    x8.flatten()[:k].copy_(q)
+    xs[0] = s
+
+
+def gemv_fp8_q(x8: torch.Tensor, xs: torch.Tensor, w_q: torch.Tensor,
+               w_scale: torch.Tensor, out: torch.Tensor) -> torch.Tensor:
+    """Pre-quantized fp8 decode GEMV (x8/xs from quant_norm_fp8)."""
+    if x8.is_cuda:
+        _require_hip().gemv_fp8_q(x8, xs, w_q, w_scale, out)
+        return out
+    k = w_q.shape[1]
+    xf = x8.flatten()[:k].view(torch.float8_e4m3fn).float() * xs[0]
+    out.copy_((xf @ dequantize_fp8(w_q, w_scale).t()).to(out.dtype)
+              .reshape(out.shape))
+    return out
+
+
+def gemv_fp8_res(x: torch.Tensor, w_q: torch.Tensor, w_scale: torch.Tensor,
+                 x8: torch.Tensor, xs: torch.Tensor,
+                 resid: torch.Tensor) -> torch.Tensor:
+    """fp8 decode fusion: resid += x @ W^T in place (quantize + GEMV with
+    residual epilogue). NOT valid under TP."""
+    if _on_gpu(x):
+        _require_hip().gemv_fp8_res(x, w_q, w_scale, x8, xs, resid)
+        return resid
+    y = gemm_fp8(x, w_q, w_scale)
+    resid.add_(y.to(resid.dtype).reshape(resid.shape))
+    return resid
+
+
+def gemv_fp8_gateup(x8: torch.Tensor, xs: torch.Tensor, w_q: torch.Tensor,
+                    w_scale: torch.Tensor, out: torch.Tensor) -> torch.Tensor:
+    """Pre-quantized fused fp8 gate_up GEMV + SwiGLU."""
+    if x8.is_cuda:
+        _require_hip().gemv_fp8_gateup(x8, xs, w_q, w_scale, out)
+        return out
+    f = w_q.shape[0] // 2
+    k = w_q.shape[1]
+    xf = x8.flatten()[:k].view(torch.float8_e4m3fn).float() * xs[0]
+    gu = xf @ dequantize_fp8(w_q, w_scale).t()
+    out.copy_(torch_ref.swiglu(gu[:f].unsqueeze(0), gu[f:].unsqueeze(0))
+              .to(out.dtype).reshape(out.shape))
+    return out
+
+
 def sample(
     logits: torch.Tensor,
     temperature: float = 0.7,

@@ -64,6 +64,12 @@ void launch_gemm_fp8(const uint8_t*, const float*, const uint8_t*, const float*,
                      ushort_t*, int, int, int, hipStream_t);
 void launch_gemm_fp8_256(const uint8_t*, const float*, const uint8_t*,
                          const float*, ushort_t*, int, int, int, hipStream_t);
+void launch_quant_norm_fp8(const ushort_t*, const ushort_t*, uint8_t*, float*,
+                           int, float, hipStream_t);
+void launch_gemv_fp8_res(const uint8_t*, const float*, const uint8_t*,
+                         const float*, ushort_t*, int, int, hipStream_t);
+void launch_gemv_fp8_gateup(const uint8_t*, const float*, const uint8_t*,
+                            const float*, ushort_t*, int, int, hipStream_t);
 void launch_gemv_fp8(const uint8_t*, const float*, const uint8_t*, const float*,
                      ushort_t*, int, int, hipStream_t);
 void launch_quant_fp8(const ushort_t*, uint8_t*, float*, int, int, hipStream_t);
@@ -626,6 +632,64 @@ void gemv_fp8(torch::Tensor x, torch::Tensor w8, torch::Tensor wsc,
                   uptr_mut(out), K, N, cur_stream());
 }
 
+// fp8 decode fusion (mirror of the bf16 gemv_norm/gemv_res fusion):
+
+// rmsnorm + rowwise e4m3 quantize in one launch: x8 = fp8(rmsnorm(x,wln)),
+// xs = the rowwise scale. Replaces (add_)rmsnorm + quant_fp8 on the fp8
+// decode path.
+void quant_norm_fp8(torch::Tensor x, torch::Tensor wln, torch::Tensor x8,
+                    torch::Tensor xs, double eps) {
+  CHECK_BF16_CUDA(x);
+  CHECK_BF16_CUDA(wln);
+  auto xc = x.contiguous();
+  const int K = xc.numel();
+  TORCH_CHECK((long)wln.numel() == (long)K && K % 8 == 0);
+  TORCH_CHECK(x8.numel() >= K && x8.scalar_type() == at::kByte);
+  TORCH_CHECK(xs.numel() >= 1 && xs.scalar_type() == at::kFloat);
+  launch_quant_norm_fp8(uptr(xc), uptr(wln), x8.data_ptr<uint8_t>(),
+                        xs.data_ptr<float>(), K, (float)eps, cur_stream());
+}
+
+// pre-quantized fp8 GEMV (x8/xs from quant_norm_fp8): out = x8 @ w8^T
+void gemv_fp8_q(torch::Tensor x8, torch::Tensor xs, torch::Tensor w8,
+                torch::Tensor wsc, torch::Tensor out) {
+  const int N = w8.size(0), K = w8.size(1);
+  TORCH_CHECK(x8.numel() >= K && x8.scalar_type() == at::kByte);
+  TORCH_CHECK((long)out.numel() == (long)N);
+  launch_gemv_fp8(x8.data_ptr<uint8_t>(), xs.data_ptr<float>(),
+                  w8.data_ptr<uint8_t>(), wsc.data_ptr<float>(),
+                  uptr_mut(out), K, N, cur_stream());
+}
+
+// quantize + residual-epilogue fp8 GEMV: resid += (x @ w8^T) (in place).
+// NOT valid under TP (the all-reduce needs the raw partial product).
+void gemv_fp8_res(torch::Tensor x, torch::Tensor w8, torch::Tensor wsc,
+                  torch::Tensor x8, torch::Tensor xs, torch::Tensor resid) {
+  CHECK_BF16_CUDA(x);
+  CHECK_BF16_CUDA(resid);
+  auto xc = x.contiguous();
+  const int K = xc.numel(), N = w8.size(0);
+  TORCH_CHECK((long)resid.numel() == (long)N && resid.is_contiguous());
+  TORCH_CHECK(N <= 8192, "gemv_fp8_res: w32 kernel only (hidden <= 8192)");
+  launch_quant_fp8(uptr(xc), x8.data_ptr<uint8_t>(), xs.data_ptr<float>(),
+                   1, K, cur_stream());
+  launch_gemv_fp8_res(x8.data_ptr<uint8_t>(), xs.data_ptr<float>(),
+                      w8.data_ptr<uint8_t>(), wsc.data_ptr<float>(),
+                      uptr_mut(resid), K, N, cur_stream());
+}
+
+// pre-quantized fused fp8 gate_up GEMV + SwiGLU
+void gemv_fp8_gateup(torch::Tensor x8, torch::Tensor xs, torch::Tensor w8,
+                     torch::Tensor wsc, torch::Tensor act) {
+  CHECK_BF16_CUDA(act);
+  const int K = w8.size(1), F2 = w8.size(0);
+  TORCH_CHECK(F2 % 2 == 0 && x8.numel() >= K);
+  TORCH_CHECK((long)act.numel() == (long)(F2 / 2));
+  launch_gemv_fp8_gateup(x8.data_ptr<uint8_t>(), xs.data_ptr<float>(),
+                         w8.data_ptr<uint8_t>(), wsc.data_ptr<float>(),
+                         uptr_mut(act), K, F2 / 2, cur_stream());
+}
+
 int64_t sample(torch::Tensor logits, double temp, double top_p, int64_t seed) {
   CHECK_BF16_CUDA(logits);
   TORCH_CHECK(top_p >= 1.0, "kernel sample handles top_p == 1 (nucleus is a cold path)");
@@ -751,6 +815,13 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("gemm_variant", &gemm_variant, "force GEMM kernel 128/256 (A/B)");
   m.def("gemm_fp8", &gemm_fp8, "fp8 e4m3 MFMA GEMM (rowwise scales)");
   m.def("gemv_fp8", &gemv_fp8, "fp8 decode GEMV (rowwise scales)");
+  m.def("quant_norm_fp8", &quant_norm_fp8,
+        "fused rmsnorm + rowwise e4m3 quantize (fp8 decode fusion)");
+  m.def("gemv_fp8_q", &gemv_fp8_q, "pre-quantized fp8 decode GEMV");
+  m.def("gemv_fp8_res", &gemv_fp8_res,
+        "fp8 GEMV with residual-add epilogue (fp8 decode fusion)");
+  m.def("gemv_fp8_gateup", &gemv_fp8_gateup,
+        "pre-quantized fp8 gate_up GEMV + SwiGLU (fp8 decode fusion)");
   m.def("quant_fp8", &quant_fp8, "rowwise bf16 -> e4m3 quantizer");
   m.def("mfma_rate", &mfma_rate, "MFMA issue-rate microbench (bf16/fp8)");
   m.def("rope_inplace_ds", &rope_inplace_ds, "graph-mode RoPE (device pos)");

@@ -429,3 +429,168 @@ extern "C" void launch_quant_fp8(const ushort_t *x, uint8_t *q, float *scale,
                                  int M, int K, hipStream_t stream) {
   quant_fp8_row_kernel<<<dim3(M), 256, 0, stream>>>(x, q, scale, K);
 }
+
+// ---------------------------------------------------------------------------
+// fp8 decode norm/residual fusion (mirror of the bf16 gemv.hip fusion):
+//   quant_norm: rmsnorm + rowwise e4m3 quantize in ONE kernel — the rms
+//   scalar commutes, so one pass accumulates sum(x^2) AND amax(x*wln);
+//   the quantizer scale is amax*rms/448 and the second pass writes
+//   fp8(x*wln*rms/scale). Replaces the separate (add_)rmsnorm + quant
+//   launches between every pair of fp8 decode GEMVs.
+// ---------------------------------------------------------------------------
+
+extern "C" __global__ void __launch_bounds__(256)
+quant_norm_fp8_kernel(const ushort_t *__restrict__ x,
+                      const ushort_t *__restrict__ wln,
+                      uint8_t *__restrict__ q, float *__restrict__ scale,
+                      int K, float eps) {
+  __shared__ float scratch[16];
+  const int nv = K / 8;
+
+  float amax = 0.f, s2 = 0.f;
+  for (int i = threadIdx.x; i < nv; i += blockDim.x) {
+    const f32x8 v = unpack8(((const bf16x8 *)x)[i]);
+    const f32x8 l = unpack8(((const bf16x8 *)wln)[i]);
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      amax = fmaxf(amax, fabsf(v.v[j] * l.v[j]));
+      s2 = fmaf(v.v[j], v.v[j], s2);
+    }
+  }
+  // sequential reuse of one scratch is safe: each reduce barriers before
+  // and after its scratch reads
+  amax = block_reduce_max(amax, scratch);
+  s2 = block_reduce_sum(s2, scratch);
+
+  const float rms = rsqrtf(s2 / K + eps);
+  const float am = amax * rms;
+  const float s = (am > 0.f) ? am / 448.f : 1.f;
+  const float inv = rms / s;  // one fused multiplier: x*wln*rms -> /s
+  if (threadIdx.x == 0) scale[0] = s;
+
+  for (int i = threadIdx.x; i < nv; i += blockDim.x) {
+    const f32x8 v = unpack8(((const bf16x8 *)x)[i]);
+    const f32x8 l = unpack8(((const bf16x8 *)wln)[i]);
+#pragma unroll
+    for (int j = 0; j < 8; ++j)
+      q[i * 8 + j] = f32_to_fp8(v.v[j] * l.v[j] * inv);
+  }
+}
+
+// residual-add epilogue fp8 GEMV: resid[n] += (x8 . w8[n]) * xs * wsc[n]
+extern "C" __global__ void __launch_bounds__(256)
+gemv_fp8_res_w32_kernel(const uint8_t *__restrict__ x,
+                        const float *__restrict__ xs,
+                        const uint8_t *__restrict__ w,
+                        const float *__restrict__ wsc,
+                        ushort_t *__restrict__ resid, int K, int N) {
+  const int lane = threadIdx.x & (WAVE - 1);
+  const int wid = threadIdx.x / WAVE;
+  const int rg = lane >> 5;
+  const int sl = lane & 31;
+  const int n = blockIdx.x * 8 + wid * 2 + rg;
+  if (n >= N) return;
+
+  const uint8_t *wr = w + (size_t)n * K;
+  const int nc = K / 16;
+  const float r0 = bf16_to_f32(resid[n]);  // prefetch (tail-latency)
+  const float ws_n = wsc[n];
+
+  float acc = 0.f;
+  int c = sl;
+  for (; c + 96 < nc; c += 128) {
+    uint4 wv[4], xv[4];
+#pragma unroll
+    for (int u = 0; u < 4; ++u) {
+      wv[u] = *(const uint4 *)(wr + (size_t)(c + 32 * u) * 16);
+      xv[u] = *(const uint4 *)(x + (size_t)(c + 32 * u) * 16);
+    }
+#pragma unroll
+    for (int u = 0; u < 4; ++u) acc += dot16_fp8(xv[u], wv[u]);
+  }
+  for (; c < nc; c += 32) {
+    const uint4 wv = *(const uint4 *)(wr + (size_t)c * 16);
+    const uint4 xv = *(const uint4 *)(x + (size_t)c * 16);
+    acc += dot16_fp8(xv, wv);
+  }
+
+#pragma unroll
+  for (int off = 16; off > 0; off >>= 1) acc += __shfl_xor(acc, off, WAVE);
+  if (sl == 0) resid[n] = f32_to_bf16(r0 + acc * xs[0] * ws_n);
+}
+
+// fused fp8 gate_up GEMV + SwiGLU: act[n] = silu(g)*u with
+// g = (x8 . wg[n])*xs*wsc[n], u = (x8 . wu[n])*xs*wsc[n+F]
+extern "C" __global__ void __launch_bounds__(256)
+gemv_fp8_gateup_kernel(const uint8_t *__restrict__ x,
+                       const float *__restrict__ xs,
+                       const uint8_t *__restrict__ w,
+                       const float *__restrict__ wsc,
+                       ushort_t *__restrict__ act, int K, int F) {
+  const int lane = threadIdx.x & (WAVE - 1);
+  const int wid = threadIdx.x / WAVE;
+  const int rg = lane >> 5;
+  const int sl = lane & 31;
+  const int n = blockIdx.x * 8 + wid * 2 + rg;
+  if (n >= F) return;
+
+  const uint8_t *wg = w + (size_t)n * K;
+  const uint8_t *wu = w + (size_t)(n + F) * K;
+  const int nc = K / 16;
+
+  float ag = 0.f, au = 0.f;
+  int c = sl;
+  for (; c + 96 < nc; c += 128) {
+    uint4 gv[4], uv[4], xv[4];
+#pragma unroll
+    for (int u = 0; u < 4; ++u) {
+      gv[u] = *(const uint4 *)(wg + (size_t)(c + 32 * u) * 16);
+      uv[u] = *(const uint4 *)(wu + (size_t)(c + 32 * u) * 16);
+      xv[u] = *(const uint4 *)(x + (size_t)(c + 32 * u) * 16);
+    }
+#pragma unroll
+    for (int u = 0; u < 4; ++u) {
+      ag += dot16_fp8(xv[u], gv[u]);
+      au += dot16_fp8(xv[u], uv[u]);
+    }
+  }
+  for (; c < nc; c += 32) {
+    const uint4 xv = *(const uint4 *)(x + (size_t)c * 16);
+    ag += dot16_fp8(xv, *(const uint4 *)(wg + (size_t)c * 16));
+    au += dot16_fp8(xv, *(const uint4 *)(wu + (size_t)c * 16));
+  }
+
+#pragma unroll
+  for (int off = 16; off > 0; off >>= 1) {
+    ag += __shfl_xor(ag, off, WAVE);
+    au += __shfl_xor(au, off, WAVE);
+  }
+  if (sl == 0) {
+    const float g = ag * xs[0] * wsc[n];
+    const float u = au * xs[0] * wsc[n + F];
+    const float s = g / (1.0f + __expf(-g));  // silu(gate)
+    act[n] = f32_to_bf16(s * u);
+  }
+}
+
+extern "C" void launch_quant_norm_fp8(const ushort_t *x, const ushort_t *wln,
+                                      uint8_t *q, float *scale, int K,
+                                      float eps, hipStream_t stream) {
+  quant_norm_fp8_kernel<<<dim3(1), 256, 0, stream>>>(x, wln, q, scale, K, eps);
+}
+
+extern "C" void launch_gemv_fp8_res(const uint8_t *x, const float *xs,
+                                    const uint8_t *w, const float *wsc,
+                                    ushort_t *resid, int K, int N,
+                                    hipStream_t stream) {
+  gemv_fp8_res_w32_kernel<<<dim3((N + 7) / 8), 256, 0, stream>>>(x, xs, w, wsc,
+                                                                 resid, K, N);
+}
+
+extern "C" void launch_gemv_fp8_gateup(const uint8_t *x, const float *xs,
+                                       const uint8_t *w, const float *wsc,
+                                       ushort_t *act, int K, int F,
+                                       hipStream_t stream) {
+  gemv_fp8_gateup_kernel<<<dim3((F + 7) / 8), 256, 0, stream>>>(x, xs, w, wsc,
+                                                                act, K, F);
+}

@@ -557,6 +557,57 @@ class TestGemvNormResFusion:
         want = torch_ref.swiglu(gu[:, :f], gu[:, f:])
         _assert_close(act, want, atol=5e-2, name="gemv_gateup_norm")
 
+    def test_fp8_quant_norm(self):
+        """quant_norm_fp8 == rmsnorm then rowwise-e4m3 quantize."""
+        k = 4096
+        x = _bf(torch.randn(1, k)).to(DEV)
+        wln = _bf(torch.rand(k) + 0.5).to(DEV)
+        x8 = torch.empty(1, k, dtype=torch.uint8, device=DEV)
+        xs = torch.empty(1, dtype=torch.float32, device=DEV)
+        ops.quant_norm_fp8(x, wln, x8, xs, self.EPS)
+        deq = (x8.view(torch.float8_e4m3fn).float() * xs).cpu()
+        want = self._norm_ref(x, wln)
+        # e4m3 has ~2^-3 relative mantissa step at the top of each binade
+        err = (deq - want).abs().max().item()
+        scale = want.abs().max().item()
+        assert err <= scale * 0.08, f"quant_norm err {err} vs amax {scale}"
+
+    def test_fp8_fused_gemv_chain(self):
+        """quant_norm -> gemv_fp8_q / gemv_fp8_gateup / gemv_fp8_res vs the
+        fp32 reference of the same quantized weights."""
+        k, n, f = 2048, 1024, 1536
+        x = _bf(torch.randn(1, k)).to(DEV)
+        wln = _bf(torch.rand(k) + 0.5).to(DEV)
+        w = _bf(torch.randn(n, k) * 0.02).to(DEV)
+        wgu = _bf(torch.randn(2 * f, k) * 0.02).to(DEV)
+        wq, wsc = ops.quantize_fp8_rowwise(w)
+        gq, gsc = ops.quantize_fp8_rowwise(wgu)
+        x8 = torch.empty(1, k, dtype=torch.uint8, device=DEV)
+        xs = torch.empty(1, dtype=torch.float32, device=DEV)
+        ops.quant_norm_fp8(x, wln, x8, xs, self.EPS)
+        xref = (x8.view(torch.float8_e4m3fn).float() * xs).cpu()
+
+        out = torch.empty(1, n, dtype=torch.bfloat16, device=DEV)
+        ops.gemv_fp8_q(x8, xs, wq, wsc, out)
+        want = xref @ ops.dequantize_fp8(wq.cpu(), wsc.cpu()).t()
+        _assert_close(out, want, atol=1.5e-1, name="gemv_fp8_q")
+
+        act = torch.empty(1, f, dtype=torch.bfloat16, device=DEV)
+        ops.gemv_fp8_gateup(x8, xs, gq, gsc, act)
+        gu = xref @ ops.dequantize_fp8(gq.cpu(), gsc.cpu()).t()
+        want_act = torch_ref.swiglu(gu[:, :f], gu[:, f:])
+        _assert_close(act, want_act, atol=1e-1, name="gemv_fp8_gateup")
+
+        resid = _bf(torch.randn(1, n)).to(DEV)
+        xin = _bf(torch.randn(1, k)).to(DEV)
+        xin8 = (xin.float() / (xin.float().abs().amax() / 448.0)
+                ).to(torch.float8_e4m3fn).float().cpu() * (
+            xin.float().abs().amax().cpu() / 448.0)
+        want_res = resid.float().cpu() + \
+            xin8 @ ops.dequantize_fp8(wq.cpu(), wsc.cpu()).t()
+        ops.gemv_fp8_res(xin, wq, wsc, x8, xs, resid)
+        _assert_close(resid, want_res, atol=1.5e-1, name="gemv_fp8_res")
+
     def test_fused_decode_step_matches_unfused_forward(self):
         """decode_step_ws (fused GEMV path) vs decode_one (unfused kernel
         sequence) on the same prefilled cache: same logits direction."""
