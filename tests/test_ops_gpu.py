@@ -636,3 +636,33 @@ class TestGemvNormResFusion:
         ).item()
         assert cos > 0.995, f"fused/unfused decode cosine {cos}"
         assert int(ref.argmax()) == int(got.argmax())
+
+    def test_fused_fp8_decode_step_matches_unfused_forward(self):
+        """fp8 decode_step_ws (quant_norm/res/gateup fusion) vs the
+        unfused fp8 decode_one kernel sequence."""
+        from adversarial_spec_amd.models import LlamaModel
+        from adversarial_spec_amd.models.config import LlamaConfig
+
+        cfg = LlamaConfig(
+            name="fuse-f8", dim=512, n_layers=3, n_heads=4, n_kv_heads=2,
+            ffn_dim=1024, vocab_size=1024, max_seq_len=512,
+            rope_theta=10000.0,
+        )
+        m = LlamaModel(cfg, device=DEV, dtype=torch.bfloat16, seed=5)
+        m.init_random().quantize_fp8()
+        toks = torch.arange(1, 33, device=DEV)
+        c1 = m.new_cache(256)
+        m.prefill(toks[:-1], c1)
+        ref = m.decode_one(int(toks[-1]), c1).float()
+
+        c2 = m.new_cache(256)
+        m.prefill(toks[:-1], c2)
+        W = m.new_decode_ws()
+        W.tok_long.fill_(int(toks[-1]))
+        pos_state = torch.tensor([c2.seq_len], dtype=torch.int32, device=DEV)
+        got = m.decode_step_ws(c2, pos_state, 256, W)[0].float()
+        cos = torch.nn.functional.cosine_similarity(
+            ref.unsqueeze(0), got.unsqueeze(0)
+        ).item()
+        assert cos > 0.99, f"fp8 fused/unfused decode cosine {cos}"
+        assert int(ref.argmax()) == int(got.argmax())
