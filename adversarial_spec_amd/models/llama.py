@@ -526,9 +526,8 @@ class LlamaModel:
             Qd = self.layers_q[i] if fp8 else None
             if fused_f8:
                 qw = Qd.get("wqkv")
-                ops.quant_norm_fp8(W.resid, L.attn_norm, W.x8, W.xs,
-                                   c.norm_eps)
-                ops.gemv_fp8_q(W.x8, W.xs, qw.q, qw.s, W.qkv)
+                ops.gemv_fp8_norm(W.resid, L.attn_norm, qw.q, qw.s,
+                                  c.norm_eps, W.qkv)
             elif fused_bf:
                 ops.gemv_norm(W.resid, L.attn_norm, L.wqkv, c.norm_eps,
                               out=W.qkv)
@@ -545,14 +544,13 @@ class LlamaModel:
                 identity=cache.identity, split_blocks=self.split_blocks,
             )
             if fused_f8:
-                ops.gemv_fp8_res(W.attn.view(1, h * hd), Qd.get("wo").q,
-                                 Qd.get("wo").s, W.x8, W.xs, W.resid)
+                ops.gemv_fp8_resl(W.attn.view(1, h * hd), Qd.get("wo").q,
+                                  Qd.get("wo").s, W.resid)
                 gw = Qd.get("w_gate_up")
-                ops.quant_norm_fp8(W.resid, L.mlp_norm, W.x8, W.xs,
-                                   c.norm_eps)
-                ops.gemv_fp8_gateup(W.x8, W.xs, gw.q, gw.s, W.act)
-                ops.gemv_fp8_res(W.act, Qd.get("w_down").q,
-                                 Qd.get("w_down").s, W.x8, W.xs, W.resid)
+                ops.gemv_fp8_gateup_norm(W.resid, L.mlp_norm, gw.q, gw.s,
+                                         c.norm_eps, W.act)
+                ops.gemv_fp8_resl(W.act, Qd.get("w_down").q,
+                                  Qd.get("w_down").s, W.resid)
                 continue
             if fused_bf:
                 ops.gemv_res(W.attn.view(1, h * hd), L.wo, W.resid)
@@ -580,10 +578,8 @@ class LlamaModel:
             ops.add_rmsnorm(W.resid2, W.mlp_out, nxt, c.norm_eps,
                             out_resid=W.resid, out_y=W.normed)
         if fused_f8:
-            ops.quant_norm_fp8(W.resid, self.final_norm, W.x8, W.xs,
-                               c.norm_eps)
-            ops.gemv_fp8_q(W.x8, W.xs, self.lm_head_q.q, self.lm_head_q.s,
-                           W.logits)
+            ops.gemv_fp8_norm(W.resid, self.final_norm, self.lm_head_q.q,
+                              self.lm_head_q.s, c.norm_eps, W.logits)
         elif fused_bf:
             ops.gemv_norm(W.resid, self.final_norm, self.lm_head, c.norm_eps,
                           out=W.logits)

@@ -358,6 +358,52 @@ def gemv_fp8_gateup(x8: torch.Tensor, xs: torch.Tensor, w_q: torch.Tensor,
     return out
 
 
+# LDS-staged single-launch fp8 fused GEMVs (each GEMV block quantizes its
+# own activation copy into LDS; removes the serial 1-block quant kernels):
+
+def gemv_fp8_norm(x: torch.Tensor, wln: torch.Tensor, w_q: torch.Tensor,
+                  w_scale: torch.Tensor, eps: float,
+                  out: torch.Tensor) -> torch.Tensor:
+    """out = rmsnorm(x, wln, eps) @ W^T in ONE launch (fp8 weights)."""
+    if _on_gpu(x):
+        _require_hip().gemv_fp8_norm(x, wln, w_q, w_scale, eps, out)
+        return out
+    normed = (x.float() * _rms_ref(x, eps) * wln.float())
+    s = float(normed.abs().amax().clamp_min(1e-12)) / 448.0
+    n8 = (normed / s).to(torch.float8_e4m3fn).float() * s
+    out.copy_((n8 @ dequantize_fp8(w_q, w_scale).t()).to(out.dtype)
+              .reshape(out.shape))
+    return out
+
+
+def gemv_fp8_resl(x: torch.Tensor, w_q: torch.Tensor, w_scale: torch.Tensor,
+                  resid: torch.Tensor) -> torch.Tensor:
+    """resid += x @ W^T in place, one launch (fp8 weights, x quantized
+    in-kernel). NOT valid under TP."""
+    if _on_gpu(x):
+        _require_hip().gemv_fp8_resl(x, w_q, w_scale, resid)
+        return resid
+    resid.add_(gemm_fp8(x, w_q, w_scale).to(resid.dtype).reshape(resid.shape))
+    return resid
+
+
+def gemv_fp8_gateup_norm(x: torch.Tensor, wln: torch.Tensor,
+                         w_q: torch.Tensor, w_scale: torch.Tensor,
+                         eps: float, out: torch.Tensor) -> torch.Tensor:
+    """out = swiglu(rmsnorm(x, wln) @ [Wg|Wu]^T), one launch."""
+    if _on_gpu(x):
+        _require_hip().gemv_fp8_gateup_norm(x, wln, w_q, w_scale, eps, out)
+        return out
+    f = w_q.shape[0] // 2
+    normed = (x.float() * _rms_ref(x, eps) * wln.float())
+    s = float(normed.abs().amax().clamp_min(1e-12)) / 448.0
+    n8 = (normed / s).to(torch.float8_e4m3fn).float() * s
+    gu = n8 @ dequantize_fp8(w_q, w_scale).t()
+    out.copy_(torch_ref.swiglu(gu[..., :f], gu[..., f:]).to(out.dtype)
+              .reshape(out.shape))
+    return out
+
+
 def sample(
     logits: torch.Tensor,
     temperature: float = 0.7,

@@ -66,6 +66,14 @@ void launch_gemm_fp8_256(const uint8_t*, const float*, const uint8_t*,
                          const float*, ushort_t*, int, int, int, hipStream_t);
 void launch_quant_norm_fp8(const ushort_t*, const ushort_t*, uint8_t*, float*,
                            int, float, hipStream_t);
+void launch_gemv_fp8_norm(const ushort_t*, const ushort_t*, const uint8_t*,
+                          const float*, ushort_t*, int, int, float,
+                          hipStream_t);
+void launch_gemv_fp8_resl(const ushort_t*, const uint8_t*, const float*,
+                          ushort_t*, int, int, hipStream_t);
+void launch_gemv_fp8_gateup_norm(const ushort_t*, const ushort_t*,
+                                 const uint8_t*, const float*, ushort_t*,
+                                 int, int, float, hipStream_t);
 void launch_gemv_fp8_res(const uint8_t*, const float*, const uint8_t*,
                          const float*, ushort_t*, int, int, hipStream_t);
 void launch_gemv_fp8_gateup(const uint8_t*, const float*, const uint8_t*,
@@ -690,6 +698,61 @@ void gemv_fp8_gateup(torch::Tensor x8, torch::Tensor xs, torch::Tensor w8,
                          uptr_mut(act), K, F2 / 2, cur_stream());
 }
 
+// LDS-staged single-launch fp8 fused GEMVs (each block quantizes its own
+// activation copy into LDS — removes the serial 1-block quant kernels
+// from the decode critical path):
+
+// y = rmsnorm(x, wln) @ w8^T (quantize + GEMV fused)
+void gemv_fp8_norm(torch::Tensor x, torch::Tensor wln, torch::Tensor w8,
+                   torch::Tensor wsc, double eps, torch::Tensor out) {
+  CHECK_BF16_CUDA(x);
+  CHECK_BF16_CUDA(wln);
+  auto xc = x.contiguous();
+  const int N = w8.size(0), K = w8.size(1);
+  TORCH_CHECK((long)xc.numel() == (long)K && K % 16 == 0);
+  TORCH_CHECK((long)wln.numel() == (long)K);
+  TORCH_CHECK((long)out.numel() == (long)N);
+  TORCH_CHECK(K <= 65536, "gemv_fp8_norm: LDS staging caps K at 65536");
+  launch_gemv_fp8_norm(uptr(xc), uptr(wln), w8.data_ptr<uint8_t>(),
+                       wsc.data_ptr<float>(), uptr_mut(out), K, N,
+                       (float)eps, cur_stream());
+}
+
+// resid += x @ w8^T (quantize + GEMV + residual epilogue, in place).
+// NOT valid under TP (the all-reduce needs the raw partial product).
+void gemv_fp8_resl(torch::Tensor x, torch::Tensor w8, torch::Tensor wsc,
+                   torch::Tensor resid) {
+  CHECK_BF16_CUDA(x);
+  CHECK_BF16_CUDA(resid);
+  auto xc = x.contiguous();
+  const int N = w8.size(0), K = w8.size(1);
+  TORCH_CHECK((long)xc.numel() == (long)K && K % 16 == 0);
+  TORCH_CHECK((long)resid.numel() == (long)N && resid.is_contiguous());
+  TORCH_CHECK(N <= 8192, "gemv_fp8_resl: w32 kernel only (hidden <= 8192)");
+  TORCH_CHECK(K <= 65536, "gemv_fp8_resl: LDS staging caps K at 65536");
+  launch_gemv_fp8_resl(uptr(xc), w8.data_ptr<uint8_t>(),
+                       wsc.data_ptr<float>(), uptr_mut(resid), K, N,
+                       cur_stream());
+}
+
+// act = swiglu(rmsnorm(x, wln) @ [Wg|Wu]^T) — fp8 MLP front half, 1 launch
+void gemv_fp8_gateup_norm(torch::Tensor x, torch::Tensor wln,
+                          torch::Tensor w8, torch::Tensor wsc, double eps,
+                          torch::Tensor act) {
+  CHECK_BF16_CUDA(x);
+  CHECK_BF16_CUDA(wln);
+  CHECK_BF16_CUDA(act);
+  auto xc = x.contiguous();
+  const int K = w8.size(1), F2 = w8.size(0);
+  TORCH_CHECK(F2 % 2 == 0 && (long)xc.numel() == (long)K && K % 16 == 0);
+  TORCH_CHECK((long)wln.numel() == (long)K);
+  TORCH_CHECK((long)act.numel() == (long)(F2 / 2));
+  TORCH_CHECK(K <= 65536, "gemv_fp8_gateup_norm: LDS caps K at 65536");
+  launch_gemv_fp8_gateup_norm(uptr(xc), uptr(wln), w8.data_ptr<uint8_t>(),
+                              wsc.data_ptr<float>(), uptr_mut(act), K,
+                              F2 / 2, (float)eps, cur_stream());
+}
+
 int64_t sample(torch::Tensor logits, double temp, double top_p, int64_t seed) {
   CHECK_BF16_CUDA(logits);
   TORCH_CHECK(top_p >= 1.0, "kernel sample handles top_p == 1 (nucleus is a cold path)");
@@ -822,6 +885,12 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "fp8 GEMV with residual-add epilogue (fp8 decode fusion)");
   m.def("gemv_fp8_gateup", &gemv_fp8_gateup,
         "pre-quantized fp8 gate_up GEMV + SwiGLU (fp8 decode fusion)");
+  m.def("gemv_fp8_norm", &gemv_fp8_norm,
+        "rmsnorm + quantize + fp8 GEMV, one launch (LDS-staged)");
+  m.def("gemv_fp8_resl", &gemv_fp8_resl,
+        "quantize + fp8 GEMV + residual epilogue, one launch (LDS-staged)");
+  m.def("gemv_fp8_gateup_norm", &gemv_fp8_gateup_norm,
+        "rmsnorm + quantize + fp8 gate_up GEMV + SwiGLU, one launch");
   m.def("quant_fp8", &quant_fp8, "rowwise bf16 -> e4m3 quantizer");
   m.def("mfma_rate", &mfma_rate, "MFMA issue-rate microbench (bf16/fp8)");
   m.def("rope_inplace_ds", &rope_inplace_ds, "graph-mode RoPE (device pos)");

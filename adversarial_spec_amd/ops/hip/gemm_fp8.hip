@@ -573,6 +573,249 @@ gemv_fp8_gateup_kernel(const uint8_t *__restrict__ x,
   }
 }
 
+// ---------------------------------------------------------------------------
+// LDS-staged fp8 fused GEMVs: the 1-block quant kernels above are SERIAL
+// latency on the decode critical path (~3-4 us each, 3+/layer), exactly
+// the launch-bound pattern the bf16 fusion removed. Here every GEMV block
+// quantizes its own copy of the activation into LDS (redundant VALU —
+// idle at the weight-streaming roofline) and dots from ds_reads, so the
+// whole layer is one kernel per projection again.
+// ---------------------------------------------------------------------------
+
+// cooperative stage: xl[] = e4m3(x * [wln *] rms / s); returns the
+// dequant scale s. One barrier after the writes (the reduce's barriers
+// cover only the first pass).
+DEVINL float fp8norm_stage(const ushort_t *__restrict__ x,
+                           const ushort_t *__restrict__ wln, uint8_t *xl,
+                           float *red, int K, float eps, bool norm) {
+  const int nv = K / 8;
+  float amax = 0.f, s2 = 0.f;
+  for (int i = threadIdx.x; i < nv; i += blockDim.x) {
+    const f32x8 v = unpack8(((const bf16x8 *)x)[i]);
+    f32x8 l;
+    if (norm) l = unpack8(((const bf16x8 *)wln)[i]);
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      const float xv = norm ? v.v[j] * l.v[j] : v.v[j];
+      amax = fmaxf(amax, fabsf(xv));
+      if (norm) s2 = fmaf(v.v[j], v.v[j], s2);
+    }
+  }
+  amax = block_reduce_max(amax, red);
+  float mult = 1.f;  // applied before quantize
+  if (norm) {
+    s2 = block_reduce_sum(s2, red);
+    mult = rsqrtf(s2 / K + eps);
+  }
+  const float am = amax * mult;
+  const float s = (am > 0.f) ? am / 448.f : 1.f;
+  const float inv = mult / s;
+  for (int i = threadIdx.x; i < nv; i += blockDim.x) {
+    const f32x8 v = unpack8(((const bf16x8 *)x)[i]);
+    f32x8 l;
+    if (norm) l = unpack8(((const bf16x8 *)wln)[i]);
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      const float xv = norm ? v.v[j] * l.v[j] : v.v[j];
+      xl[i * 8 + j] = f32_to_fp8(xv * inv);
+    }
+  }
+  __syncthreads();  // xl visible to the dot loops
+  return s;
+}
+
+// fp8 w32 dot loop against LDS-staged activation (shared by the fused
+// kernels): 4-deep weight stream, uint4 ds_reads for x.
+DEVINL float fp8_dot_lds_w32(const uint8_t *xl, const uint8_t *wr, int nc,
+                             int sl) {
+  float acc = 0.f;
+  int c = sl;
+  for (; c + 96 < nc; c += 128) {
+    uint4 wv[4];
+#pragma unroll
+    for (int u = 0; u < 4; ++u)
+      wv[u] = *(const uint4 *)(wr + (size_t)(c + 32 * u) * 16);
+#pragma unroll
+    for (int u = 0; u < 4; ++u)
+      acc += dot16_fp8(*(const uint4 *)(xl + (size_t)(c + 32 * u) * 16),
+                       wv[u]);
+  }
+  for (; c < nc; c += 32)
+    acc += dot16_fp8(*(const uint4 *)(xl + (size_t)c * 16),
+                     *(const uint4 *)(wr + (size_t)c * 16));
+  return acc;
+}
+
+// y = (rmsnorm(x,wln) @ w8^T) — rmsnorm + quantize + GEMV in ONE launch
+extern "C" __global__ void __launch_bounds__(256)
+gemv_fp8_norm_w32_kernel(const ushort_t *__restrict__ x,
+                         const ushort_t *__restrict__ wln,
+                         const uint8_t *__restrict__ w,
+                         const float *__restrict__ wsc,
+                         ushort_t *__restrict__ y, int K, int N, float eps) {
+  extern __shared__ uint8_t xl8[];
+  __shared__ float red[16];
+  const int lane = threadIdx.x & (WAVE - 1);
+  const int wid = threadIdx.x / WAVE;
+  const int rg = lane >> 5;
+  const int sl = lane & 31;
+  const int n0 = blockIdx.x * 8 + wid * 2 + rg;
+  const int n = n0 < N ? n0 : N - 1;  // no early return before barriers
+
+  const float s = fp8norm_stage(x, wln, xl8, red, K, eps, true);
+  float acc = fp8_dot_lds_w32(xl8, w + (size_t)n * K, K / 16, sl);
+#pragma unroll
+  for (int off = 16; off > 0; off >>= 1) acc += __shfl_xor(acc, off, WAVE);
+  if (sl == 0 && n0 < N) y[n0] = f32_to_bf16(acc * s * wsc[n0]);
+}
+
+// 16-lane-group variant for LARGE N (lm_head)
+extern "C" __global__ void __launch_bounds__(256)
+gemv_fp8_norm_kernel(const ushort_t *__restrict__ x,
+                     const ushort_t *__restrict__ wln,
+                     const uint8_t *__restrict__ w,
+                     const float *__restrict__ wsc,
+                     ushort_t *__restrict__ y, int K, int N, float eps) {
+  extern __shared__ uint8_t xl8[];
+  __shared__ float red[16];
+  const int lane = threadIdx.x & (WAVE - 1);
+  const int wid = threadIdx.x / WAVE;
+  const int rg = lane >> 4;
+  const int sl = lane & 15;
+  const int n0 = blockIdx.x * 16 + wid * 4 + rg;
+  const int n = n0 < N ? n0 : N - 1;
+
+  const float s = fp8norm_stage(x, wln, xl8, red, K, eps, true);
+  const uint8_t *wr = w + (size_t)n * K;
+  const int nc = K / 16;
+  float acc = 0.f;
+  int c = sl;
+  for (; c + 48 < nc; c += 64) {
+    uint4 wv[4];
+#pragma unroll
+    for (int u = 0; u < 4; ++u)
+      wv[u] = *(const uint4 *)(wr + (size_t)(c + 16 * u) * 16);
+#pragma unroll
+    for (int u = 0; u < 4; ++u)
+      acc += dot16_fp8(*(const uint4 *)(xl8 + (size_t)(c + 16 * u) * 16),
+                       wv[u]);
+  }
+  for (; c < nc; c += 16)
+    acc += dot16_fp8(*(const uint4 *)(xl8 + (size_t)c * 16),
+                     *(const uint4 *)(wr + (size_t)c * 16));
+#pragma unroll
+  for (int off = 8; off > 0; off >>= 1) acc += __shfl_xor(acc, off, WAVE);
+  if (sl == 0 && n0 < N) y[n0] = f32_to_bf16(acc * s * wsc[n0]);
+}
+
+// resid += (x @ w8^T): quantize-in-LDS + GEMV + residual epilogue
+extern "C" __global__ void __launch_bounds__(256)
+gemv_fp8_resl_w32_kernel(const ushort_t *__restrict__ x,
+                         const uint8_t *__restrict__ w,
+                         const float *__restrict__ wsc,
+                         ushort_t *__restrict__ resid, int K, int N) {
+  extern __shared__ uint8_t xl8[];
+  __shared__ float red[16];
+  const int lane = threadIdx.x & (WAVE - 1);
+  const int wid = threadIdx.x / WAVE;
+  const int rg = lane >> 5;
+  const int sl = lane & 31;
+  const int n0 = blockIdx.x * 8 + wid * 2 + rg;
+  const int n = n0 < N ? n0 : N - 1;
+
+  const float r0 = bf16_to_f32(resid[n]);  // prefetch
+  const float s = fp8norm_stage(x, nullptr, xl8, red, K, 0.f, false);
+  float acc = fp8_dot_lds_w32(xl8, w + (size_t)n * K, K / 16, sl);
+#pragma unroll
+  for (int off = 16; off > 0; off >>= 1) acc += __shfl_xor(acc, off, WAVE);
+  if (sl == 0 && n0 < N) resid[n0] = f32_to_bf16(r0 + acc * s * wsc[n0]);
+}
+
+// act = swiglu(rmsnorm(x,wln) @ [Wg|Wu]^T): the whole fp8 MLP front half
+extern "C" __global__ void __launch_bounds__(256)
+gemv_fp8_gateup_norm_kernel(const ushort_t *__restrict__ x,
+                            const ushort_t *__restrict__ wln,
+                            const uint8_t *__restrict__ w,
+                            const float *__restrict__ wsc,
+                            ushort_t *__restrict__ act, int K, int F,
+                            float eps) {
+  extern __shared__ uint8_t xl8[];
+  __shared__ float red[16];
+  const int lane = threadIdx.x & (WAVE - 1);
+  const int wid = threadIdx.x / WAVE;
+  const int rg = lane >> 5;
+  const int sl = lane & 31;
+  const int n0 = blockIdx.x * 8 + wid * 2 + rg;
+  const int n = n0 < F ? n0 : F - 1;
+
+  const float s = fp8norm_stage(x, wln, xl8, red, K, eps, true);
+  const uint8_t *wg = w + (size_t)n * K;
+  const uint8_t *wu = w + (size_t)(n + F) * K;
+  const int nc = K / 16;
+  float ag = 0.f, au = 0.f;
+  int c = sl;
+  for (; c + 96 < nc; c += 128) {
+    uint4 gv[4], uv[4];
+#pragma unroll
+    for (int u = 0; u < 4; ++u) {
+      gv[u] = *(const uint4 *)(wg + (size_t)(c + 32 * u) * 16);
+      uv[u] = *(const uint4 *)(wu + (size_t)(c + 32 * u) * 16);
+    }
+#pragma unroll
+    for (int u = 0; u < 4; ++u) {
+      const uint4 xv = *(const uint4 *)(xl8 + (size_t)(c + 32 * u) * 16);
+      ag += dot16_fp8(xv, gv[u]);
+      au += dot16_fp8(xv, uv[u]);
+    }
+  }
+  for (; c < nc; c += 32) {
+    const uint4 xv = *(const uint4 *)(xl8 + (size_t)c * 16);
+    ag += dot16_fp8(xv, *(const uint4 *)(wg + (size_t)c * 16));
+    au += dot16_fp8(xv, *(const uint4 *)(wu + (size_t)c * 16));
+  }
+#pragma unroll
+  for (int off = 16; off > 0; off >>= 1) {
+    ag += __shfl_xor(ag, off, WAVE);
+    au += __shfl_xor(au, off, WAVE);
+  }
+  if (sl == 0 && n0 < F) {
+    const float g = ag * s * wsc[n0];
+    const float u = au * s * wsc[n0 + F];
+    const float sg = g / (1.0f + __expf(-g));  // silu(gate)
+    act[n0] = f32_to_bf16(sg * u);
+  }
+}
+
+extern "C" void launch_gemv_fp8_norm(const ushort_t *x, const ushort_t *wln,
+                                     const uint8_t *w, const float *wsc,
+                                     ushort_t *y, int K, int N, float eps,
+                                     hipStream_t stream) {
+  const size_t lds = (size_t)K;  // e4m3 staging
+  if (N <= 8192)
+    gemv_fp8_norm_w32_kernel<<<dim3((N + 7) / 8), 256, lds, stream>>>(
+        x, wln, w, wsc, y, K, N, eps);
+  else
+    gemv_fp8_norm_kernel<<<dim3((N + 15) / 16), 256, lds, stream>>>(
+        x, wln, w, wsc, y, K, N, eps);
+}
+
+extern "C" void launch_gemv_fp8_resl(const ushort_t *x, const uint8_t *w,
+                                     const float *wsc, ushort_t *resid,
+                                     int K, int N, hipStream_t stream) {
+  gemv_fp8_resl_w32_kernel<<<dim3((N + 7) / 8), 256, (size_t)K, stream>>>(
+      x, w, wsc, resid, K, N);
+}
+
+extern "C" void launch_gemv_fp8_gateup_norm(const ushort_t *x,
+                                            const ushort_t *wln,
+                                            const uint8_t *w,
+                                            const float *wsc, ushort_t *act,
+                                            int K, int F, float eps,
+                                            hipStream_t stream) {
+  gemv_fp8_gateup_norm_kernel<<<dim3((F + 7) / 8), 256, (size_t)K, stream>>>(
+      x, wln, w, wsc, act, K, F, eps);
+}
+
 extern "C" void launch_quant_norm_fp8(const ushort_t *x, const ushort_t *wln,
                                       uint8_t *q, float *scale, int K,
                                       float eps, hipStream_t stream) {

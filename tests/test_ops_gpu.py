@@ -608,6 +608,41 @@ class TestGemvNormResFusion:
         ops.gemv_fp8_res(xin, wq, wsc, x8, xs, resid)
         _assert_close(resid, want_res, atol=1.5e-1, name="gemv_fp8_res")
 
+    def test_fp8_lds_fused_kernels(self):
+        """Single-launch LDS-staged fp8 GEMVs (norm / res / gateup_norm)
+        vs the fp32 reference of the same quantized weights."""
+        k, n, f = 2048, 1024, 1536
+        x = _bf(torch.randn(1, k)).to(DEV)
+        wln = _bf(torch.rand(k) + 0.5).to(DEV)
+        w = _bf(torch.randn(n, k) * 0.02).to(DEV)
+        wgu = _bf(torch.randn(2 * f, k) * 0.02).to(DEV)
+        wq, wsc = ops.quantize_fp8_rowwise(w)
+        gq, gsc = ops.quantize_fp8_rowwise(wgu)
+
+        normed = self._norm_ref(x, wln)
+        s = float(normed.abs().amax()) / 448.0
+        n8 = (normed / s).to(torch.float8_e4m3fn).float() * s
+        wd = ops.dequantize_fp8(wq.cpu(), wsc.cpu())
+        gd = ops.dequantize_fp8(gq.cpu(), gsc.cpu())
+
+        out = torch.empty(1, n, dtype=torch.bfloat16, device=DEV)
+        ops.gemv_fp8_norm(x, wln, wq, wsc, self.EPS, out)
+        _assert_close(out, n8 @ wd.t(), atol=1.5e-1, name="gemv_fp8_norm")
+
+        act = torch.empty(1, f, dtype=torch.bfloat16, device=DEV)
+        ops.gemv_fp8_gateup_norm(x, wln, gq, gsc, self.EPS, act)
+        gu = n8 @ gd.t()
+        _assert_close(act, torch_ref.swiglu(gu[:, :f], gu[:, f:]),
+                      atol=1e-1, name="gemv_fp8_gateup_norm")
+
+        resid = _bf(torch.randn(1, n)).to(DEV)
+        xin = _bf(torch.randn(1, k)).to(DEV)
+        si = float(xin.float().abs().amax()) / 448.0
+        xi8 = (xin.float().cpu() / si).to(torch.float8_e4m3fn).float() * si
+        want = resid.float().cpu() + xi8 @ wd.t()
+        ops.gemv_fp8_resl(xin, wq, wsc, resid)
+        _assert_close(resid, want, atol=1.5e-1, name="gemv_fp8_resl")
+
     def test_fused_decode_step_matches_unfused_forward(self):
         """decode_step_ws (fused GEMV path) vs decode_one (unfused kernel
         sequence) on the same prefilled cache: same logits direction."""
