@@ -22,6 +22,7 @@ Replaces the reference's remote `completion()` call.
 from __future__ import annotations
 
 import math
+import os
 from dataclasses import dataclass
 from typing import Optional
 
@@ -126,6 +127,16 @@ class DecodeWorkspace:
         kmax = max(c.dim, h * hd, c.ffn_dim)
         self.x8 = torch.empty(1, kmax, dtype=torch.uint8, device=dev)
         self.xs = torch.empty(1, dtype=torch.float32, device=dev)
+
+
+# fp8 fused-decode variant (A/B-measured, profiles/r02_measurements.md):
+# default = quant_norm trio (one 1-block quant_norm + plain fp8 GEMVs) —
+# best under the production 3-opponent concurrency (0.816 critiques/s vs
+# 0.744 for the LDS trio). ADVSPEC_FP8_LDS=1 switches to the LDS-staged
+# single-launch trio, which wins SOLO decode (3.51 vs 3.70 ms/tok): the
+# per-block redundant staging work is free on an idle chip but costs
+# shared VALU/L2 when co-resident opponents saturate the device.
+_FP8_LDS = os.environ.get("ADVSPEC_FP8_LDS") == "1"
 
 
 class LlamaModel:
@@ -526,8 +537,13 @@ class LlamaModel:
             Qd = self.layers_q[i] if fp8 else None
             if fused_f8:
                 qw = Qd.get("wqkv")
-                ops.gemv_fp8_norm(W.resid, L.attn_norm, qw.q, qw.s,
-                                  c.norm_eps, W.qkv)
+                if _FP8_LDS:
+                    ops.gemv_fp8_norm(W.resid, L.attn_norm, qw.q, qw.s,
+                                      c.norm_eps, W.qkv)
+                else:
+                    ops.quant_norm_fp8(W.resid, L.attn_norm, W.x8, W.xs,
+                                       c.norm_eps)
+                    ops.gemv_fp8_q(W.x8, W.xs, qw.q, qw.s, W.qkv)
             elif fused_bf:
                 ops.gemv_norm(W.resid, L.attn_norm, L.wqkv, c.norm_eps,
                               out=W.qkv)
@@ -544,13 +560,25 @@ class LlamaModel:
                 identity=cache.identity, split_blocks=self.split_blocks,
             )
             if fused_f8:
-                ops.gemv_fp8_resl(W.attn.view(1, h * hd), Qd.get("wo").q,
-                                  Qd.get("wo").s, W.resid)
                 gw = Qd.get("w_gate_up")
-                ops.gemv_fp8_gateup_norm(W.resid, L.mlp_norm, gw.q, gw.s,
-                                         c.norm_eps, W.act)
-                ops.gemv_fp8_resl(W.act, Qd.get("w_down").q,
-                                  Qd.get("w_down").s, W.resid)
+                if _FP8_LDS:
+                    ops.gemv_fp8_resl(W.attn.view(1, h * hd),
+                                      Qd.get("wo").q, Qd.get("wo").s,
+                                      W.resid)
+                    ops.gemv_fp8_gateup_norm(W.resid, L.mlp_norm, gw.q,
+                                             gw.s, c.norm_eps, W.act)
+                    ops.gemv_fp8_resl(W.act, Qd.get("w_down").q,
+                                      Qd.get("w_down").s, W.resid)
+                else:
+                    ops.gemv_fp8_res(W.attn.view(1, h * hd),
+                                     Qd.get("wo").q, Qd.get("wo").s,
+                                     W.x8, W.xs, W.resid)
+                    ops.quant_norm_fp8(W.resid, L.mlp_norm, W.x8, W.xs,
+                                       c.norm_eps)
+                    ops.gemv_fp8_gateup(W.x8, W.xs, gw.q, gw.s, W.act)
+                    ops.gemv_fp8_res(W.act, Qd.get("w_down").q,
+                                     Qd.get("w_down").s, W.x8, W.xs,
+                                     W.resid)
                 continue
             if fused_bf:
                 ops.gemv_res(W.attn.view(1, h * hd), L.wo, W.resid)
@@ -578,8 +606,15 @@ class LlamaModel:
             ops.add_rmsnorm(W.resid2, W.mlp_out, nxt, c.norm_eps,
                             out_resid=W.resid, out_y=W.normed)
         if fused_f8:
-            ops.gemv_fp8_norm(W.resid, self.final_norm, self.lm_head_q.q,
-                              self.lm_head_q.s, c.norm_eps, W.logits)
+            if _FP8_LDS:
+                ops.gemv_fp8_norm(W.resid, self.final_norm,
+                                  self.lm_head_q.q, self.lm_head_q.s,
+                                  c.norm_eps, W.logits)
+            else:
+                ops.quant_norm_fp8(W.resid, self.final_norm, W.x8, W.xs,
+                                   c.norm_eps)
+                ops.gemv_fp8_q(W.x8, W.xs, self.lm_head_q.q,
+                               self.lm_head_q.s, W.logits)
         elif fused_bf:
             ops.gemv_norm(W.resid, self.final_norm, self.lm_head, c.norm_eps,
                           out=W.logits)
