@@ -22,4 +22,4 @@ Layers (MI355X mapping of SURVEY.md §1):
   utils/      - timing, synthetic spec generation
 """
 
-__version__ = "1.1.0"
+__version__ = "1.2.0"
