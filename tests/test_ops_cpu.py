@@ -177,3 +177,88 @@ class TestSampling:
         for _ in range(300):
             counts[ops.sample(logits, 1.0, generator=g)] += 1
         assert counts[0] > counts[1]
+
+
+class TestFusedGemvCPURefs:
+    """CPU reference implementations of the decode-fusion ops (the GPU
+    kernels are tested in test_ops_gpu.py; these pin the CPU fallback
+    contract that numerics tests compare against)."""
+
+    EPS = 1e-5
+
+    def _norm(self, x, wln):
+        xf = x.float()
+        rms = torch.rsqrt(xf.pow(2).mean(-1, keepdim=True) + self.EPS)
+        return xf * rms * wln.float()
+
+    def test_gemv_norm_matches_rmsnorm_then_matmul(self):
+        torch.manual_seed(0)
+        x = torch.randn(1, 64)
+        wln = torch.rand(64) + 0.5
+        w = torch.randn(32, 64)
+        y = ops.gemv_norm(x, wln, w, self.EPS)
+        want = self._norm(x, wln) @ w.t()
+        assert torch.allclose(y.float(), want, atol=1e-5)
+
+    def test_gemv_res_adds_in_place(self):
+        torch.manual_seed(1)
+        x = torch.randn(1, 64)
+        w = torch.randn(32, 64)
+        resid = torch.randn(1, 32)
+        want = resid + x @ w.t()
+        out = ops.gemv_res(x, w, resid)
+        assert out is resid
+        assert torch.allclose(resid, want, atol=1e-5)
+
+    def test_gemv_gateup_norm_matches_unfused(self):
+        torch.manual_seed(2)
+        f = 16
+        x = torch.randn(1, 64)
+        wln = torch.rand(64) + 0.5
+        w = torch.randn(2 * f, 64)
+        out = torch.empty(1, f)
+        ops.gemv_gateup_norm(x, wln, w, self.EPS, out)
+        gu = self._norm(x, wln) @ w.t()
+        want = torch_ref.swiglu(gu[:, :f], gu[:, f:])
+        assert torch.allclose(out, want, atol=1e-5)
+
+    def test_fp8_quant_norm_round_trip(self):
+        torch.manual_seed(3)
+        k = 64
+        x = torch.randn(1, k)
+        wln = torch.rand(k) + 0.5
+        x8 = torch.empty(1, k, dtype=torch.uint8)
+        xs = torch.empty(1, dtype=torch.float32)
+        ops.quant_norm_fp8(x, wln, x8, xs, self.EPS)
+        deq = x8.view(torch.float8_e4m3fn).float() * xs
+        want = self._norm(x, wln)
+        err = (deq - want).abs().max().item()
+        assert err <= want.abs().max().item() * 0.08
+
+    def test_fp8_fused_chain_cpu(self):
+        torch.manual_seed(4)
+        k, n, f = 64, 32, 16
+        x = torch.randn(1, k)
+        wln = torch.rand(k) + 0.5
+        wq, wsc = ops.quantize_fp8_rowwise(torch.randn(n, k) * 0.1)
+        gq, gsc = ops.quantize_fp8_rowwise(torch.randn(2 * f, k) * 0.1)
+
+        out = torch.empty(1, n)
+        ops.gemv_fp8_norm(x, wln, wq, wsc, self.EPS, out)
+        assert out.abs().sum() > 0  # sane, detailed numerics on GPU
+
+        act = torch.empty(1, f)
+        ops.gemv_fp8_gateup_norm(x, wln, gq, gsc, self.EPS, act)
+
+        resid = torch.randn(1, n)
+        before = resid.clone()
+        ops.gemv_fp8_resl(x, wq, wsc, resid)
+        assert not torch.allclose(resid, before)
+
+        x8 = torch.empty(1, k, dtype=torch.uint8)
+        xs = torch.empty(1, dtype=torch.float32)
+        ops.quant_norm_fp8(x, wln, x8, xs, self.EPS)
+        out2 = torch.empty(1, n)
+        ops.gemv_fp8_q(x8, xs, wq, wsc, out2)
+        # pre-quantized GEMV on the quant_norm output == the one-shot op
+        assert torch.allclose(out, out2, atol=1e-4)
