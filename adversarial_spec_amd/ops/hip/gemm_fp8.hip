@@ -278,20 +278,17 @@ gemv_fp8_kernel(const uint8_t *__restrict__ x, const float *__restrict__ xs,
   const uint8_t *wr = w + (size_t)n * K;
   const int nc = K / 16;  // 16 B chunks of 16 fp8 (K % 16 == 0)
 
-  // 8-deep: fp8 rows are HALF the chunks of bf16 at equal K, so a lane's
-  // whole row is only nc/16 chunks (16 at K=4096) — at 4-deep the kernel
-  // was pure load latency (~3.5 TB/s vs the bf16 GEMV's 6.9)
   float acc = 0.f;
   int c = sl;
-  for (; c + 112 < nc; c += 128) {
-    uint4 wv[8], xv[8];
+  for (; c + 48 < nc; c += 64) {
+    uint4 wv[4], xv[4];
 #pragma unroll
-    for (int u = 0; u < 8; ++u) {
+    for (int u = 0; u < 4; ++u) {
       wv[u] = *(const uint4 *)(wr + (size_t)(c + 16 * u) * 16);
       xv[u] = *(const uint4 *)(x + (size_t)(c + 16 * u) * 16);
     }
 #pragma unroll
-    for (int u = 0; u < 8; ++u) acc += dot16_fp8(xv[u], wv[u]);
+    for (int u = 0; u < 4; ++u) acc += dot16_fp8(xv[u], wv[u]);
   }
   for (; c < nc; c += 16) {
     const uint4 wv = *(const uint4 *)(wr + (size_t)c * 16);
@@ -384,18 +381,17 @@ gemv_fp8_kernel_w32(const uint8_t *__restrict__ x, const float *__restrict__ xs,
   const uint8_t *wr = w + (size_t)n * K;
   const int nc = K / 16;
 
-  // 8-deep (see gemv_fp8_kernel comment: half the chunks of bf16)
   float acc = 0.f;
   int c = sl;
-  for (; c + 224 < nc; c += 256) {
-    uint4 wv[8], xv[8];
+  for (; c + 96 < nc; c += 128) {
+    uint4 wv[4], xv[4];
 #pragma unroll
-    for (int u = 0; u < 8; ++u) {
+    for (int u = 0; u < 4; ++u) {
       wv[u] = *(const uint4 *)(wr + (size_t)(c + 32 * u) * 16);
       xv[u] = *(const uint4 *)(x + (size_t)(c + 32 * u) * 16);
     }
 #pragma unroll
-    for (int u = 0; u < 8; ++u) acc += dot16_fp8(xv[u], wv[u]);
+    for (int u = 0; u < 4; ++u) acc += dot16_fp8(xv[u], wv[u]);
   }
   for (; c < nc; c += 32) {
     const uint4 wv = *(const uint4 *)(wr + (size_t)c * 16);
@@ -500,18 +496,17 @@ gemv_fp8_res_w32_kernel(const uint8_t *__restrict__ x,
   const float r0 = bf16_to_f32(resid[n]);  // prefetch (tail-latency)
   const float ws_n = wsc[n];
 
-  // 8-deep (see gemv_fp8_kernel comment: half the chunks of bf16)
   float acc = 0.f;
   int c = sl;
-  for (; c + 224 < nc; c += 256) {
-    uint4 wv[8], xv[8];
+  for (; c + 96 < nc; c += 128) {
+    uint4 wv[4], xv[4];
 #pragma unroll
-    for (int u = 0; u < 8; ++u) {
+    for (int u = 0; u < 4; ++u) {
       wv[u] = *(const uint4 *)(wr + (size_t)(c + 32 * u) * 16);
       xv[u] = *(const uint4 *)(x + (size_t)(c + 32 * u) * 16);
     }
 #pragma unroll
-    for (int u = 0; u < 8; ++u) acc += dot16_fp8(xv[u], wv[u]);
+    for (int u = 0; u < 4; ++u) acc += dot16_fp8(xv[u], wv[u]);
   }
   for (; c < nc; c += 32) {
     const uint4 wv = *(const uint4 *)(wr + (size_t)c * 16);
