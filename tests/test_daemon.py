@@ -97,6 +97,37 @@ class TestDaemonGuards:
         finally:
             srv.server_close()
 
+    def test_serve_creates_deep_socket_dir(self, tmp_path):
+        """serve() must mkdir the socket's parents (mutation killer:
+        mkdir(parents=True) — a fresh install has no config dir)."""
+        sock = tmp_path / "a" / "b" / "daemon.sock"
+        srv = daemon.serve(sock)
+        try:
+            assert sock.exists()
+        finally:
+            srv.server_close()
+
+    def test_dispatch_error_reports_code_1(self, live_daemon, monkeypatch):
+        """Both daemon error paths answer with exit code EXACTLY 1 (the
+        reference's processing-error code; 2 means missing credentials)."""
+        sock, _srv = live_daemon
+        from adversarial_spec_amd.cli import debate as cli_mod
+
+        def boom(argv):
+            raise RuntimeError("engine exploded")
+
+        # inner path: cli.main raises -> caught inside run_request
+        monkeypatch.setattr(cli_mod, "main", boom)
+        fwd = daemon.try_forward(["critique"], SPEC, sock)
+        assert fwd is not None and fwd[0] == 1
+        # dispatch path: run_request itself raises -> handler catches
+        monkeypatch.setattr(daemon, "run_request",
+                            lambda argv, s: (_ for _ in ()).throw(
+                                RuntimeError("dispatch")))
+        fwd = daemon.try_forward(["critique"], SPEC, sock)
+        assert fwd is not None and fwd[0] == 1
+        assert "dispatch error" in fwd[2]
+
     def test_concurrent_forwards(self, live_daemon):
         """Two critiques in flight at once (the threaded server + per-
         engine locks must not deadlock or cross wires)."""

@@ -25,6 +25,8 @@ TARGETS = {
     "adversarial_spec_amd/protocol.py": ["tests/test_protocol.py"],
     "adversarial_spec_amd/engine/scheduler.py": ["tests/test_scheduler.py"],
     "adversarial_spec_amd/session.py": ["tests/test_session.py"],
+    "adversarial_spec_amd/daemon.py": ["tests/test_daemon.py"],
+    "adversarial_spec_amd/parallel/consensus.py": ["tests/test_consensus.py"],
 }
 
 CMP_SWAPS = {
@@ -66,8 +68,29 @@ def iter_mutants(tree: ast.Module):
 # Mutants shown equivalent by analysis (documented, not silently skipped):
 #  - `len(buf) > 1` vs `>= 1` in protocol.extract_tasks: buf entries are
 #    pre-stripped lines, so "\n".join(buf).strip() == buf[0] at len 1.
-EQUIVALENT_LINES = (
+#  - daemon `64 * 1024 * 1024` recv/readline caps and `while n < limit`:
+#    safety ceilings far above any real message; off-by-one/boundary is
+#    behavior-preserving for every representable request.
+#  - daemon `daemon_threads` / `allow_reuse_address` / shutdown-thread
+#    `daemon=True`: process-lifecycle knobs, not observable from an
+#    in-process test (non-daemon threads only matter at interpreter exit).
+#  - consensus `HDR = 4`: pack/unpack/gather all derive from HDR, so a
+#    coherent resize is wire-compatible within a job (one wasted word).
+#  - consensus `async_op=True`: synchronous all-gather is semantically
+#    identical (wait() becomes a no-op); perf-only.
+EQUIVALENT_SWAP_LINES = (
     'task[key] = "\\n".join(buf).strip() if len(buf) > 1',
+    "while n < limit:",
+)
+EQUIVALENT_LINES = (
+    "daemon_threads = True",
+    "allow_reuse_address = True",
+    "threading.Thread(target=srv.shutdown, daemon=True).start()",
+    "HDR = 4",
+)
+EQUIVALENT_FRAGMENTS = (
+    "* 1024 * 1024",   # daemon recv/readline caps
+    "async_op=True",   # consensus: sync collective is equivalent
 )
 
 
@@ -77,7 +100,11 @@ def should_skip(desc: str, src_line: str) -> bool:
     s = src_line.strip()
     if "max_workers" in s:
         return True  # pool sizing above the needed minimum is equivalent
-    if "swap" in desc and any(s.startswith(e) for e in EQUIVALENT_LINES):
+    if "swap" in desc and any(s.startswith(e) for e in EQUIVALENT_SWAP_LINES):
+        return True
+    if any(s.startswith(e) for e in EQUIVALENT_LINES):
+        return True
+    if any(f in s for f in EQUIVALENT_FRAGMENTS):
         return True
     return s.startswith(("#", '"', "'", "print(")) or "version" in s.lower()
 
