@@ -35,11 +35,22 @@ def main():
     logits = torch.empty(1, V, dtype=torch.bfloat16, device=DEV)
     xf = bf(1, F)
 
+    # fp8 path (quant trio default): which bound — cvt-issue or wait?
+    q_qkv, s_qkv = ops.quantize_fp8_rowwise(w_qkv)
+    q_gu, s_gu = ops.quantize_fp8_rowwise(w_gu)
+    q_dn, s_dn = ops.quantize_fp8_rowwise(w_dn)
+    x8 = torch.empty(1, max(K, F), dtype=torch.uint8, device=DEV)
+    xs = torch.empty(1, dtype=torch.float32, device=DEV)
+
     for _ in range(ITERS):
         ops.gemv_norm(x, wln, w_qkv, EPS, out=qkv)
         ops.gemv_gateup_norm(x, wln, w_gu, EPS, act)
         ops.gemv_res(xf, w_dn, resid)
         ops.gemv_norm(x, wln, w_lm, EPS, out=logits)
+        ops.quant_norm_fp8(x, wln, x8, xs, EPS)
+        ops.gemv_fp8_q(x8, xs, q_qkv, s_qkv, qkv)
+        ops.gemv_fp8_gateup(x8, xs, q_gu, s_gu, act)
+        ops.gemv_fp8_res(xf, q_dn, s_dn, x8, xs, resid)
     torch.cuda.synchronize()
     print("pmc target done", ITERS)
 
