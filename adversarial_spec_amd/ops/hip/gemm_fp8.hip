@@ -366,12 +366,7 @@ extern "C" void launch_gemm_fp8(const uint8_t *a, const float *asc,
   gemm_fp8_kernel<<<grid, 256, 0, stream>>>(a, asc, b, bsc, c, M, N, K);
 }
 
-// 32-lane-group, TWO-rows-per-group variant for SMALL N. At equal K an
-// fp8 row is HALF the 16 B chunks of bf16, so a one-row group has half
-// the loads in flight over the same latency structure — PMC shows these
-// kernels wait-dominated (SQ_WAIT/SQ_BUSY ~14x) at ~3.5 TB/s, half the
-// bf16 GEMV's rate. Pairing rows restores the per-wave weight-byte depth
-// (two streams, like the gateup kernel).
+// 32-lanes-per-row variant for SMALL N (same rationale as gemv.hip)
 extern "C" __global__ void __launch_bounds__(256)
 gemv_fp8_kernel_w32(const uint8_t *__restrict__ x, const float *__restrict__ xs,
                     const uint8_t *__restrict__ w, const float *__restrict__ wsc,
@@ -380,45 +375,33 @@ gemv_fp8_kernel_w32(const uint8_t *__restrict__ x, const float *__restrict__ xs,
   const int wid = threadIdx.x / WAVE;
   const int rg = lane >> 5;
   const int sl = lane & 31;
-  const int n0 = (blockIdx.x * 8 + wid * 2 + rg) * 2;
-  if (n0 >= N) return;
-  const int n1 = min(n0 + 1, N - 1);
+  const int n = blockIdx.x * 8 + wid * 2 + rg;
+  if (n >= N) return;
 
-  const uint8_t *w0 = w + (size_t)n0 * K;
-  const uint8_t *w1 = w + (size_t)n1 * K;
+  const uint8_t *wr = w + (size_t)n * K;
   const int nc = K / 16;
 
-  float a0 = 0.f, a1 = 0.f;
+  float acc = 0.f;
   int c = sl;
   for (; c + 96 < nc; c += 128) {
-    uint4 wv0[4], wv1[4], xv[4];
+    uint4 wv[4], xv[4];
 #pragma unroll
     for (int u = 0; u < 4; ++u) {
-      wv0[u] = *(const uint4 *)(w0 + (size_t)(c + 32 * u) * 16);
-      wv1[u] = *(const uint4 *)(w1 + (size_t)(c + 32 * u) * 16);
+      wv[u] = *(const uint4 *)(wr + (size_t)(c + 32 * u) * 16);
       xv[u] = *(const uint4 *)(x + (size_t)(c + 32 * u) * 16);
     }
 #pragma unroll
-    for (int u = 0; u < 4; ++u) {
-      a0 += dot16_fp8(xv[u], wv0[u]);
-      a1 += dot16_fp8(xv[u], wv1[u]);
-    }
+    for (int u = 0; u < 4; ++u) acc += dot16_fp8(xv[u], wv[u]);
   }
   for (; c < nc; c += 32) {
+    const uint4 wv = *(const uint4 *)(wr + (size_t)c * 16);
     const uint4 xv = *(const uint4 *)(x + (size_t)c * 16);
-    a0 += dot16_fp8(xv, *(const uint4 *)(w0 + (size_t)c * 16));
-    a1 += dot16_fp8(xv, *(const uint4 *)(w1 + (size_t)c * 16));
+    acc += dot16_fp8(xv, wv);
   }
 
 #pragma unroll
-  for (int off = 16; off > 0; off >>= 1) {
-    a0 += __shfl_xor(a0, off, WAVE);
-    a1 += __shfl_xor(a1, off, WAVE);
-  }
-  if (sl == 0) {
-    y[n0] = f32_to_bf16(a0 * xs[0] * wsc[n0]);
-    if (n0 + 1 < N) y[n0 + 1] = f32_to_bf16(a1 * xs[0] * wsc[n0 + 1]);
-  }
+  for (int off = 16; off > 0; off >>= 1) acc += __shfl_xor(acc, off, WAVE);
+  if (sl == 0) y[n] = f32_to_bf16(acc * xs[0] * wsc[n]);
 }
 
 extern "C" void launch_gemv_fp8(const uint8_t *x, const float *xs,
@@ -434,9 +417,8 @@ extern "C" void launch_gemv_fp8(const uint8_t *x, const float *xs,
   // (stage W and x through LDS in full 128-B lines via glds, 8-B
   // fragment reads from LDS) — future work, not a drop-in.
   if (N <= 8192) {
-    gemv_fp8_kernel_w32<<<dim3((N + 15) / 16), 256, 0, stream>>>(x, xs, w,
-                                                                 wsc, y, K,
-                                                                 N);
+    gemv_fp8_kernel_w32<<<dim3((N + 7) / 8), 256, 0, stream>>>(x, xs, w, wsc,
+                                                               y, K, N);
   } else {
     gemv_fp8_kernel<<<dim3((N + 15) / 16), 256, 0, stream>>>(x, xs, w, wsc, y,
                                                              K, N);
@@ -506,48 +488,35 @@ gemv_fp8_res_w32_kernel(const uint8_t *__restrict__ x,
   const int wid = threadIdx.x / WAVE;
   const int rg = lane >> 5;
   const int sl = lane & 31;
-  // two rows per 32-lane group (see gemv_fp8_kernel_w32 comment)
-  const int n0 = (blockIdx.x * 8 + wid * 2 + rg) * 2;
-  if (n0 >= N) return;
-  const int n1 = min(n0 + 1, N - 1);
+  const int n = blockIdx.x * 8 + wid * 2 + rg;
+  if (n >= N) return;
 
-  const uint8_t *w0 = w + (size_t)n0 * K;
-  const uint8_t *w1 = w + (size_t)n1 * K;
+  const uint8_t *wr = w + (size_t)n * K;
   const int nc = K / 16;
-  const float r0 = bf16_to_f32(resid[n0]);  // prefetch (tail-latency)
-  const float r1 = bf16_to_f32(resid[n1]);
+  const float r0 = bf16_to_f32(resid[n]);  // prefetch (tail-latency)
+  const float ws_n = wsc[n];
 
-  float a0 = 0.f, a1 = 0.f;
+  float acc = 0.f;
   int c = sl;
   for (; c + 96 < nc; c += 128) {
-    uint4 wv0[4], wv1[4], xv[4];
+    uint4 wv[4], xv[4];
 #pragma unroll
     for (int u = 0; u < 4; ++u) {
-      wv0[u] = *(const uint4 *)(w0 + (size_t)(c + 32 * u) * 16);
-      wv1[u] = *(const uint4 *)(w1 + (size_t)(c + 32 * u) * 16);
+      wv[u] = *(const uint4 *)(wr + (size_t)(c + 32 * u) * 16);
       xv[u] = *(const uint4 *)(x + (size_t)(c + 32 * u) * 16);
     }
 #pragma unroll
-    for (int u = 0; u < 4; ++u) {
-      a0 += dot16_fp8(xv[u], wv0[u]);
-      a1 += dot16_fp8(xv[u], wv1[u]);
-    }
+    for (int u = 0; u < 4; ++u) acc += dot16_fp8(xv[u], wv[u]);
   }
   for (; c < nc; c += 32) {
+    const uint4 wv = *(const uint4 *)(wr + (size_t)c * 16);
     const uint4 xv = *(const uint4 *)(x + (size_t)c * 16);
-    a0 += dot16_fp8(xv, *(const uint4 *)(w0 + (size_t)c * 16));
-    a1 += dot16_fp8(xv, *(const uint4 *)(w1 + (size_t)c * 16));
+    acc += dot16_fp8(xv, wv);
   }
 
 #pragma unroll
-  for (int off = 16; off > 0; off >>= 1) {
-    a0 += __shfl_xor(a0, off, WAVE);
-    a1 += __shfl_xor(a1, off, WAVE);
-  }
-  if (sl == 0) {
-    resid[n0] = f32_to_bf16(r0 + a0 * xs[0] * wsc[n0]);
-    if (n0 + 1 < N) resid[n0 + 1] = f32_to_bf16(r1 + a1 * xs[0] * wsc[n0 + 1]);
-  }
+  for (int off = 16; off > 0; off >>= 1) acc += __shfl_xor(acc, off, WAVE);
+  if (sl == 0) resid[n] = f32_to_bf16(r0 + acc * xs[0] * ws_n);
 }
 
 // fused fp8 gate_up GEMV + SwiGLU: act[n] = silu(g)*u with
@@ -857,8 +826,8 @@ extern "C" void launch_gemv_fp8_res(const uint8_t *x, const float *xs,
                                     const uint8_t *w, const float *wsc,
                                     ushort_t *resid, int K, int N,
                                     hipStream_t stream) {
-  gemv_fp8_res_w32_kernel<<<dim3((N + 15) / 16), 256, 0, stream>>>(
-      x, xs, w, wsc, resid, K, N);
+  gemv_fp8_res_w32_kernel<<<dim3((N + 7) / 8), 256, 0, stream>>>(x, xs, w, wsc,
+                                                                 resid, K, N);
 }
 
 extern "C" void launch_gemv_fp8_gateup(const uint8_t *x, const float *xs,
