@@ -188,7 +188,7 @@ def main(argv: Optional[list[str]] = None) -> int:
             print(reply)
             return 0
         return 1
-    return 1
+    return 1  # unreachable: argparse choices covers every command
 
 
 if __name__ == "__main__":

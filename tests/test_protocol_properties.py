@@ -122,3 +122,10 @@ class TestSplitMessageProperties:
         # newline-boundary cuts drop ONLY the boundary newlines
         joined = "".join(split_message(text, limit))
         assert joined.replace("\n", "") == text.replace("\n", "")
+
+    @given(text=st.text(min_size=1, max_size=2000), limit=st.integers(10, 100))
+    @settings(max_examples=80, deadline=None)
+    def test_no_empty_chunks(self, text, limit):
+        # an empty chunk would become an empty Telegram send
+        for c in split_message(text, limit):
+            assert c != ""

@@ -155,3 +155,88 @@ class TestCLI:
         with patch.object(telegram, "send_long_message", return_value=True):
             assert telegram.main(["send", "--message", "hello"]) == 0
         capsys.readouterr()
+
+
+class TestMutationKillers:
+    """Pin behaviors the example tests left unobserved (mutation-driven)."""
+
+    def test_poll_offset_semantics(self):
+        """getUpdates must start EXACTLY after the watermark (off-by-one
+        re-reads or skips an update) and ack past the consumed update."""
+        calls = []
+
+        def fake_api(token, method, params=None, timeout=35):
+            calls.append(dict(params or {}))
+            return {"ok": True, "result": [
+                {"update_id": 10,
+                 "message": {"chat": {"id": 42}, "text": "ok"}}]}
+
+        with patch.object(telegram, "api_call", side_effect=fake_api):
+            assert telegram.poll_for_reply("t", "42", 9, timeout=5) == "ok"
+        assert calls[0]["offset"] == 10   # watermark + 1, exactly
+        assert calls[-1]["offset"] == 11  # ack = consumed update_id + 1
+
+    def test_discover_offset_semantics(self):
+        calls = []
+
+        def fake_api(token, method, params=None, timeout=35):
+            calls.append(dict(params or {}))
+            if len(calls) == 1:  # get_last_update_id probe
+                return {"ok": True, "result": [{"update_id": 4}]}
+            return {"ok": True, "result": [
+                {"update_id": 5, "message": {"chat": {"id": 7}}}]}
+
+        with patch.object(telegram, "api_call", side_effect=fake_api):
+            assert telegram.discover_chat_id("t", wait=5) == "7"
+        assert calls[1]["offset"] == 5  # last_update_id + 1, exactly
+
+    def test_send_long_failure_propagates(self):
+        """A failed chunk must fail the WHOLE send (silent-success bug)."""
+        with patch.object(telegram, "send_message", return_value=False):
+            assert telegram.send_long_message("t", "c", "hello") is False
+
+    def test_cli_exit_codes_exact(self, clean_env, capsys):
+        clean_env.setenv("TELEGRAM_BOT_TOKEN", "t")
+        # setup finds nothing -> 1
+        with patch.object(telegram, "discover_chat_id", return_value=None):
+            assert telegram.main(["setup", "--timeout", "1"]) == 1
+        # chat id missing for send -> 2
+        assert telegram.main(["send", "--message", "x"]) == 2
+        clean_env.setenv("TELEGRAM_CHAT_ID", "c")
+        # send failure -> 1
+        with patch.object(telegram, "send_long_message", return_value=False):
+            assert telegram.main(["send", "--message", "x"]) == 1
+        # poll dispatch: reply -> 0 and prints it; none -> 1
+        with patch.object(telegram, "get_last_update_id", return_value=0), \
+             patch.object(telegram, "poll_for_reply", return_value="yes"):
+            assert telegram.main(["poll"]) == 0
+        assert "yes" in capsys.readouterr().out
+        with patch.object(telegram, "get_last_update_id", return_value=0), \
+             patch.object(telegram, "poll_for_reply", return_value=None):
+            assert telegram.main(["poll"]) == 1
+        capsys.readouterr()
+
+    def test_split_no_empty_chunk_on_leading_newline(self):
+        """rfind can return 0 (newline at position 0): the cut<=0 guard
+        must treat it as no-boundary, never emitting an empty chunk."""
+        chunks = telegram.split_message("\n" + "x" * 50, 10)
+        assert all(c != "" for c in chunks)
+        assert "".join(chunks).replace("\n", "") == "x" * 50
+
+    def test_discover_advances_past_chatless_updates(self):
+        """Updates without a chat id must be acked exactly one past their
+        update_id, or discover re-reads them forever."""
+        calls = []
+
+        def fake_api(token, method, params=None, timeout=35):
+            calls.append(dict(params or {}))
+            if len(calls) == 1:
+                return {"ok": True, "result": []}  # watermark probe -> 0
+            if len(calls) == 2:
+                return {"ok": True, "result": [{"update_id": 3}]}  # chatless
+            return {"ok": True, "result": [
+                {"update_id": 4, "message": {"chat": {"id": 9}}}]}
+
+        with patch.object(telegram, "api_call", side_effect=fake_api):
+            assert telegram.discover_chat_id("t", wait=30) == "9"
+        assert calls[2]["offset"] == 4  # 3 + 1, exactly

@@ -27,6 +27,8 @@ TARGETS = {
     "adversarial_spec_amd/session.py": ["tests/test_session.py"],
     "adversarial_spec_amd/daemon.py": ["tests/test_daemon.py"],
     "adversarial_spec_amd/parallel/consensus.py": ["tests/test_consensus.py"],
+    "adversarial_spec_amd/telegram.py": [
+        "tests/test_telegram.py", "tests/test_protocol_properties.py"],
 }
 
 CMP_SWAPS = {
@@ -81,6 +83,13 @@ def iter_mutants(tree: ast.Module):
 EQUIVALENT_SWAP_LINES = (
     'task[key] = "\\n".join(buf).strip() if len(buf) > 1',
     "while n < limit:",
+    # telegram boundary swaps are instant/exact-limit equivalent:
+    #  - `time.time() < deadline` vs <= differs only at one instant;
+    #  - split_message at len(text)==limit produces [text] on both the
+    #    early-return and the loop path (rfind->limit cut).
+    "while time.time() < deadline:",
+    "if len(text) <= limit:",
+    "while len(rest) > limit:",
 )
 EQUIVALENT_LINES = (
     "daemon_threads = True",
@@ -91,6 +100,18 @@ EQUIVALENT_LINES = (
 EQUIVALENT_FRAGMENTS = (
     "* 1024 * 1024",   # daemon recv/readline caps
     "async_op=True",   # consensus: sync collective is equivalent
+    # telegram long-poll timing knobs: any nearby value behaves identically
+    # through mocked api_call (and in production is a tolerance, not a
+    # contract — the reference hard-codes similar values)
+    "LONG_POLL_SLICE = 30",
+    "timeout: int = 35",
+    "timeout: int = 60",
+    "wait: int = 60",
+    "timeout=poll + 5",
+    "remaining = max(1, int(deadline",
+    '{"timeout": min(LONG_POLL_SLICE',
+    'default=60, help="Poll time',
+    "# unreachable",   # defensive dead returns (argparse-choices covered)
 )
 
 
