@@ -164,3 +164,73 @@ class TestValidateCredentials:
     def test_unknown_scheme_passes(self, clean_env, isolated_paths):
         valid, invalid = providers.validate_model_credentials(["weird/model"])
         assert valid == ["weird/model"]
+
+
+class TestMutationKillers:
+    """Pin exact exit codes / defaults the example tests left unobserved."""
+
+    def test_bedrock_error_paths_exit_1_exactly(self, isolated_paths, capsys):
+        h = providers.handle_bedrock_command
+        assert h("add-model", None, None, None) == 1
+        assert h("remove-model", None, None, None) == 1
+        assert h("remove-model", "not-there", None, None) == 1
+        assert h("alias", "name-only", None, None) == 1
+        assert h("definitely-unknown", None, None, None) == 1
+        capsys.readouterr()
+
+    def test_bedrock_list_models_exits_0(self, isolated_paths, capsys):
+        assert providers.handle_bedrock_command("list-models", None, None,
+                                                None) == 0
+        out = capsys.readouterr().out
+        assert "llama-3-8b" in out
+
+    def test_local_error_paths_exit_1_exactly(self, isolated_paths, capsys):
+        h = providers.handle_local_command
+        assert h("add-model", None, None, None, None) == 1
+        assert h("add-model", "no-such-arch-xyz", None, None, None) == 1
+        assert h("remove-model", None, None, None, None) == 1
+        assert h("remove-model", "not-there", None, None, None) == 1
+        assert h("alias", "name-only", None, None, None) == 1
+        assert h("alias", "my", "no-such-arch-xyz", None, None) == 1
+        assert h("definitely-unknown", None, None, None, None) == 1
+        capsys.readouterr()
+
+    def test_local_list_models_exits_0(self, isolated_paths, capsys):
+        assert providers.handle_local_command("list-models", None, None,
+                                              None, None) == 0
+        assert "llama-3-8b" in capsys.readouterr().out
+
+    def test_bedrock_defaults_disabled(self, isolated_paths, capsys):
+        """A fresh config materialized by any bedrock write must default to
+        enabled: False (opt-in)."""
+        assert providers.handle_bedrock_command("add-model", "llama-3-8b",
+                                                None, None) == 0
+        cfg = providers.load_global_config()
+        assert cfg["bedrock"]["enabled"] is False
+        capsys.readouterr()
+
+    def test_config_saves_survive_existing_and_deep_dirs(
+            self, isolated_paths, tmp_path, monkeypatch):
+        """save_global_config / save_profile must create parents (a fresh
+        install has NO ~/.claude/adversarial-spec chain) AND tolerate the
+        directory already existing (second save)."""
+        monkeypatch.setattr(providers, "GLOBAL_CONFIG_PATH",
+                            tmp_path / "deep" / "chain" / "config.json")
+        monkeypatch.setattr(providers, "PROFILES_DIR",
+                            tmp_path / "deep" / "profiles" / "nested")
+        providers.save_global_config({"a": 1})
+        providers.save_global_config({"a": 2})  # dir exists now
+        assert providers.load_global_config() == {"a": 2}
+        providers.save_profile("p1", {"models": ["x"]})
+        providers.save_profile("p1", {"models": ["y"]})
+        assert providers.load_profile("p1")["models"] == ["y"]
+
+    def test_local_engine_available_false_on_torch_error(self, monkeypatch):
+        monkeypatch.delenv("ADVSPEC_FORCE_LOCAL", raising=False)
+        import torch
+
+        def boom():
+            raise RuntimeError("no device")
+
+        monkeypatch.setattr(torch.cuda, "is_available", boom)
+        assert providers.local_engine_available() is False

@@ -29,6 +29,7 @@ TARGETS = {
     "adversarial_spec_amd/parallel/consensus.py": ["tests/test_consensus.py"],
     "adversarial_spec_amd/telegram.py": [
         "tests/test_telegram.py", "tests/test_protocol_properties.py"],
+    "adversarial_spec_amd/providers.py": ["tests/test_providers.py"],
 }
 
 CMP_SWAPS = {
@@ -112,6 +113,7 @@ EQUIVALENT_FRAGMENTS = (
     '{"timeout": min(LONG_POLL_SLICE',
     'default=60, help="Poll time',
     "# unreachable",   # defensive dead returns (argparse-choices covered)
+    "indent=2",        # JSON pretty-print width: formatting only
 )
 
 
