@@ -30,6 +30,7 @@ TARGETS = {
     "adversarial_spec_amd/telegram.py": [
         "tests/test_telegram.py", "tests/test_protocol_properties.py"],
     "adversarial_spec_amd/providers.py": ["tests/test_providers.py"],
+    "adversarial_spec_amd/engine/backend.py": ["tests/test_backends.py"],
 }
 
 CMP_SWAPS = {
@@ -114,6 +115,7 @@ EQUIVALENT_FRAGMENTS = (
     'default=60, help="Poll time',
     "# unreachable",   # defensive dead returns (argparse-choices covered)
     "indent=2",        # JSON pretty-print width: formatting only
+    "head = user_message[:64]",  # stub round-parse window: simulation knob
 )
 
 
