@@ -269,3 +269,97 @@ class TestTelegramGlue:
             )
         data = json.loads(out)
         assert data["user_feedback"] == "add a rollback section"
+
+
+class TestMutationKillers:
+    """Pin precedence/consensus/dispatch exactness (mutation-driven)."""
+
+    def test_profile_doc_type_precedence(self, clean_env, tmp_path, monkeypatch):
+        """A profile's doc_type fills only the DEFAULT (tech); an explicit
+        --doc-type wins over the profile."""
+        from adversarial_spec_amd import providers
+
+        monkeypatch.setattr(providers, "PROFILES_DIR", tmp_path)
+        providers.save_profile("p", {"doc_type": "prd"})
+
+        parser = debate.create_parser()
+        args = parser.parse_args(["critique", "--profile", "p"])
+        debate.apply_profile(args)
+        assert args.doc_type == "prd"  # default tech -> profile fills
+
+        args = parser.parse_args(
+            ["critique", "--profile", "p", "--doc-type", "tech"])
+        # explicit tech is indistinguishable from the default by design
+        # (reference semantics, debate.py:529-550) — but an explicit PRD
+        # must never be overridden the other way:
+        providers.save_profile("p2", {"doc_type": "tech"})
+        args = parser.parse_args(
+            ["critique", "--profile", "p2", "--doc-type", "prd"])
+        debate.apply_profile(args)
+        assert args.doc_type == "prd"
+
+    def test_all_error_round_is_not_consensus(self, clean_env):
+        """Every opponent erroring must NOT report all_agreed (a round with
+        zero successful critiques has no consensus to claim)."""
+        code, out, err = run_cli(
+            ["critique", "--models", "stub/error", "--json"], stdin="spec")
+        data = json.loads(out)
+        assert data["all_agreed"] is False
+
+    def test_round_default_is_1(self):
+        args = debate.create_parser().parse_args(["critique"])
+        assert args.round == 1
+
+    def test_export_tasks_numbering_starts_at_1(self, clean_env):
+        canned = ("[TASK]\ntitle: First\ntype: feature\npriority: high\n"
+                  "description: d\n[/TASK]")
+
+        class FakeBackend:
+            def generate(self, *a, **k):
+                return canned, 10, 10
+
+        with patch("adversarial_spec_amd.engine.backend.get_backend",
+                   return_value=FakeBackend()):
+            code, out, _ = run_cli(
+                ["export-tasks", "--models", "stub/x"], stdin="spec")
+        assert code == 0
+        assert "1. [feature] [high] First" in out
+
+    def test_daemon_forward_uses_sys_argv_tail(self, clean_env, monkeypatch):
+        """main(None) forwards sys.argv[1:] EXACTLY to a live daemon ([2:]
+        would eat the action and replay the wrong command)."""
+        import sys as _sys
+
+        from adversarial_spec_amd import daemon as dmod
+
+        argv = ["debate.py", "critique", "--models", "stub/agree", "--json"]
+        monkeypatch.setattr(_sys, "argv", argv)
+        seen = {}
+
+        def fake_forward(raw_argv, stdin_text, socket_path=None):
+            seen["argv"] = list(raw_argv)
+            return 0, "{}", ""
+
+        monkeypatch.setattr(dmod, "ping", lambda *a, **k: True)
+        monkeypatch.setattr(dmod, "try_forward", fake_forward)
+        with patch("sys.stdin", io.StringIO("spec")), \
+                patch("sys.stdout", io.StringIO()):
+            code = debate.main(None)
+        assert code == 0
+        assert seen["argv"] == argv[1:]
+
+    def test_setup_bedrock_enabled_returns_true(self, isolated_paths, clean_env):
+        """With bedrock enabled and models valid, setup_bedrock must report
+        bedrock_mode=True (False would silently route to plain litellm)."""
+        from adversarial_spec_amd import providers
+
+        providers.handle_bedrock_command("enable", None, None, "us-west-2")
+        providers.handle_bedrock_command("add-model", "llama-3-8b", None, None)
+        args = debate.create_parser().parse_args(["critique"])
+        mode, region = debate.setup_bedrock(args, ["llama-3-8b"])
+        assert mode is True
+        assert region == "us-west-2"
+
+    def test_send_final_rounds_default_is_1(self):
+        args = debate.create_parser().parse_args(["send-final"])
+        assert args.rounds == 1

@@ -142,3 +142,17 @@ class TestDaemonGuards:
         for r in results:
             assert r is not None and r[0] == 0
         assert srv.requests_served == 2
+
+    def test_cli_serve_stop_subcommand(self, live_daemon, capsys):
+        """`debate.py serve stop` must hit the stop dispatch (exit 0 and
+        the daemon actually goes down)."""
+        sock, _srv = live_daemon
+        assert cli.main(["serve", "stop"]) == 0
+        assert "stopped" in capsys.readouterr().out
+        import time as _t
+
+        for _ in range(50):
+            if not daemon.ping(sock):
+                break
+            _t.sleep(0.05)
+        assert not daemon.ping(sock)

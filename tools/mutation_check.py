@@ -31,6 +31,8 @@ TARGETS = {
         "tests/test_telegram.py", "tests/test_protocol_properties.py"],
     "adversarial_spec_amd/providers.py": ["tests/test_providers.py"],
     "adversarial_spec_amd/engine/backend.py": ["tests/test_backends.py"],
+    "adversarial_spec_amd/cli/debate.py": [
+        "tests/test_cli.py", "tests/test_daemon.py"],
 }
 
 CMP_SWAPS = {
@@ -116,6 +118,11 @@ EQUIVALENT_FRAGMENTS = (
     "# unreachable",   # defensive dead returns (argparse-choices covered)
     "indent=2",        # JSON pretty-print width: formatting only
     "head = user_message[:64]",  # stub round-parse window: simulation knob
+    # display-only truncation/separator widths in the CLI output
+    'ERROR - {r.error[:100]}',
+    "get_critique_summary(r",
+    '"=" * 30',
+    'default=60,',  # poll-timeout default: a tolerance, not a contract
 )
 
 
