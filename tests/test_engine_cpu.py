@@ -326,3 +326,76 @@ class TestEngineInternals:
         c3 = eng._get_cache(c1.max_seq + 1)  # grows: graph invalidated
         assert c3 is not c1 and eng._graph_state is None
         assert c3.max_seq % 2048 == 0 and c3.max_seq >= c1.max_seq + 1
+
+
+class TestStopScan:
+    """Direct unit tests for the BPE-safe [/SPEC] stop detector (the
+    engine tests only exercise it end-to-end)."""
+
+    def _tok(self):
+        from adversarial_spec_amd.engine.tokenizer import build_tokenizer
+
+        return build_tokenizer(1024)
+
+    def test_tag_split_across_pushes(self):
+        from adversarial_spec_amd.engine.local import _StopScan
+
+        tok = self._tok()
+        scan = _StopScan(tok)
+        ids = tok.encode("text before [/SP")
+        hits = [scan.push(t) for t in ids]
+        assert not any(hits)  # partial tag never fires
+        for t in tok.encode("EC] after"):
+            if scan.push(t):
+                return
+        raise AssertionError("split [/SPEC] tag never detected")
+
+    def test_open_tag_does_not_stop(self):
+        from adversarial_spec_amd.engine.local import _StopScan
+
+        tok = self._tok()
+        scan = _StopScan(tok)
+        assert not any(scan.push(t) for t in tok.encode("x [SPEC] body y"))
+
+    def test_window_slides(self):
+        from adversarial_spec_amd.engine.local import _StopScan
+
+        tok = self._tok()
+        scan = _StopScan(tok)
+        # long stream without the tag: window must stay bounded
+        for t in tok.encode("a" * 500):
+            assert not scan.push(t)
+        assert len(scan._ids) <= scan.WINDOW
+
+
+class TestArchitectureConstants:
+    """Pin the preset dimensions the kernel dispatch assumes: the decode
+    fusion requires hidden <= 8192 (gemv_res w32) and qkv N <= 8192 for
+    every non-TP fused arch, and K % 16 == 0 for the fp8 staging."""
+
+    def test_preset_shapes(self):
+        from adversarial_spec_amd.models.config import PRESETS
+
+        want = {
+            "llama-3-8b": (4096, 32, 32, 8, 14336, 128256),
+            "llama-3-70b": (8192, 80, 64, 8, 28672, 128256),
+            "mistral-7b": (4096, 32, 32, 8, 14336, 32000),
+        }
+        for name, (d, nl, h, kh, f, v) in want.items():
+            c = PRESETS[name]
+            assert (c.dim, c.n_layers, c.n_heads, c.n_kv_heads,
+                    c.ffn_dim, c.vocab_size) == (d, nl, h, kh, f, v), name
+
+    def test_fusion_dispatch_bounds(self):
+        from adversarial_spec_amd.models.config import PRESETS
+
+        for name, c in PRESETS.items():
+            qkv_n = (c.n_heads + 2 * c.n_kv_heads) * c.head_dim
+            if name == "llama-3-70b":
+                # 70B runs fp8/TP; its qkv exceeds the w32 norm kernel's
+                # range and dispatches to the 16-lane variant
+                assert qkv_n > 8192
+            else:
+                assert qkv_n <= 8192, f"{name}: qkv {qkv_n} breaks w32 fusion"
+            assert c.dim <= 8192, f"{name}: hidden breaks gemv_res w32"
+            assert c.dim % 16 == 0 and c.ffn_dim % 16 == 0, name
