@@ -129,18 +129,23 @@ class TestDaemonGuards:
         assert "dispatch error" in fwd[2]
 
     def test_concurrent_forwards(self, live_daemon):
-        """Two critiques in flight at once (the threaded server + per-
-        engine locks must not deadlock or cross wires)."""
+        """Two critiques in flight at once (the threaded server + request
+        lock must not deadlock or cross wires). Uses the raw socket
+        request: with an IN-PROCESS server the ADVSPEC_IN_DAEMON guard is
+        briefly visible to the client threads too (same environment — a
+        real daemon is a separate process), so try_forward could
+        legitimately decline mid-race."""
         import concurrent.futures as cf
 
         sock, srv = live_daemon
-        argv = ["critique", "--models", "stub/agree", "--json"]
+        payload = {"argv": ["critique", "--models", "stub/agree", "--json"],
+                   "stdin": SPEC}
         with cf.ThreadPoolExecutor(2) as pool:
-            futs = [pool.submit(daemon.try_forward, argv, SPEC, sock)
+            futs = [pool.submit(daemon._request, payload, sock)
                     for _ in range(2)]
             results = [f.result(timeout=60) for f in futs]
         for r in results:
-            assert r is not None and r[0] == 0
+            assert r is not None and r["code"] == 0
         assert srv.requests_served == 2
 
     def test_cli_serve_stop_subcommand(self, live_daemon, capsys):
